@@ -1,0 +1,23 @@
+"""demodel-amd: an MI355X-native model/dataset pull engine.
+
+Caching, syncing, distributing middleware for models and datasets —
+the capability contract of moeru-ai/demodel (reference: /root/reference,
+see cmd/demodel/main.go:59) rebuilt from scratch MI355X-first:
+
+  * TLS-MITM / HF_ENDPOINT / OLLAMA_HOST-compatible proxy front-end
+    (asyncio + per-host leaf certs minted natively via libcrypto),
+  * byte-compatible on-disk response cache (.cache/{sha256} + .meta,
+    bodies kept in original Content-Encoding — reference
+    CONTRIBUTING.md:53-153),
+  * a GPU landing pipeline: chunked blob downloads stream through a
+    pinned host ring into HBM3E via hipMemcpyAsync on side streams,
+    with hand-written CDNA4 (gfx950) HIP kernels for SHA-256 verify,
+    gzip/zstd inflate, safetensors tensor-scatter and GGUF q4 dequant,
+  * RCCL-over-xGMI fan-out: broadcast of a pulled model to all GPUs of
+    a node, and sharded pulls reassembled with all-gather overlapped
+    with the next chunk's download.
+"""
+
+__version__ = "0.1.0"
+
+from .config import Config, load_config  # noqa: F401

@@ -1,0 +1,133 @@
+"""The ``demodel`` command-line interface.
+
+Parity with the reference CLI (cmd/demodel/main.go:56-81):
+
+* ``demodel`` / ``demodel start`` — run the proxy (start.go:218-230)
+* ``demodel init``                — create/load the CA (init.go:156-168)
+* ``demodel export-ca [--for X]`` — print or install the CA
+  (export_ca.go:22-120; presets python-ssl, python-certifi, plus the
+  README-promised ``openssl`` the reference never implemented)
+
+New subcommands beyond the reference:
+
+* ``demodel pull <spec>``  — pull a model/dataset directly (no client
+  needed): hf://org/repo[@rev] or ollama://name[:tag], with optional GPU
+  landing (--gpu / --gpus).
+* ``demodel verify <spec>`` — re-verify cached blobs against digests.
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import sys
+
+from .config import load_config
+
+
+def _cmd_start(args) -> int:
+    from .proxy.server import run_proxy
+
+    cfg = load_config()
+    if args.port is not None:
+        cfg.port = args.port
+    try:
+        asyncio.run(run_proxy(cfg))
+    except KeyboardInterrupt:
+        pass
+    return 0
+
+
+def _cmd_init(args) -> int:
+    from . import ca as ca_mod
+
+    cfg = load_config()
+    ca_mod.read_or_new_ca(cfg.ca_use_ecdsa)
+    crt, key = ca_mod.cert_paths()
+    print(f"CA certificate: {crt}")
+    print(f"CA private key: {key}")
+    if args.install:
+        ok = ca_mod.install_system()
+        print("system trust store: " + ("installed" if ok else "skipped"))
+    return 0
+
+
+def _cmd_export_ca(args) -> int:
+    from . import ca as ca_mod
+
+    out = ca_mod.export_ca(args.dest)
+    if args.dest is None:
+        sys.stdout.write(out)
+    else:
+        print(f"CA exported to {out}")
+    return 0
+
+
+def _cmd_pull(args) -> int:
+    from .engine.pull import pull_spec
+
+    cfg = load_config()
+    result = asyncio.run(pull_spec(
+        args.spec, cfg,
+        endpoint=args.endpoint,
+        gpu=args.gpu,
+        out_dir=args.out,
+    ))
+    print(json.dumps(result, indent=2, default=str))
+    return 0
+
+
+def _cmd_verify(args) -> int:
+    from .engine.pull import verify_cache
+
+    cfg = load_config()
+    result = verify_cache(cfg, uri=args.uri)
+    print(json.dumps(result, indent=2))
+    return 0 if result.get("ok", False) else 1
+
+
+def main(argv: list[str] | None = None) -> int:
+    p = argparse.ArgumentParser(
+        prog="demodel",
+        description="Caching, syncing, distributing middleware for models "
+                    "and datasets — MI355X-native.")
+    sub = p.add_subparsers(dest="cmd")
+
+    sp = sub.add_parser("start", help="run the caching proxy")
+    sp.add_argument("--port", type=int, default=None)
+    sp.set_defaults(fn=_cmd_start)
+
+    ip = sub.add_parser("init", help="create or load the demodel CA")
+    ip.add_argument("--install", action="store_true",
+                    help="also install into the system trust store")
+    ip.set_defaults(fn=_cmd_init)
+
+    ep = sub.add_parser("export-ca", help="export the CA certificate")
+    ep.add_argument("--for", dest="dest", default=None,
+                    choices=["python-ssl", "python-certifi", "openssl"])
+    ep.set_defaults(fn=_cmd_export_ca)
+
+    pp = sub.add_parser("pull", help="pull a model/dataset through the engine")
+    pp.add_argument("spec", help="hf://org/repo[@rev] or ollama://name[:tag]")
+    pp.add_argument("--endpoint", default=None,
+                    help="upstream endpoint override (e.g. a fake origin)")
+    pp.add_argument("--gpu", action="store_true",
+                    help="land blobs in GPU HBM via the HIP pipeline")
+    pp.add_argument("--out", default=None, help="materialize files here")
+    pp.set_defaults(fn=_cmd_pull)
+
+    vp = sub.add_parser("verify", help="re-verify cached blobs")
+    vp.add_argument("--uri", default=None)
+    vp.set_defaults(fn=_cmd_verify)
+
+    args = p.parse_args(argv)
+    if not hasattr(args, "fn"):
+        # bare `demodel` runs the proxy, like the reference root command
+        # (main.go:68-70)
+        return _cmd_start(argparse.Namespace(port=None))
+    return args.fn(args)
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())

@@ -1,0 +1,481 @@
+"""The demodel-amd proxy front-end.
+
+Asyncio re-design of the reference's proxy core (goproxy-based ``start()``
+in cmd/demodel/start.go:167-216) with the same three behavioural hook
+points:
+
+1. **CONNECT policy** (start.go:183-196) — MITM vs blind tunnel, decided
+   by ``Config.should_mitm`` (env surface of main.go:15-42, bug-fixed).
+2. **Request hook** (start.go:197-200) — cache lookup; a hit answers the
+   client without contacting the origin.
+3. **Response hook** (start.go:201-204) — cache fill, streamed as a tee
+   while forwarding, plus structured transfer logs.
+
+Beyond the reference (which only proxied via HTTP_PROXY), the same
+listener also serves **origin-form requests directly**, so
+``HF_ENDPOINT=http://host:port`` and Ollama-registry-style clients work
+without any proxy env vars; in that reverse mode redirects are followed
+server-side so CDN hops stay inside the engine and blobs are cached under
+their canonical URI.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import ssl
+from urllib.parse import urlsplit
+
+from ..cache import CacheStore
+from ..certs import LeafStore
+from ..config import Config
+from ..utils.log import TransferLog, get_logger
+from . import http1
+from .http1 import ProtocolError, RequestHead, ResponseHead
+
+log = get_logger("proxy")
+
+_PROXY_HEADERS = ("proxy-connection", "proxy-authorization")
+
+
+class ProxyServer:
+    def __init__(self, cfg: Config, leafs: LeafStore | None = None,
+                 cache: CacheStore | None = None):
+        self.cfg = cfg
+        self.leafs = leafs
+        self.cache = cache or CacheStore(cfg.cache_dir,
+                                         chunk_bytes=cfg.chunk_bytes)
+        self.transfers = TransferLog()
+        self._server: asyncio.AbstractServer | None = None
+        self.port: int | None = None
+        # reverse-mode routing table: path-prefix -> upstream base
+        self.reverse_routes: list[tuple[str, str]] = [
+            ("/v2/", "https://registry.ollama.ai"),
+            ("/", "https://huggingface.co"),
+        ]
+        self._upstream_ssl: ssl.SSLContext | None = None
+
+    # ------------------------------------------------------------------ #
+    # lifecycle
+
+    async def start(self) -> int:
+        self._server = await asyncio.start_server(
+            self._handle_client, self.cfg.host, self.cfg.port,
+            limit=http1.MAX_HEAD,
+        )
+        self.port = self._server.sockets[0].getsockname()[1]
+        log.info("listening on %s:%d", self.cfg.host, self.port)
+        return self.port
+
+    async def close(self) -> None:
+        if self._server:
+            self._server.close()
+            await self._server.wait_closed()
+
+    def upstream_ssl(self) -> ssl.SSLContext:
+        if self._upstream_ssl is None:
+            ctx = ssl.create_default_context()
+            if self.cfg.upstream_cafile:
+                ctx.load_verify_locations(cafile=self.cfg.upstream_cafile)
+            if self.cfg.upstream_insecure:
+                ctx.check_hostname = False
+                ctx.verify_mode = ssl.CERT_NONE
+            self._upstream_ssl = ctx
+        return self._upstream_ssl
+
+    # ------------------------------------------------------------------ #
+    # connection handling
+
+    async def _handle_client(self, reader: asyncio.StreamReader,
+                             writer: asyncio.StreamWriter) -> None:
+        try:
+            await self._client_loop(reader, writer, tls_host=None)
+        except (ProtocolError, ConnectionResetError, BrokenPipeError,
+                asyncio.IncompleteReadError, ssl.SSLError) as e:
+            log.debug("client connection ended: %r", e)
+        except Exception:
+            log.exception("unhandled proxy error")
+        finally:
+            try:
+                writer.close()
+                await writer.wait_closed()
+            except Exception:
+                pass
+
+    async def _client_loop(self, reader, writer, tls_host: str | None):
+        """Serve HTTP/1.1 requests on one (possibly TLS) client conn."""
+        while True:
+            head = await http1.read_request_head(reader)
+            if head is None:
+                return
+            if head.method == "CONNECT":
+                handled = await self._handle_connect(head, reader, writer)
+                if not handled:
+                    return  # tunneled or refused; conn consumed either way
+                # MITM established — continue loop on the upgraded stream
+                reader, writer, tls_host = handled
+                continue
+            close = await self._handle_request(head, reader, writer, tls_host)
+            if close:
+                return
+
+    async def _handle_connect(self, head: RequestHead, reader, writer):
+        hostport = head.target
+        host = hostport.rsplit(":", 1)[0]
+        if self.cfg.should_mitm(hostport) and self.leafs is not None:
+            # Close the ClientHello race: stop plaintext reads before 200.
+            writer.transport.pause_reading()
+            writer.write(b"HTTP/1.1 200 Connection Established\r\n\r\n")
+            await writer.drain()
+            loop = asyncio.get_running_loop()
+            transport = writer.transport
+            protocol = transport.get_protocol()
+            ctx = self.leafs.server_context(host)
+            new_transport = await loop.start_tls(
+                transport, protocol, ctx, server_side=True)
+            new_writer = asyncio.StreamWriter(new_transport, protocol,
+                                              reader, loop)
+            log.debug("MITM established for %s", hostport)
+            return reader, new_writer, hostport
+        # blind tunnel (reference: goproxy.OkConnect path)
+        await self._tunnel(hostport, reader, writer)
+        return None
+
+    async def _tunnel(self, hostport: str, reader, writer) -> None:
+        host, _, port = hostport.rpartition(":")
+        try:
+            up_r, up_w = await asyncio.open_connection(host, int(port or 443))
+        except OSError as e:
+            writer.write(f"HTTP/1.1 502 Bad Gateway\r\n\r\n".encode())
+            await writer.drain()
+            log.info("tunnel to %s failed: %s", hostport, e)
+            return
+        writer.write(b"HTTP/1.1 200 Connection Established\r\n\r\n")
+        await writer.drain()
+
+        async def pump(src: asyncio.StreamReader, dst: asyncio.StreamWriter):
+            try:
+                while True:
+                    data = await src.read(http1.CHUNK)
+                    if not data:
+                        break
+                    dst.write(data)
+                    await dst.drain()
+            except (ConnectionResetError, BrokenPipeError):
+                pass
+            finally:
+                try:
+                    dst.write_eof()
+                except (OSError, RuntimeError):
+                    pass
+
+        await asyncio.gather(pump(reader, up_w), pump(up_r, writer))
+        up_w.close()
+
+    # ------------------------------------------------------------------ #
+    # request proxying
+
+    def _canonical_uri(self, head: RequestHead, tls_host: str | None
+                       ) -> tuple[str, str, int, bool, str]:
+        """Return (uri, host, port, is_tls, path) for this request."""
+        t = head.target
+        if t.startswith("http://") or t.startswith("https://"):
+            # absolute-form: plain forward proxy (reference's HTTP path)
+            u = urlsplit(t)
+            is_tls = u.scheme == "https"
+            port = u.port or (443 if is_tls else 80)
+            path = u.path or "/"
+            if u.query:
+                path += "?" + u.query
+            return self._mk_uri(u.scheme, u.hostname, port, path), \
+                u.hostname, port, is_tls, path
+        if tls_host is not None:
+            # origin-form inside a MITM'd tunnel
+            host, _, port_s = tls_host.rpartition(":")
+            port = int(port_s or 443)
+            return self._mk_uri("https", host, port, t), host, port, True, t
+        # origin-form on the plain listener: reverse mode (HF_ENDPOINT)
+        for prefix, base in self.reverse_routes:
+            if t.startswith(prefix):
+                u = urlsplit(base)
+                is_tls = u.scheme == "https"
+                port = u.port or (443 if is_tls else 80)
+                return self._mk_uri(u.scheme, u.hostname, port, t), \
+                    u.hostname, port, is_tls, t
+        raise ProtocolError(f"no reverse route for {t!r}")
+
+    @staticmethod
+    def _absolute_uri(location: str, base_uri: str) -> str:
+        """Resolve a Location header against the request URI."""
+        if location.startswith("http://") or location.startswith("https://"):
+            return location
+        b = urlsplit(base_uri)
+        if location.startswith("/"):
+            return f"{b.scheme}://{b.netloc}{location}"
+        base_path = b.path.rsplit("/", 1)[0]
+        return f"{b.scheme}://{b.netloc}{base_path}/{location}"
+
+    @staticmethod
+    def _mk_uri(scheme: str, host: str, port: int, path: str) -> str:
+        default = 443 if scheme == "https" else 80
+        netloc = host if port == default else f"{host}:{port}"
+        return f"{scheme}://{netloc}{path}"
+
+    async def _handle_request(self, head: RequestHead, reader, writer,
+                              tls_host: str | None) -> bool:
+        uri, host, port, is_tls, path = self._canonical_uri(head, tls_host)
+        reverse_mode = tls_host is None and "://" not in head.target
+
+        # drain request body up-front (pulls have none; PUT/POST pass through
+        # un-cached below via buffered body)
+        req_mode, req_len = http1.body_mode(head, method=head.method)
+        req_body = b""
+        if req_mode != "none":
+            parts = []
+            async for chunk in http1.iter_body(reader, req_mode, req_len):
+                parts.append(chunk)
+            req_body = b"".join(parts)
+
+        client_wants_close = (head.get("connection", "").lower() == "close")
+
+        # ---- hook 2: cache lookup --------------------------------------
+        # In reverse mode a cached redirect chain is followed inside the
+        # cache, so a fully cached pull replays with zero upstream traffic.
+        if self.cache.cacheable(head.method, 200):
+            lookup_uri = uri
+            for _ in range(6):
+                hit = self.cache.lookup(lookup_uri)
+                if hit is None:
+                    break
+                if reverse_mode and 300 <= hit.status < 400:
+                    loc = dict((k.lower(), v) for k, v in hit.headers
+                               ).get("location")
+                    if not loc:
+                        break
+                    lookup_uri = self._absolute_uri(loc, lookup_uri)
+                    continue
+                await self._serve_cached(hit, writer, head)
+                self.transfers.record(event="hit", uri=lookup_uri,
+                                      bytes=hit.body_size)
+                log.info("HIT  %s %s (%d bytes)", head.method, lookup_uri,
+                         hit.body_size)
+                return client_wants_close
+        # ------------------------------------------------------------------
+
+        redirects = 0
+        carry: dict[str, str] = {}
+        while True:
+            result = await self._forward_once(
+                head, req_body, uri, host, port, is_tls, path, writer,
+                follow_redirect=reverse_mode and redirects < 5,
+                carry=carry,
+            )
+            if result is None:
+                return client_wants_close
+            # internal redirect follow (reverse mode only)
+            uri, host, port, is_tls, path = result
+            redirects += 1
+            head = RequestHead(head.method, path, head.version,
+                               [(k, v) for k, v in head.headers
+                                if k.lower() not in
+                                ("host", "content-length",
+                                 "transfer-encoding", "authorization")])
+            head.replace("Host",
+                         host if port in (80, 443) else f"{host}:{port}")
+            req_body = b""
+
+    # Metadata headers the HF client reads off the hub's redirect hop; when
+    # the proxy follows the hop internally they must surface on the final
+    # response (and be cached with it) or hf_hub_download refuses the file.
+    _CARRY_HEADERS = ("x-repo-commit", "x-linked-etag", "x-linked-size",
+                      "etag", "x-request-id")
+
+    async def _forward_once(self, head: RequestHead, req_body: bytes,
+                            uri: str, host: str, port: int, is_tls: bool,
+                            path: str, writer,
+                            follow_redirect: bool,
+                            carry: dict[str, str] | None = None):
+        """Forward one request upstream; returns redirect target or None."""
+        up_head = RequestHead(head.method, path, "HTTP/1.1",
+                              list(head.headers))
+        for h in _PROXY_HEADERS:
+            up_head.remove(h)
+        up_head.replace("Host",
+                        host if port in (80, 443) else f"{host}:{port}")
+        up_head.replace("Connection", "close")
+        up_head.remove("accept-encoding")
+        # identity keeps cached bytes byte-exact AND client-agnostic; clients
+        # that asked for gzip still get valid identity responses.
+        if req_body:
+            up_head.replace("Content-Length", str(len(req_body)))
+
+        sslctx = self.upstream_ssl() if is_tls else None
+        try:
+            up_r, up_w = await asyncio.open_connection(
+                host, port, ssl=sslctx,
+                server_hostname=host if is_tls else None)
+        except (OSError, ssl.SSLError) as e:
+            log.info("upstream connect %s:%d failed: %s", host, port, e)
+            err = (b"HTTP/1.1 502 Bad Gateway\r\nContent-Length: 0\r\n"
+                   b"Connection: keep-alive\r\n\r\n")
+            writer.write(err)
+            await writer.drain()
+            return None
+
+        try:
+            up_w.write(http1.serialize_request(up_head))
+            if req_body:
+                up_w.write(req_body)
+            await up_w.drain()
+            resp = await http1.read_response_head(up_r)
+
+            if (follow_redirect and resp.status in (301, 302, 303, 307, 308)
+                    and head.method in ("GET", "HEAD")):
+                loc = resp.get("location")
+                if loc:
+                    # harvest metadata headers to re-surface on the final hop
+                    if carry is not None:
+                        for k, v in resp.headers:
+                            if (k.lower() in self._CARRY_HEADERS
+                                    and k.lower() not in carry):
+                                carry[k.lower()] = v
+                    # cache the redirect hop so a later replay can walk the
+                    # chain offline, then follow it internally
+                    mode, length = http1.body_mode(resp, status=resp.status)
+                    if head.method == "HEAD":
+                        mode, length = "none", 0
+                    cw = None
+                    if (head.method == "GET"
+                            and self.cache.cacheable("GET", resp.status)):
+                        cw = self.cache.writer(uri, resp.status, resp.reason,
+                                               resp.headers)
+                    async for chunk in http1.iter_body(up_r, mode, length):
+                        if cw:
+                            cw.write(chunk)
+                    if cw:
+                        cw.finalize()
+                    target = self._absolute_uri(loc, uri)
+                    u = urlsplit(target)
+                    r_tls = u.scheme == "https"
+                    r_port = u.port or (443 if r_tls else 80)
+                    r_path = u.path or "/"
+                    if u.query:
+                        r_path += "?" + u.query
+                    return (self._mk_uri(u.scheme, u.hostname, r_port,
+                                         r_path),
+                            u.hostname, r_port, r_tls, r_path)
+
+            if carry:
+                present = {k.lower() for k, _ in resp.headers}
+                for k, v in carry.items():
+                    if k not in present:
+                        resp.headers.append((k, v))
+            await self._stream_response(head, resp, up_r, writer, uri)
+            return None
+        finally:
+            try:
+                up_w.close()
+            except Exception:
+                pass
+
+    async def _stream_response(self, req: RequestHead, resp: ResponseHead,
+                               up_r, writer, uri: str) -> None:
+        mode, length = http1.body_mode(resp, status=resp.status)
+        if req.method == "HEAD":
+            mode, length = "none", 0
+
+        cache_writer = None
+        if (self.cache.cacheable(req.method, resp.status)
+                and req.get("range") is None):
+            cache_writer = self.cache.writer(uri, resp.status, resp.reason,
+                                             resp.headers)
+
+        out = ResponseHead("HTTP/1.1", resp.status, resp.reason,
+                           [(k, v) for k, v in resp.headers
+                            if k.lower() not in ("connection",
+                                                 "keep-alive",
+                                                 "transfer-encoding")])
+        total = 0
+        try:
+            if mode in ("chunked", "eof"):
+                # re-frame as chunked toward the client (length unknown)
+                out.replace("Transfer-Encoding", "chunked")
+                out.replace("Connection", "keep-alive")
+                writer.write(http1.serialize_response(out))
+                if req.method != "HEAD":
+                    async for chunk in http1.iter_body(up_r, mode, length):
+                        total += len(chunk)
+                        if cache_writer:
+                            cache_writer.write(chunk)
+                        writer.write(b"%x\r\n" % len(chunk) + chunk + b"\r\n")
+                        await writer.drain()
+                    writer.write(b"0\r\n\r\n")
+                else:
+                    writer.write(b"0\r\n\r\n")
+                await writer.drain()
+            else:
+                out.replace("Connection", "keep-alive")
+                writer.write(http1.serialize_response(out))
+                if req.method != "HEAD":
+                    async for chunk in http1.iter_body(up_r, mode, length):
+                        total += len(chunk)
+                        if cache_writer:
+                            cache_writer.write(chunk)
+                        writer.write(chunk)
+                        await writer.drain()
+                await writer.drain()
+        except BaseException:
+            if cache_writer:
+                cache_writer.abort()
+            raise
+        # HEAD responses carry no body: don't poison the cache with an
+        # empty entry for a URI whose GET has content.
+        if cache_writer:
+            if req.method == "HEAD":
+                cache_writer.abort()
+            else:
+                cache_writer.finalize()
+        # ---- hook 3: response hook ------------------------------------
+        self.transfers.record(event="miss", uri=uri, status=resp.status,
+                              bytes=total)
+        log.info("MISS %s %s -> %d (%d bytes)", req.method, uri,
+                 resp.status, total)
+
+    async def _serve_cached(self, hit, writer, req: RequestHead) -> None:
+        out = ResponseHead("HTTP/1.1", hit.status, hit.reason or "OK",
+                           [(k, v) for k, v in hit.headers
+                            if k.lower() not in ("connection", "keep-alive",
+                                                 "transfer-encoding",
+                                                 "content-length")])
+        out.replace("Content-Length", str(hit.body_size))
+        out.replace("Connection", "keep-alive")
+        out.replace("X-Demodel-Cache", "HIT")
+        writer.write(http1.serialize_response(out))
+        if req.method != "HEAD":
+            loop = asyncio.get_running_loop()
+            await writer.drain()
+            with hit.open_body() as f:
+                # zero-copy page-cache -> socket where possible
+                try:
+                    await loop.sendfile(writer.transport, f, fallback=True)
+                except (NotImplementedError, RuntimeError, OSError):
+                    f.seek(0)
+                    while True:
+                        data = f.read(http1.CHUNK)
+                        if not data:
+                            break
+                        writer.write(data)
+                        await writer.drain()
+        await writer.drain()
+
+
+async def run_proxy(cfg: Config) -> None:
+    from . import server as _self  # noqa: F401
+    from ..ca import read_or_new_ca
+
+    ca = read_or_new_ca(cfg.ca_use_ecdsa)
+    leafs = LeafStore(ca)
+    srv = ProxyServer(cfg, leafs=leafs)
+    await srv.start()
+    assert srv._server is not None
+    async with srv._server:
+        await srv._server.serve_forever()

@@ -1,0 +1,319 @@
+"""In-process fake Hugging Face Hub / Ollama registry origin.
+
+Serves the minimal protocol surface the real clients use, so the proxy and
+pull engine can be exercised with zero network (SURVEY.md §4's test plan;
+the protocol shapes come from the reference's worked example —
+CONTRIBUTING.md:127-153 for the Ollama manifest — and the public HF Hub
+HTTP API as exercised by huggingface_hub):
+
+* HF:  ``GET /api/models/{repo}[/revision/{rev}]`` (repo info + siblings),
+       ``GET|HEAD /{repo}/resolve/{rev}/{file}`` with
+       ``X-Repo-Commit`` / ``ETag`` / ``Content-Length`` headers and an
+       optional 302 hop to a /cdn/ path (mirrors the hub->CDN redirect),
+       Range support.
+* Ollama: ``GET /v2/{name}/manifests/{tag}`` (Docker-v2 manifest JSON,
+       optionally gzip Content-Encoding like the reference capture),
+       ``GET|HEAD /v2/{name}/blobs/sha256:{digest}``.
+
+Also doubles as the synthetic origin for bench.py: file-backed blobs are
+served via sendfile so the origin is not the bottleneck.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import gzip
+import hashlib
+import json
+import os
+import ssl
+
+from ..proxy import http1
+from ..proxy.http1 import RequestHead, ResponseHead
+
+
+class FakeOrigin:
+    def __init__(self, root: str, tls_ctx: ssl.SSLContext | None = None,
+                 redirect_blobs: bool = True, gzip_manifests: bool = True):
+        self.root = root  # directory holding blob files
+        self.tls_ctx = tls_ctx
+        self.redirect_blobs = redirect_blobs
+        self.gzip_manifests = gzip_manifests
+        # repo_id -> {"sha": commit, "files": {rfilename: abspath}}
+        self.hf_repos: dict[str, dict] = {}
+        # name -> {"tag": ..., "manifest": dict}; blobs by digest
+        self.ollama_manifests: dict[tuple[str, str], dict] = {}
+        self.blobs: dict[str, str] = {}  # "sha256:<hex>" -> abspath
+        self.port: int | None = None
+        self._server = None
+        self.requests: list[str] = []  # log of "<METHOD> <path>"
+
+    # ------------------------------------------------------------------ #
+    # content registration
+
+    def add_hf_repo(self, repo_id: str, files: dict[str, str],
+                    commit: str | None = None) -> str:
+        commit = commit or hashlib.sha1(repo_id.encode()).hexdigest()
+        self.hf_repos[repo_id] = {"sha": commit, "files": dict(files)}
+        return commit
+
+    def add_ollama_model(self, name: str, tag: str,
+                         layers: list[tuple[str, str]]) -> dict:
+        """layers: list of (media_type, blob_path). Returns the manifest."""
+        def register(path: str) -> tuple[str, int]:
+            h = hashlib.sha256()
+            with open(path, "rb") as f:
+                for chunk in iter(lambda: f.read(1 << 20), b""):
+                    h.update(chunk)
+            digest = "sha256:" + h.hexdigest()
+            self.blobs[digest] = path
+            return digest, os.path.getsize(path)
+
+        config_path = os.path.join(self.root, f"{name.replace('/', '_')}-config.json")
+        with open(config_path, "w") as f:
+            json.dump({"model_format": "gguf", "model_family": "llama"}, f)
+        cfg_digest, cfg_size = register(config_path)
+        manifest = {
+            "schemaVersion": 2,
+            "mediaType": "application/vnd.docker.distribution.manifest.v2+json",
+            "config": {
+                "mediaType": "application/vnd.docker.container.image.v1+json",
+                "digest": cfg_digest,
+                "size": cfg_size,
+            },
+            "layers": [],
+        }
+        for media_type, path in layers:
+            digest, size = register(path)
+            manifest["layers"].append(
+                {"mediaType": media_type, "digest": digest, "size": size})
+        self.ollama_manifests[(name, tag)] = manifest
+        return manifest
+
+    # ------------------------------------------------------------------ #
+    # server
+
+    async def start(self, host: str = "127.0.0.1", port: int = 0) -> int:
+        self._server = await asyncio.start_server(
+            self._handle, host, port, ssl=self.tls_ctx, limit=http1.MAX_HEAD)
+        self.port = self._server.sockets[0].getsockname()[1]
+        return self.port
+
+    async def close(self) -> None:
+        if self._server:
+            self._server.close()
+            await self._server.wait_closed()
+
+    async def _handle(self, reader, writer):
+        try:
+            while True:
+                head = await http1.read_request_head(reader)
+                if head is None:
+                    return
+                self.requests.append(f"{head.method} {head.target}")
+                await self._dispatch(head, writer)
+                if head.get("connection", "").lower() == "close":
+                    return
+        except (http1.ProtocolError, ConnectionResetError,
+                asyncio.IncompleteReadError, ssl.SSLError):
+            pass
+        finally:
+            try:
+                writer.close()
+                await writer.wait_closed()
+            except Exception:
+                pass
+
+    async def _dispatch(self, head: RequestHead, writer):
+        path = head.target.split("?")[0]
+        parts = [p for p in path.split("/") if p]
+
+        # ---- Ollama registry v2 ----
+        if parts and parts[0] == "v2":
+            if len(parts) >= 4 and parts[-2] == "manifests":
+                name = "/".join(parts[1:-2])
+                tag = parts[-1]
+                m = self.ollama_manifests.get((name, tag))
+                if m is None:
+                    return await self._error(writer, 404)
+                body = json.dumps(m).encode()
+                hdrs = [("Content-Type",
+                         "application/vnd.docker.distribution.manifest.v2+json"),
+                        ("Docker-Content-Digest",
+                         "sha256:" + hashlib.sha256(body).hexdigest())]
+                if self.gzip_manifests:
+                    body = gzip.compress(body)
+                    hdrs.append(("Content-Encoding", "gzip"))
+                return await self._reply(writer, head, 200, hdrs, body)
+            if len(parts) >= 4 and parts[-2] == "blobs":
+                digest = parts[-1]
+                blob = self.blobs.get(digest)
+                if blob is None:
+                    return await self._error(writer, 404)
+                return await self._serve_file(writer, head, blob, etag=digest)
+            return await self._error(writer, 404)
+
+        # ---- HF api: /api/models/{repo}/tree/{rev} (paginated file list,
+        # what huggingface_hub 1.x snapshot_download walks) ----
+        if parts[:2] == ["api", "models"] and "tree" in parts:
+            i = parts.index("tree")
+            repo_id = "/".join(parts[2:i])
+            repo = self.hf_repos.get(repo_id)
+            if repo is None:
+                return await self._error(writer, 404)
+            items = []
+            for name in sorted(repo["files"]):
+                fpath = repo["files"][name]
+                size = os.path.getsize(fpath)
+                etag = self._file_etag(fpath)
+                items.append({
+                    "type": "file", "path": name, "size": size,
+                    "oid": etag[:40],
+                    "lfs": {"oid": etag, "size": size,
+                            "pointerSize": 134},
+                })
+            body = json.dumps(items).encode()
+            return await self._reply(
+                writer, head, 200,
+                [("Content-Type", "application/json")], body)
+
+        # ---- HF api: /api/models/{repo}[/revision/{rev}] ----
+        if parts[:2] == ["api", "models"]:
+            rest = parts[2:]
+            rev = "main"
+            if "revision" in rest:
+                i = rest.index("revision")
+                rev = rest[i + 1] if i + 1 < len(rest) else "main"
+                rest = rest[:i]
+            repo_id = "/".join(rest)
+            repo = self.hf_repos.get(repo_id)
+            if repo is None:
+                return await self._error(writer, 404)
+            info = {
+                "_id": repo["sha"][:24], "id": repo_id,
+                "modelId": repo_id, "sha": repo["sha"],
+                "private": False, "gated": False, "disabled": False,
+                "downloads": 0, "likes": 0, "tags": [],
+                "siblings": [{"rfilename": name}
+                             for name in sorted(repo["files"])],
+            }
+            body = json.dumps(info).encode()
+            return await self._reply(
+                writer, head, 200,
+                [("Content-Type", "application/json")], body)
+
+        # ---- HF resolve: /{repo}/resolve/{rev}/{path...} ----
+        if "resolve" in parts:
+            i = parts.index("resolve")
+            repo_id = "/".join(parts[:i])
+            rev = parts[i + 1] if i + 1 < len(parts) else "main"
+            fname = "/".join(parts[i + 2:])
+            repo = self.hf_repos.get(repo_id)
+            if repo is None or fname not in repo["files"]:
+                return await self._error(writer, 404)
+            fpath = repo["files"][fname]
+            etag = self._file_etag(fpath)
+            extra = [("X-Repo-Commit", repo["sha"]),
+                     ("X-Linked-Etag", f'"{etag}"'),
+                     ("X-Linked-Size", str(os.path.getsize(fpath)))]
+            if self.redirect_blobs:
+                loc = f"/cdn/{repo_id}/{repo['sha']}/{fname}"
+                return await self._reply(
+                    writer, head, 302,
+                    extra + [("Location", loc),
+                             ("Content-Type", "text/plain")],
+                    b"redirect")
+            return await self._serve_file(writer, head, fpath, etag=etag,
+                                          extra=extra)
+
+        # ---- CDN: /cdn/{repo}/{commit}/{path...} ----
+        if parts and parts[0] == "cdn":
+            repo_id = "/".join(parts[1:3])
+            fname = "/".join(parts[4:])
+            repo = self.hf_repos.get(repo_id)
+            if repo is None or fname not in repo["files"]:
+                return await self._error(writer, 404)
+            fpath = repo["files"][fname]
+            return await self._serve_file(writer, head, fpath,
+                                          etag=self._file_etag(fpath))
+
+        return await self._error(writer, 404)
+
+    _etag_cache: dict[tuple[str, float], str] = {}
+
+    def _file_etag(self, path: str) -> str:
+        key = (path, os.path.getmtime(path))
+        hit = self._etag_cache.get(key)
+        if hit:
+            return hit
+        h = hashlib.sha256()
+        with open(path, "rb") as f:
+            for chunk in iter(lambda: f.read(1 << 20), b""):
+                h.update(chunk)
+        self._etag_cache[key] = h.hexdigest()
+        return self._etag_cache[key]
+
+    async def _error(self, writer, status: int):
+        body = json.dumps({"error": status}).encode()
+        head = ResponseHead("HTTP/1.1", status, "Error",
+                            [("Content-Type", "application/json"),
+                             ("Content-Length", str(len(body)))])
+        writer.write(http1.serialize_response(head) + body)
+        await writer.drain()
+
+    async def _reply(self, writer, req: RequestHead, status: int,
+                     headers: list[tuple[str, str]], body: bytes):
+        headers = headers + [("Content-Length", str(len(body))),
+                             ("Accept-Ranges", "bytes")]
+        reason = {200: "OK", 302: "Found"}.get(status, "OK")
+        writer.write(http1.serialize_response(
+            ResponseHead("HTTP/1.1", status, reason, headers)))
+        if req.method != "HEAD":
+            writer.write(body)
+        await writer.drain()
+
+    async def _serve_file(self, writer, req: RequestHead, path: str,
+                          etag: str, extra: list[tuple[str, str]] | None = None):
+        size = os.path.getsize(path)
+        start, end = 0, size - 1
+        status, reason = 200, "OK"
+        rng = req.get("range")
+        if rng and rng.startswith("bytes="):
+            spec = rng[len("bytes="):].split(",")[0]
+            s, _, e = spec.partition("-")
+            if s:
+                start = int(s)
+                end = int(e) if e else size - 1
+            else:  # suffix range
+                start = max(0, size - int(e))
+            status, reason = 206, "Partial Content"
+        length = end - start + 1
+        headers = [("Content-Type", "application/octet-stream"),
+                   ("Content-Length", str(length)),
+                   ("Accept-Ranges", "bytes"),
+                   ("ETag", f'"{etag}"')] + (extra or [])
+        if status == 206:
+            headers.append(("Content-Range", f"bytes {start}-{end}/{size}"))
+        writer.write(http1.serialize_response(
+            ResponseHead("HTTP/1.1", status, reason, headers)))
+        await writer.drain()
+        if req.method == "HEAD":
+            return
+        loop = asyncio.get_running_loop()
+        with open(path, "rb") as f:
+            f.seek(start)
+            if self.tls_ctx is None:
+                try:
+                    await loop.sendfile(writer.transport, f, offset=start,
+                                        count=length, fallback=True)
+                    return
+                except (NotImplementedError, OSError):
+                    f.seek(start)
+            remaining = length
+            while remaining > 0:
+                data = f.read(min(http1.CHUNK, remaining))
+                if not data:
+                    break
+                remaining -= len(data)
+                writer.write(data)
+                await writer.drain()
