@@ -1,0 +1,180 @@
+#!/usr/bin/env python3
+"""demodel-amd flagship benchmark: Llama-3-8B synthetic safetensors pull
+-> HBM with GPU chunk-verified landing (BASELINE.json configs 2/3).
+
+Each *step* pulls the whole model (all shards, ~16 GB bf16) from a
+per-rank loopback origin through the engine's landing pipeline into this
+rank's HBM, with GPU SHA-256 chunk verification and zero-copy tensor
+views materialized.  value = whole-job GB/s into HBM (sum over ranks).
+
+Run directly (1 GPU) or under torch.distributed.run with one rank per
+GPU; weak scaling (per-rank work fixed).
+"""
+
+import argparse
+import json
+import os
+import pathlib
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                "tests"))
+
+GEOMS = {
+    "llama3-8b": ("LLAMA3_8B", 4),
+    "llama3-70b": ("LLAMA3_70B", 16),
+    "tiny": (None, 2),
+}
+
+
+def log(msg):
+    print(f"[bench r{os.environ.get('RANK', '0')}] {msg}", file=sys.stderr,
+          flush=True)
+
+
+def make_model_files(model: str, data_dir: str):
+    from demodel_amd.testing import synth
+
+    if model == "tiny":
+        geom = {"hidden": 256, "inter": 688, "layers": 4, "heads": 8,
+                "kv_heads": 4, "vocab": 32000}
+        return synth.write_shards(data_dir, geom, 2)
+    geom_name, n_shards = GEOMS[model]
+    geom = getattr(synth, geom_name)
+    return synth.write_shards(data_dir, geom, n_shards)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--model", default="llama3-8b",
+                    choices=list(GEOMS))
+    ap.add_argument("--workers", type=int, default=4)
+    ap.add_argument("--verify", default="chunked",
+                    choices=["chunked", "digest", "gpu-digest", "off"])
+    ap.add_argument("--slab-mib", type=int, default=64)
+    ap.add_argument("--data-dir", default=None)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    import torch
+
+    have_gpu = torch.cuda.is_available()
+    if have_gpu:
+        torch.cuda.set_device(local_rank)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group(
+            backend="nccl" if have_gpu else "gloo")
+
+    from demodel_amd.engine import pull as pull_mod
+    from demodel_amd.engine.pull import LanderPool
+    from demodel_amd.testing.origin import FakeOrigin
+    from helpers import LoopThread
+
+    # ---- setup (untimed): synth model + per-rank loopback origin -------
+    data_dir = args.data_dir or os.path.join(
+        os.environ.get("TMPDIR", "/tmp"), f"demodel_bench_{args.model}")
+    t = time.time()
+    files = make_model_files(args.model, data_dir)
+    log(f"model files ready in {time.time() - t:.1f}s "
+        f"({sum(os.path.getsize(p) for p in files.values()) / 1e9:.2f} GB)")
+
+    lt = LoopThread()
+    origin = FakeOrigin(data_dir, redirect_blobs=True)
+    origin.add_hf_repo("bench/model", files)
+    port = lt.call(origin.start())
+    endpoint = f"http://127.0.0.1:{port}"
+    total_bytes = sum(os.path.getsize(p) for p in files.values())
+
+    landers = LanderPool(local_rank if have_gpu else 0,
+                         slab_bytes=args.slab_mib << 20)
+
+    def sync():
+        if have_gpu:
+            torch.cuda.synchronize()
+        if dist:
+            dist.barrier()
+
+    def one_step():
+        res = pull_mod.pull_hf(
+            "bench/model", endpoint=endpoint, workers=args.workers,
+            verify=args.verify, landers=landers)
+        assert res.total_bytes == total_bytes, res.total_bytes
+        # model-ready: materialize the tensor views
+        n_t = len(res.tensors()) if args.model != "tiny" or True else 0
+        if have_gpu:
+            torch.cuda.synchronize()
+        return res, n_t
+
+    for i in range(args.warmup):
+        res, _ = one_step()
+        log(f"warmup {i}: {res.seconds_to_ready:.2f}s "
+            f"({res.gbps:.2f} GB/s)")
+
+    sync()
+    t0 = time.perf_counter()
+    per_step = []
+    for i in range(args.steps):
+        res, _ = one_step()
+        per_step.append(res.seconds_to_ready)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    if dist:
+        et = torch.tensor([elapsed])
+        if have_gpu:
+            et = et.cuda()
+        dist.all_reduce(et, op=dist.ReduceOp.MAX)
+        elapsed = float(et.item())
+
+    steps_bytes = total_bytes * args.steps * world
+    value = steps_bytes / elapsed / 1e9
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "pull_gbps_into_hbm",
+            "value": round(value, 3),
+            "unit": "GB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "Llama-3-8B" if args.model == "llama3-8b"
+                else args.model,
+                "bytes_per_model": total_bytes,
+                "files": len(files),
+                "verify": args.verify,
+                "seconds_to_ready": round(ms_per_step / 1000.0, 3),
+                "parallelism": f"independent-pull dp{world}",
+                "device": "cuda" if have_gpu else "cpu",
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    lt.call(origin.close())
+    lt.stop()
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -66,14 +66,13 @@ def _cmd_export_ca(args) -> int:
 
 def _cmd_pull(args) -> int:
     from .engine.pull import pull_spec
+    from .gpu import have_gpu
 
-    cfg = load_config()
-    result = asyncio.run(pull_spec(
-        args.spec, cfg,
-        endpoint=args.endpoint,
-        gpu=args.gpu,
-        out_dir=args.out,
-    ))
+    if args.gpu and not have_gpu():
+        print("error: --gpu requested but no AMD GPU is available",
+              file=sys.stderr)
+        return 1
+    result = pull_spec(args.spec, endpoint=args.endpoint, out_dir=args.out)
     print(json.dumps(result, indent=2, default=str))
     return 0
 

@@ -1,0 +1,195 @@
+// CDNA4 GGUF dequantization kernels: GGML quant
+// blocks -> bf16, run while a pulled Ollama blob (reference protocol:
+// CONTRIBUTING.md:127-153, application/vnd.ollama.image.model layers)
+// still sits in HBM.  Block layouts follow the public GGML formats:
+//   q4_0:  18 B / 32 elems  (f16 d, 16 B nibbles; v = d*(q-8))
+//   q8_0:  34 B / 32 elems  (f16 d, 32 int8;     v = d*q)
+//   q4_K: 144 B / 256 elems (f16 d, f16 dmin, 12 B 6-bit scales, 128 B)
+//   q6_K: 210 B / 256 elems (128 B ql, 64 B qh, 16 int8 scales, f16 d)
+//
+// Mapping: q4_0/q8_0 one lane per 32-elem block; q4_K/q6_K one wave64 per
+// 256-elem superblock (4 elems/lane) — scales decoded per-lane from the
+// packed bytes, no LDS needed, stores coalesced in 64 B/lane groups.
+
+#include <hip/hip_runtime.h>
+
+namespace {
+
+__device__ __forceinline__ float f16_to_f32(const uint8_t* p) {
+  uint16_t h = (uint16_t)p[0] | ((uint16_t)p[1] << 8);
+  uint32_t sign = (uint32_t)(h & 0x8000u) << 16;
+  uint32_t exp = (h >> 10) & 0x1F;
+  uint32_t man = h & 0x3FF;
+  uint32_t f;
+  if (exp == 0) {
+    if (man == 0) {
+      f = sign;
+    } else {
+      int e = 0;
+      while (!(man & 0x400)) { man <<= 1; ++e; }
+      f = sign | ((uint32_t)(127 - 15 - e) << 23) | ((man & 0x3FF) << 13);
+    }
+  } else if (exp == 31) {
+    f = sign | 0x7F800000u | (man << 13);
+  } else {
+    f = sign | ((exp - 15 + 127) << 23) | (man << 13);
+  }
+  return __uint_as_float(f);
+}
+
+__device__ __forceinline__ uint16_t f32_to_bf16(float v) {
+  uint32_t x = __float_as_uint(v);
+  if ((x & 0x7fffffffu) > 0x7f800000u) return (uint16_t)((x >> 16) | 0x40);
+  uint32_t lsb = (x >> 16) & 1u;
+  return (uint16_t)((x + 0x7fffu + lsb) >> 16);
+}
+
+// ---- q4_0: one lane per block ------------------------------------------
+
+__global__ void dequant_q4_0(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_blocks) {
+  int64_t b0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t b = b0; b < n_blocks; b += stride) {
+    const uint8_t* q = src + b * 18;
+    float d = f16_to_f32(q);
+    uint16_t* o = dst + b * 32;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      uint8_t byte = q[2 + j];
+      o[j] = f32_to_bf16(d * (float)((int)(byte & 0xF) - 8));
+      o[j + 16] = f32_to_bf16(d * (float)((int)(byte >> 4) - 8));
+    }
+  }
+}
+
+// ---- q8_0: one lane per block ------------------------------------------
+
+__global__ void dequant_q8_0(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_blocks) {
+  int64_t b0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t b = b0; b < n_blocks; b += stride) {
+    const uint8_t* q = src + b * 34;
+    float d = f16_to_f32(q);
+    uint16_t* o = dst + b * 32;
+#pragma unroll
+    for (int j = 0; j < 32; ++j)
+      o[j] = f32_to_bf16(d * (float)(int8_t)q[2 + j]);
+  }
+}
+
+// ---- q4_K: one wave per 256-elem superblock, 4 elems/lane ---------------
+// scale/min unpack follows GGML get_scale_min_k4.
+
+__device__ __forceinline__ void scale_min_k4(int j, const uint8_t* s,
+                                             uint8_t* d, uint8_t* m) {
+  if (j < 4) {
+    *d = s[j] & 63;
+    *m = s[j + 4] & 63;
+  } else {
+    *d = (s[j + 4] & 0xF) | ((s[j - 4] >> 6) << 4);
+    *m = (s[j + 4] >> 4) | ((s[j] >> 6) << 4);
+  }
+}
+
+__global__ void dequant_q4_K(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_sblocks) {
+  int64_t sb0 = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  int64_t stride = (int64_t)gridDim.x * (blockDim.x / 64);
+  int lane = threadIdx.x & 63;
+  for (int64_t sb = sb0; sb < n_sblocks; sb += stride) {
+    const uint8_t* blk = src + sb * 144;
+    float d = f16_to_f32(blk);
+    float dmin = f16_to_f32(blk + 2);
+    const uint8_t* scales = blk + 4;
+    const uint8_t* qs = blk + 16;
+    uint16_t* o = dst + sb * 256;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int j = lane * 4 + r * 64;   // element index 0..255
+      int sub = j >> 5;            // 32-elem sub-block 0..7
+      uint8_t sc, mn;
+      scale_min_k4(sub, scales, &sc, &mn);
+      int l = j & 31;
+      int pair = j >> 6;           // 64-elem pair index 0..3
+      uint8_t byte = qs[pair * 32 + l];
+      int nib = ((j >> 5) & 1) ? (byte >> 4) : (byte & 0xF);
+      o[j] = f32_to_bf16(d * (float)sc * (float)nib -
+                         dmin * (float)mn);
+    }
+  }
+}
+
+// ---- q6_K: one wave per superblock, 4 elems/lane ------------------------
+
+__global__ void dequant_q6_K(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_sblocks) {
+  int64_t sb0 = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  int64_t stride = (int64_t)gridDim.x * (blockDim.x / 64);
+  int lane = threadIdx.x & 63;
+  for (int64_t sb = sb0; sb < n_sblocks; sb += stride) {
+    const uint8_t* blk = src + sb * 210;
+    const uint8_t* ql = blk;
+    const uint8_t* qh = blk + 128;
+    const int8_t* scales = (const int8_t*)(blk + 192);
+    float d = f16_to_f32(blk + 208);
+    uint16_t* o = dst + sb * 256;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int j = lane * 4 + r * 64;   // element 0..255
+      int n = j >> 7;              // 128-half 0..1
+      int rr = j & 127;
+      int half = rr >> 5;          // 0..3
+      int l = rr & 31;
+      int q;
+      const uint8_t* qln = ql + n * 64;
+      const uint8_t* qhn = qh + n * 32;
+      if (half == 0)
+        q = (int)((qln[l] & 0xF) | (((qhn[l] >> 0) & 3) << 4)) - 32;
+      else if (half == 1)
+        q = (int)((qln[32 + l] & 0xF) | (((qhn[l] >> 2) & 3) << 4)) - 32;
+      else if (half == 2)
+        q = (int)((qln[l] >> 4) | (((qhn[l] >> 4) & 3) << 4)) - 32;
+      else
+        q = (int)((qln[32 + l] >> 4) | (((qhn[l] >> 6) & 3) << 4)) - 32;
+      int sc = scales[n * 8 + half * 2 + (l >> 4)];
+      o[j] = f32_to_bf16(d * (float)sc * (float)q);
+    }
+  }
+}
+
+}  // namespace
+
+// qtype ids follow GGML: 2=q4_0, 8=q8_0, 12=q4_K, 14=q6_K
+extern "C" void launch_gguf_dequant(int qtype, const void* src,
+                                    uint16_t* dst, int64_t n_blocks,
+                                    hipStream_t stream) {
+  if (n_blocks <= 0) return;
+  const uint8_t* s = (const uint8_t*)src;
+  if (qtype == 2 || qtype == 8) {
+    int64_t want = (n_blocks + 255) / 256;
+    int blocks = want > 8192 ? 8192 : (int)want;
+    if (qtype == 2)
+      hipLaunchKernelGGL(dequant_q4_0, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else
+      hipLaunchKernelGGL(dequant_q8_0, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    return;
+  }
+  if (qtype == 12 || qtype == 14) {
+    // 4 waves per 256-thread workgroup, one superblock per wave
+    int64_t want = (n_blocks + 3) / 4;
+    int blocks = want > 8192 ? 8192 : (int)want;
+    if (qtype == 12)
+      hipLaunchKernelGGL(dequant_q4_K, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else
+      hipLaunchKernelGGL(dequant_q6_K, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    return;
+  }
+  // unknown qtype: trap loudly rather than silently skip
+  abort();
+}

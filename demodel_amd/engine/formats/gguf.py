@@ -1,0 +1,340 @@
+"""GGUF container parsing, synthesis, and GPU dequantization.
+
+The Ollama model layer (mediaType application/vnd.ollama.image.model,
+reference CONTRIBUTING.md:141-146) is a GGUF file.  This module parses the
+header from a landed blob's head bytes, sizes each quantized tensor, and
+drives the CDNA4 dequant kernels (csrc/gguf_dequant.hip) straight from the
+blob's HBM region into fresh bf16 tensors — the blob never round-trips to
+host.
+
+Only the container layout + the quant types the kernels support are
+implemented; unknown quant types fail loudly.
+"""
+
+from __future__ import annotations
+
+import os
+import struct
+from dataclasses import dataclass, field
+
+MAGIC = 0x46554747  # 'GGUF' little-endian
+
+# ggml type id -> (name, block_elems, block_bytes)
+GGML_TYPES = {
+    0: ("f32", 1, 4),
+    1: ("f16", 1, 2),
+    2: ("q4_0", 32, 18),
+    8: ("q8_0", 32, 34),
+    12: ("q4_K", 256, 144),
+    14: ("q6_K", 256, 210),
+    30: ("bf16", 1, 2),
+}
+
+_VT_FMT = {0: "<B", 1: "<b", 2: "<H", 3: "<h", 4: "<I", 5: "<i",
+           6: "<f", 7: "<B", 10: "<Q", 11: "<q", 12: "<d"}
+
+
+@dataclass
+class GGUFTensor:
+    name: str
+    dims: tuple[int, ...]     # ggml order: dims[0] fastest-varying
+    type_id: int
+    offset: int               # into the data section
+
+    @property
+    def type_name(self) -> str:
+        return GGML_TYPES[self.type_id][0]
+
+    @property
+    def n_elems(self) -> int:
+        n = 1
+        for d in self.dims:
+            n *= d
+        return n
+
+    @property
+    def nbytes(self) -> int:
+        _, be, bb = GGML_TYPES[self.type_id]
+        assert self.n_elems % be == 0, (self.name, self.dims)
+        return self.n_elems // be * bb
+
+    @property
+    def n_blocks(self) -> int:
+        _, be, _ = GGML_TYPES[self.type_id]
+        return self.n_elems // be
+
+
+@dataclass
+class GGUFModel:
+    version: int
+    kv: dict
+    tensors: list[GGUFTensor]
+    data_offset: int          # absolute file offset of the data section
+    alignment: int
+    blob: object = None       # LandedBlob when parsed from one
+
+    def tensor(self, name: str) -> GGUFTensor:
+        for t in self.tensors:
+            if t.name == name:
+                return t
+        raise KeyError(name)
+
+
+class _Reader:
+    def __init__(self, data: bytes):
+        self.d = data
+        self.o = 0
+
+    def take(self, fmt: str):
+        v = struct.unpack_from(fmt, self.d, self.o)
+        self.o += struct.calcsize(fmt)
+        return v[0] if len(v) == 1 else v
+
+    def take_str(self) -> str:
+        n = self.take("<Q")
+        s = self.d[self.o:self.o + n].decode("utf-8", "replace")
+        self.o += n
+        return s
+
+    def take_value(self, vt: int):
+        if vt in _VT_FMT:
+            return self.take(_VT_FMT[vt])
+        if vt == 8:
+            return self.take_str()
+        if vt == 9:
+            et = self.take("<I")
+            n = self.take("<Q")
+            return [self.take_value(et) for _ in range(n)]
+        raise ValueError(f"unknown gguf value type {vt}")
+
+
+def parse_bytes(head: bytes) -> GGUFModel:
+    r = _Reader(head)
+    magic = r.take("<I")
+    if magic != MAGIC:
+        raise ValueError(f"not a GGUF file (magic {magic:#x})")
+    version = r.take("<I")
+    n_tensors = r.take("<Q")
+    n_kv = r.take("<Q")
+    kv = {}
+    for _ in range(n_kv):
+        k = r.take_str()
+        vt = r.take("<I")
+        kv[k] = r.take_value(vt)
+    tensors = []
+    for _ in range(n_tensors):
+        name = r.take_str()
+        nd = r.take("<I")
+        dims = tuple(r.take("<Q") for _ in range(nd))
+        tid = r.take("<I")
+        off = r.take("<Q")
+        if tid not in GGML_TYPES:
+            raise ValueError(
+                f"unsupported ggml tensor type {tid} for {name!r}")
+        tensors.append(GGUFTensor(name, dims, tid, off))
+    align = int(kv.get("general.alignment", 32))
+    data_offset = (r.o + align - 1) // align * align
+    return GGUFModel(version=version, kv=kv, tensors=tensors,
+                     data_offset=data_offset, alignment=align)
+
+
+def parse(blob) -> GGUFModel:
+    """Parse from a LandedBlob's head bytes."""
+    gg = parse_bytes(blob.head)
+    gg.blob = blob
+    return gg
+
+
+# ------------------------------------------------------------------ #
+# synthesis (tests + synthetic benchmarks)
+
+def build_file(path: str, tensors: list[tuple[str, tuple[int, ...], int]],
+               kv: dict | None = None, rng=None) -> GGUFModel:
+    """Write a synthetic GGUF file: (name, dims, type_id) with random
+    quant payloads."""
+    import numpy as np
+
+    rng = rng or np.random.default_rng(0)
+    kv = dict(kv or {})
+    kv.setdefault("general.architecture", "llama")
+    kv.setdefault("general.alignment", 32)
+
+    def enc_str(s: str) -> bytes:
+        b = s.encode()
+        return struct.pack("<Q", len(b)) + b
+
+    def enc_kv(k: str, v) -> bytes:
+        if isinstance(v, bool):
+            return enc_str(k) + struct.pack("<IB", 7, int(v))
+        if isinstance(v, int):
+            return enc_str(k) + struct.pack("<Iq", 11, v)
+        if isinstance(v, float):
+            return enc_str(k) + struct.pack("<If", 6, v)
+        if isinstance(v, str):
+            return enc_str(k) + struct.pack("<I", 8) + enc_str(v)
+        raise TypeError(type(v))
+
+    head = struct.pack("<IIQQ", MAGIC, 3, len(tensors), len(kv))
+    for k, v in kv.items():
+        head += enc_kv(k, v)
+    infos = []
+    off = 0
+    align = int(kv["general.alignment"])
+    for name, dims, tid in tensors:
+        t = GGUFTensor(name, dims, tid, off)
+        infos.append(t)
+        off += (t.nbytes + align - 1) // align * align
+    for t in infos:
+        head += enc_str(t.name)
+        head += struct.pack("<I", len(t.dims))
+        for d in t.dims:
+            head += struct.pack("<Q", d)
+        head += struct.pack("<IQ", t.type_id, t.offset)
+    data_offset = (len(head) + align - 1) // align * align
+    with open(path, "wb") as f:
+        f.write(head)
+        f.write(b"\0" * (data_offset - len(head)))
+        for t in infos:
+            payload = rng.integers(0, 256, size=t.nbytes,
+                                   dtype=np.uint8).tobytes()
+            # keep f16/f32 payloads finite so dequant comparisons are exact
+            if t.type_id in (0, 1, 30):
+                vals = rng.standard_normal(t.n_elems)
+                if t.type_id == 0:
+                    payload = vals.astype(np.float32).tobytes()
+                elif t.type_id == 1:
+                    payload = vals.astype(np.float16).tobytes()
+                else:
+                    payload = (vals.astype(np.float32).view(np.uint32)
+                               >> 16).astype(np.uint16).tobytes()
+            f.write(payload)
+            pad = (t.nbytes + align - 1) // align * align - t.nbytes
+            f.write(b"\0" * pad)
+    gg = parse_bytes(open(path, "rb").read(min(
+        os.path.getsize(path), 8 << 20)))
+    return gg
+
+
+# ------------------------------------------------------------------ #
+# CPU reference dequantization (tests; mirrors csrc/gguf_dequant.hip)
+
+def dequant_cpu(type_id: int, raw: bytes, n_elems: int):
+    """Reference dequant -> float32 numpy array (GGML layouts)."""
+    import numpy as np
+
+    if type_id == 0:
+        return np.frombuffer(raw, np.float32)[:n_elems].copy()
+    if type_id == 1:
+        return np.frombuffer(raw, np.float16)[:n_elems].astype(np.float32)
+    if type_id == 30:
+        u = np.frombuffer(raw, np.uint16)[:n_elems].astype(np.uint32) << 16
+        return u.view(np.float32)
+    b = np.frombuffer(raw, np.uint8)
+    if type_id == 2:  # q4_0
+        nb = n_elems // 32
+        blk = b[:nb * 18].reshape(nb, 18)
+        d = blk[:, 0:2].copy().view(np.float16).astype(np.float32)
+        qs = blk[:, 2:18]
+        lo = (qs & 0xF).astype(np.int32) - 8
+        hi = (qs >> 4).astype(np.int32) - 8
+        out = np.concatenate([lo, hi], axis=1).astype(np.float32)
+        return (out * d).reshape(-1)
+    if type_id == 8:  # q8_0
+        nb = n_elems // 32
+        blk = b[:nb * 34].reshape(nb, 34)
+        d = blk[:, 0:2].copy().view(np.float16).astype(np.float32)
+        q = blk[:, 2:34].copy().view(np.int8).astype(np.float32)
+        return (q * d).reshape(-1)
+    if type_id == 12:  # q4_K
+        nb = n_elems // 256
+        blk = b[:nb * 144].reshape(nb, 144)
+        d = blk[:, 0:2].copy().view(np.float16).astype(np.float32)[:, 0]
+        dmin = blk[:, 2:4].copy().view(np.float16).astype(np.float32)[:, 0]
+        scales = blk[:, 4:16]
+        qs = blk[:, 16:144]
+        out = np.empty((nb, 256), np.float32)
+        for j in range(8):  # 32-elem sub-blocks
+            if j < 4:
+                sc = (scales[:, j] & 63).astype(np.float32)
+                mn = (scales[:, j + 4] & 63).astype(np.float32)
+            else:
+                sc = ((scales[:, j + 4] & 0xF)
+                      | ((scales[:, j - 4] >> 6) << 4)).astype(np.float32)
+                mn = ((scales[:, j + 4] >> 4)
+                      | ((scales[:, j] >> 6) << 4)).astype(np.float32)
+            pair = j // 2            # 64-elem group
+            hi_nib = j % 2
+            q8 = qs[:, pair * 32:(pair + 1) * 32]
+            nib = (q8 >> 4) if hi_nib else (q8 & 0xF)
+            out[:, j * 32:(j + 1) * 32] = (
+                d[:, None] * sc[:, None] * nib.astype(np.float32)
+                - dmin[:, None] * mn[:, None])
+        return out.reshape(-1)
+    if type_id == 14:  # q6_K
+        nb = n_elems // 256
+        blk = b[:nb * 210].reshape(nb, 210)
+        ql = blk[:, 0:128]
+        qh = blk[:, 128:192]
+        sc = blk[:, 192:208].copy().view(np.int8)
+        d = blk[:, 208:210].copy().view(np.float16).astype(np.float32)[:, 0]
+        out = np.empty((nb, 256), np.float32)
+        for n in range(2):
+            qln = ql[:, n * 64:(n + 1) * 64]
+            qhn = qh[:, n * 32:(n + 1) * 32]
+            scn = sc[:, n * 8:(n + 1) * 8]
+            for half in range(4):
+                if half == 0:
+                    q = (qln[:, 0:32] & 0xF) | (((qhn >> 0) & 3) << 4)
+                elif half == 1:
+                    q = (qln[:, 32:64] & 0xF) | (((qhn >> 2) & 3) << 4)
+                elif half == 2:
+                    q = (qln[:, 0:32] >> 4) | (((qhn >> 4) & 3) << 4)
+                else:
+                    q = (qln[:, 32:64] >> 4) | (((qhn >> 6) & 3) << 4)
+                qv = q.astype(np.int32) - 32
+                scale_idx = half * 2  # + l//16
+                s0 = scn[:, scale_idx].astype(np.float32)
+                s1 = scn[:, scale_idx + 1].astype(np.float32)
+                seg = np.empty((nb, 32), np.float32)
+                seg[:, :16] = (qv[:, :16] * s0[:, None])
+                seg[:, 16:] = (qv[:, 16:] * s1[:, None])
+                out[:, n * 128 + half * 32: n * 128 + (half + 1) * 32] = \
+                    d[:, None] * seg
+        return out.reshape(-1)
+    raise ValueError(f"unsupported type {type_id}")
+
+
+# ------------------------------------------------------------------ #
+# GPU dequantization from a landed blob
+
+def dequant_tensor_gpu(gg: GGUFModel, t: GGUFTensor, stream=None):
+    """Dequantize one tensor from the landed blob -> torch bf16 tensor
+    (shape reversed from ggml dims: torch shape = dims[::-1])."""
+    import torch
+
+    from ...gpu import hip
+
+    h = hip()
+    blob = gg.blob
+    assert blob is not None and blob.device != "cpu", \
+        "GPU dequant needs a GPU-landed blob"
+    src = blob.buffer.ptr + gg.data_offset + t.offset
+    own_stream = stream is None
+    stream = stream or h.Stream(0)
+    if t.type_id in (1, 0, 30):
+        # f16/f32/bf16: view + cast via torch (no custom kernel needed)
+        u8 = blob.torch_u8()
+        raw = u8[gg.data_offset + t.offset:
+                 gg.data_offset + t.offset + t.nbytes]
+        if t.type_id == 30:
+            return raw.view(torch.bfloat16).view(t.dims[::-1])
+        if t.type_id == 1:
+            return raw.view(torch.float16).view(t.dims[::-1]).to(
+                torch.bfloat16)
+        return raw.view(torch.float32).view(t.dims[::-1]).to(torch.bfloat16)
+    out = h.DeviceBuffer(t.n_elems * 2)
+    h.gguf_dequant(t.type_id, src, out.ptr, t.n_blocks, stream.handle)
+    if own_stream:
+        stream.sync()
+    u8 = torch.from_dlpack(out.to_dlpack())
+    return u8.view(torch.bfloat16).view(t.dims[::-1])

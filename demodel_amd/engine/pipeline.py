@@ -1,0 +1,307 @@
+"""The HBM landing pipeline (SURVEY.md §2.3 K6 + K1).
+
+Replaces the reference proxy's hot byte loop (goproxy io.Copy,
+SURVEY.md §3.2) with the MI355X path:
+
+    socket/file ──drain (no GIL)──▶ pinned ring slab ──hipMemcpyAsync──▶
+      HBM blob buffer ──sha256_batch (GPU, chunk-parallel)──▶ verified
+
+* The ring has N slabs; slab reuse waits on that slab's H2D event, so
+  network receive of chunk k+N overlaps the copy of chunk k.
+* Verification is chunk-parallel on the GPU (64 KiB sub-chunks by
+  default) — digests land in the cache sidecar and are compared on
+  replay.  An exact whole-blob sha256 (upstream etag/digest identity) is
+  available as `host_chain=True` (hashlib pipelined on the fill thread) or
+  `gpu_chain=True` (on-device sequential chain kernel, honest-but-slow;
+  SURVEY.md §7 hard part (b)).
+* On CPU-only machines HostLander lands into page-cache memory with the
+  same interface, so the whole engine is testable without a GPU.
+"""
+
+from __future__ import annotations
+
+import hashlib
+import struct
+import time
+from dataclasses import dataclass, field
+
+from ..gpu import have_gpu, hip
+
+VERIFY_CHUNK = 64 << 10
+
+
+@dataclass
+class LandedBlob:
+    nbytes: int
+    device: str
+    buffer: object                    # _hip.DeviceBuffer | bytearray
+    chunk_digests: list[str] = field(default_factory=list)
+    verify_chunk: int = VERIFY_CHUNK
+    sha256: str | None = None         # exact whole-blob digest if computed
+    head: bytes = b""                 # first bytes (for header parsing)
+    timings: dict = field(default_factory=dict)
+
+    def torch_u8(self):
+        """The landed bytes as a zero-copy 1-D torch.uint8 tensor."""
+        import torch
+
+        if self.device == "cpu":
+            return torch.frombuffer(self.buffer, dtype=torch.uint8)
+        return torch.from_dlpack(self.buffer.to_dlpack())
+
+
+class Lander:
+    """GPU landing pipeline; one instance per device + stream pair."""
+
+    def __init__(self, device_index: int = 0, slab_bytes: int = 32 << 20,
+                 n_slabs: int = 4, verify_chunk: int = VERIFY_CHUNK,
+                 head_bytes: int = 8 << 20):
+        self._h = hip()
+        self._h.set_device(device_index)
+        self.device_index = device_index
+        self.slab_bytes = slab_bytes
+        self.verify_chunk = verify_chunk
+        self.head_bytes = head_bytes
+        self.pool = self._h.PinnedPool(slab_bytes, n_slabs)
+        self.copy_stream = self._h.Stream(0)
+        self.verify_stream = self._h.Stream(0)
+        self._slab_events = [self._h.Event() for _ in range(n_slabs)]
+        self._slab_busy = [False] * n_slabs
+
+    def land(self, fill, nbytes: int, verify: bool = True,
+             host_chain: bool = False, gpu_chain: bool = False,
+             keep_head: bool = True) -> LandedBlob:
+        """Land exactly `nbytes` from `fill` into a fresh HBM buffer.
+
+        fill(view: memoryview) -> int: write up to len(view) bytes into the
+        pinned slab view, return bytes written (0 = EOF/underrun -> error).
+        """
+        h = self._h
+        t0 = time.perf_counter()
+        buf = h.DeviceBuffer(max(nbytes, 1))
+        chain = hashlib.sha256() if host_chain else None
+        gpu_state = None
+        if gpu_chain:
+            gpu_state = h.DeviceBuffer(32)
+            h.sha256_chain_init(gpu_state.ptr, self.verify_stream.handle)
+        head = bytearray()
+        n_slabs = self.pool.n_slabs
+        off = 0
+        i = 0
+        fill_s = 0.0
+        while off < nbytes:
+            slab = i % n_slabs
+            if self._slab_busy[slab]:
+                self._slab_events[slab].sync()
+            want = min(self.slab_bytes, nbytes - off)
+            view = self.pool.slab_view(slab)[:want]
+            tf = time.perf_counter()
+            got = 0
+            while got < want:
+                n = fill(view[got:])
+                if n <= 0:
+                    raise IOError(
+                        f"blob underrun at {off + got}/{nbytes} bytes")
+                got += n
+            fill_s += time.perf_counter() - tf
+            if keep_head and len(head) < self.head_bytes:
+                take = min(want, self.head_bytes - len(head))
+                head += bytes(view[:take])
+            if chain is not None:
+                chain.update(view)
+            h.h2d_async(buf.ptr + off, self.pool.slab_ptr(slab), want,
+                        self.copy_stream.handle)
+            self._slab_events[slab].record(self.copy_stream.handle)
+            self._slab_busy[slab] = True
+            if gpu_state is not None:
+                # chain over the landed region (whole 64B blocks only; the
+                # ragged tail is folded in at finalize)
+                self._slab_events[slab].wait(self.verify_stream.handle)
+                h.sha256_chain_update(gpu_state.ptr, buf.ptr + off,
+                                      want // 64
+                                      if off + want < nbytes
+                                      else (nbytes - (nbytes % 64) - off)
+                                      // 64,
+                                      self.verify_stream.handle)
+            off += want
+            i += 1
+        blob = LandedBlob(nbytes=nbytes,
+                          device=f"cuda:{self.device_index}", buffer=buf,
+                          verify_chunk=self.verify_chunk, head=bytes(head))
+        t_fill_done = time.perf_counter()
+
+        if verify and nbytes > 0:
+            blob.chunk_digests = self._gpu_chunk_digests(buf, nbytes)
+        else:
+            self.copy_stream.sync()
+
+        if chain is not None:
+            blob.sha256 = chain.hexdigest()
+        elif gpu_state is not None:
+            blob.sha256 = self._finalize_gpu_chain(gpu_state, buf, nbytes)
+
+        t1 = time.perf_counter()
+        blob.timings = {
+            "total_s": t1 - t0,
+            "fill_s": fill_s,
+            "land_s": t_fill_done - t0,
+            "verify_s": t1 - t_fill_done,
+            "gbps": nbytes / max(t1 - t0, 1e-9) / 1e9,
+        }
+        return blob
+
+    def _gpu_chunk_digests(self, buf, nbytes: int) -> list[str]:
+        h = self._h
+        n_chunks = (nbytes + self.verify_chunk - 1) // self.verify_chunk
+        dig_dev = h.DeviceBuffer(n_chunks * 32)
+        # hash must see completed copies
+        done = h.Event()
+        done.record(self.copy_stream.handle)
+        done.wait(self.verify_stream.handle)
+        h.sha256_batch(buf.ptr, nbytes, self.verify_chunk, dig_dev.ptr,
+                       n_chunks, self.verify_stream.handle)
+        host = bytearray(n_chunks * 32)
+        import ctypes
+
+        addr = ctypes.addressof(
+            (ctypes.c_char * len(host)).from_buffer(host))
+        h.d2h_async(addr, dig_dev.ptr, n_chunks * 32,
+                    self.verify_stream.handle)
+        self.verify_stream.sync()
+        out = []
+        for c in range(n_chunks):
+            words = struct.unpack_from(">8I", host, c * 32)
+            out.append("".join(f"{w:08x}" for w in words))
+        return out
+
+    def _finalize_gpu_chain(self, state_buf, buf, nbytes: int) -> str:
+        """Fold the ragged tail + padding into the device chain state."""
+        h = self._h
+        import ctypes
+
+        raw = bytearray(32)
+        addr = ctypes.addressof((ctypes.c_char * 32).from_buffer(raw))
+        self.verify_stream.sync()
+        h.d2h_async(addr, state_buf.ptr, 32, self.verify_stream.handle)
+        tail_off = nbytes - (nbytes % 64)
+        tail = bytearray(nbytes % 64)
+        if tail:
+            taddr = ctypes.addressof(
+                (ctypes.c_char * len(tail)).from_buffer(tail))
+            h.d2h_async(taddr, buf.ptr + tail_off, len(tail),
+                        self.verify_stream.handle)
+        self.verify_stream.sync()
+        state = list(struct.unpack("<8I", raw))
+        return _host_sha256_finish(state, nbytes, bytes(tail))
+
+
+# --- pure-python SHA-256 tail finisher (for the GPU chain) ---------------
+
+_K = [
+    0x428a2f98, 0x71374491, 0xb5c0fbcf, 0xe9b5dba5, 0x3956c25b, 0x59f111f1,
+    0x923f82a4, 0xab1c5ed5, 0xd807aa98, 0x12835b01, 0x243185be, 0x550c7dc3,
+    0x72be5d74, 0x80deb1fe, 0x9bdc06a7, 0xc19bf174, 0xe49b69c1, 0xefbe4786,
+    0x0fc19dc6, 0x240ca1cc, 0x2de92c6f, 0x4a7484aa, 0x5cb0a9dc, 0x76f988da,
+    0x983e5152, 0xa831c66d, 0xb00327c8, 0xbf597fc7, 0xc6e00bf3, 0xd5a79147,
+    0x06ca6351, 0x14292967, 0x27b70a85, 0x2e1b2138, 0x4d2c6dfc, 0x53380d13,
+    0x650a7354, 0x766a0abb, 0x81c2c92e, 0x92722c85, 0xa2bfe8a1, 0xa81a664b,
+    0xc24b8b70, 0xc76c51a3, 0xd192e819, 0xd6990624, 0xf40e3585, 0x106aa070,
+    0x19a4c116, 0x1e376c08, 0x2748774c, 0x34b0bcb5, 0x391c0cb3, 0x4ed8aa4a,
+    0x5b9cca4f, 0x682e6ff3, 0x748f82ee, 0x78a5636f, 0x84c87814, 0x8cc70208,
+    0x90befffa, 0xa4506ceb, 0xbef9a3f7, 0xc67178f2,
+]
+
+
+def _compress(state: list[int], block: bytes) -> None:
+    w = list(struct.unpack(">16I", block))
+    for t in range(16, 64):
+        w15, w2 = w[t - 15], w[t - 2]
+        s0 = (_rotr(w15, 7) ^ _rotr(w15, 18) ^ (w15 >> 3))
+        s1 = (_rotr(w2, 17) ^ _rotr(w2, 19) ^ (w2 >> 10))
+        w.append((w[t - 16] + s0 + w[t - 7] + s1) & 0xffffffff)
+    a, b, c, d, e, f, g, hh = state
+    for t in range(64):
+        S1 = _rotr(e, 6) ^ _rotr(e, 11) ^ _rotr(e, 25)
+        ch = (e & f) ^ (~e & g) & 0xffffffff
+        t1 = (hh + S1 + ch + _K[t] + w[t]) & 0xffffffff
+        S0 = _rotr(a, 2) ^ _rotr(a, 13) ^ _rotr(a, 22)
+        maj = (a & b) ^ (a & c) ^ (b & c)
+        t2 = (S0 + maj) & 0xffffffff
+        hh, g, f, e = g, f, e, (d + t1) & 0xffffffff
+        d, c, b, a = c, b, a, (t1 + t2) & 0xffffffff
+    for i, v in enumerate((a, b, c, d, e, f, g, hh)):
+        state[i] = (state[i] + v) & 0xffffffff
+
+
+def _rotr(x: int, n: int) -> int:
+    return ((x >> n) | (x << (32 - n))) & 0xffffffff
+
+
+def _host_sha256_finish(state: list[int], total_len: int,
+                        tail: bytes) -> str:
+    pad = bytearray(tail)
+    pad.append(0x80)
+    while (len(pad) % 64) != 56:
+        pad.append(0)
+    pad += struct.pack(">Q", total_len * 8)
+    for i in range(0, len(pad), 64):
+        _compress(state, bytes(pad[i:i + 64]))
+    return "".join(f"{w:08x}" for w in state)
+
+
+# --- CPU fallback landing target (tests, no-GPU plumbing config) ---------
+
+class HostLander:
+    def __init__(self, slab_bytes: int = 32 << 20,
+                 verify_chunk: int = VERIFY_CHUNK, head_bytes: int = 8 << 20):
+        self.slab_bytes = slab_bytes
+        self.verify_chunk = verify_chunk
+        self.head_bytes = head_bytes
+
+    def land(self, fill, nbytes: int, verify: bool = True,
+             host_chain: bool = False, gpu_chain: bool = False,
+             keep_head: bool = True) -> LandedBlob:
+        t0 = time.perf_counter()
+        buf = bytearray(nbytes)
+        mv = memoryview(buf)
+        chain = hashlib.sha256() if (host_chain or gpu_chain) else None
+        off = 0
+        while off < nbytes:
+            want = min(self.slab_bytes, nbytes - off)
+            got = 0
+            while got < want:
+                n = fill(mv[off + got: off + want])
+                if n <= 0:
+                    raise IOError(
+                        f"blob underrun at {off + got}/{nbytes} bytes")
+                got += n
+            if chain is not None:
+                chain.update(mv[off:off + want])
+            off += want
+        blob = LandedBlob(nbytes=nbytes, device="cpu", buffer=buf,
+                          verify_chunk=self.verify_chunk,
+                          head=bytes(buf[:min(nbytes, self.head_bytes)]))
+        t_land = time.perf_counter()
+        if verify:
+            vc = self.verify_chunk
+            blob.chunk_digests = [
+                hashlib.sha256(mv[o:o + vc]).hexdigest()
+                for o in range(0, nbytes, vc)
+            ]
+        if chain is not None:
+            blob.sha256 = chain.hexdigest()
+        t1 = time.perf_counter()
+        blob.timings = {
+            "total_s": t1 - t0, "fill_s": t_land - t0,
+            "land_s": t_land - t0, "verify_s": t1 - t_land,
+            "gbps": nbytes / max(t1 - t0, 1e-9) / 1e9,
+        }
+        return blob
+
+
+def make_lander(device_index: int = 0, **kw):
+    if have_gpu():
+        return Lander(device_index=device_index, **kw)
+    return HostLander(**{k: v for k, v in kw.items()
+                         if k in ("slab_bytes", "verify_chunk",
+                                  "head_bytes")})

@@ -1,0 +1,50 @@
+"""GPU runtime loader.
+
+Loads the in-tree ``demodel_amd._hip`` extension (built by
+demodel_amd/build.py, .so travels with the repo snapshot).  On a machine
+WITH a GPU a missing/broken extension is a hard error — the engine must
+never fall back to a silent CPU path there.  On CPU-only machines
+``have_gpu()`` is False and callers use host landing targets.
+"""
+
+from __future__ import annotations
+
+import os
+
+_hip = None
+_import_error: Exception | None = None
+
+try:
+    from demodel_amd import _hip as _hip  # type: ignore
+except Exception as e:  # pragma: no cover - exercised only on broken builds
+    _import_error = e
+
+
+def _torch_has_gpu() -> bool:
+    try:
+        import torch
+
+        return torch.cuda.is_available()
+    except Exception:
+        return False
+
+
+def have_gpu() -> bool:
+    if _hip is None:
+        if _torch_has_gpu():
+            raise RuntimeError(
+                "demodel_amd._hip failed to import on a GPU machine: "
+                f"{_import_error!r}. Run python -m demodel_amd.build."
+            )
+        return False
+    return _hip.device_count() > 0
+
+
+def hip():
+    """The raw extension module; raises loudly if unavailable."""
+    if _hip is None:
+        raise RuntimeError(
+            f"demodel_amd._hip is not available: {_import_error!r}. "
+            "Build it with `python -m demodel_amd.build`."
+        )
+    return _hip

@@ -1,0 +1,120 @@
+"""Synthetic model blob generation for tests and benchmarks.
+
+There is no network in the build/bench environments, so BASELINE.json's
+configs run on synthetic blobs: random-init Llama-3-shaped safetensors
+shards (configs 2-3), GGUF quant files (config 4), and zstd parquet-like
+streams (config 5).  Files are written once and served by the in-process
+fake origin (testing/origin.py) over loopback via sendfile.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import struct
+
+import numpy as np
+
+from ..engine.formats import safetensors as st
+
+# Llama-3-8B geometry (public config): the bench's model shape.
+LLAMA3_8B = {
+    "hidden": 4096, "inter": 14336, "layers": 32,
+    "heads": 32, "kv_heads": 8, "vocab": 128256,
+}
+# Llama-3-70B geometry for the sharded 8-rank pull (config 3).
+LLAMA3_70B = {
+    "hidden": 8192, "inter": 28672, "layers": 80,
+    "heads": 64, "kv_heads": 8, "vocab": 128256,
+}
+
+
+def llama_tensor_table(geom: dict) -> list[tuple[str, tuple[int, ...]]]:
+    h, inter, L = geom["hidden"], geom["inter"], geom["layers"]
+    kvh = geom["kv_heads"] * (h // geom["heads"])
+    out = [("model.embed_tokens.weight", (geom["vocab"], h))]
+    for i in range(L):
+        p = f"model.layers.{i}."
+        out += [
+            (p + "self_attn.q_proj.weight", (h, h)),
+            (p + "self_attn.k_proj.weight", (kvh, h)),
+            (p + "self_attn.v_proj.weight", (kvh, h)),
+            (p + "self_attn.o_proj.weight", (h, h)),
+            (p + "mlp.gate_proj.weight", (inter, h)),
+            (p + "mlp.up_proj.weight", (inter, h)),
+            (p + "mlp.down_proj.weight", (h, inter)),
+            (p + "input_layernorm.weight", (h,)),
+            (p + "post_attention_layernorm.weight", (h,)),
+        ]
+    out += [("model.norm.weight", (h,)),
+            ("lm_head.weight", (geom["vocab"], h))]
+    return out
+
+
+def write_shards(out_dir: str, geom: dict, n_shards: int,
+                 seed: int = 0, reuse: bool = True) -> dict[str, str]:
+    """Write llama-shaped bf16 safetensors shards; returns
+    {rfilename: path}.  Payload is a repeated random block (content is
+    irrelevant to the pipeline; generation must not dominate setup)."""
+    os.makedirs(out_dir, exist_ok=True)
+    table = llama_tensor_table(geom)
+    # round-robin tensors into shards, balancing bytes
+    shard_specs: list[dict] = [dict() for _ in range(n_shards)]
+    shard_bytes = [0] * n_shards
+    for name, shape in table:
+        nbytes = int(np.prod(shape)) * 2
+        i = shard_bytes.index(min(shard_bytes))
+        shard_specs[i][name] = ("BF16", shape, nbytes)
+        shard_bytes[i] += nbytes
+
+    rng = np.random.default_rng(seed)
+    rand_block = rng.integers(0, 256, size=64 << 20, dtype=np.uint8)
+    rand_block = rand_block.tobytes()
+
+    files = {}
+    total = n_shards
+    for i, spec in enumerate(shard_specs):
+        fname = f"model-{i + 1:05d}-of-{total:05d}.safetensors"
+        path = os.path.join(out_dir, fname)
+        head, data_bytes = st.build_header(spec)
+        want_size = len(head) + data_bytes
+        if reuse and os.path.exists(path) and \
+                os.path.getsize(path) == want_size:
+            files[fname] = path
+            continue
+        with open(path, "wb") as f:
+            f.write(head)
+            left = data_bytes
+            while left > 0:
+                take = min(left, len(rand_block))
+                f.write(rand_block[:take])
+                left -= take
+        files[fname] = path
+    cfg_path = os.path.join(out_dir, "config.json")
+    with open(cfg_path, "w") as f:
+        json.dump({"architectures": ["LlamaForCausalLM"],
+                   "hidden_size": geom["hidden"],
+                   "num_hidden_layers": geom["layers"],
+                   "vocab_size": geom["vocab"],
+                   "torch_dtype": "bfloat16"}, f)
+    files["config.json"] = cfg_path
+    return files
+
+
+def write_gguf_model(path: str, geom: dict, qtype: int = 12,
+                     reuse: bool = True):
+    """Synthetic GGUF (default q4_K) with llama-shaped 2-D tensors."""
+    from ..engine.formats import gguf
+
+    if reuse and os.path.exists(path) and os.path.getsize(path) > 0:
+        return gguf.parse_bytes(open(path, "rb").read(8 << 20))
+    tensors = []
+    for name, shape in llama_tensor_table(geom):
+        if len(shape) == 1:
+            tensors.append((name, shape, 0))       # f32 norms
+        else:
+            # ggml dims are reversed (fastest first) and dim0 must divide
+            # the quant block size
+            d0, d1 = shape[1], shape[0]
+            tensors.append((name, (d0, d1), qtype))
+    return gguf.build_file(path, tensors)

@@ -1,0 +1,106 @@
+"""GPU landing-pipeline + pull e2e on a real MI355X: fake origin over
+loopback -> pinned ring -> HBM -> GPU chunk verify -> zero-copy views."""
+
+import hashlib
+import json
+import os
+
+import numpy as np
+import pytest
+
+from helpers import Stack
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    s = Stack(tmp_path)
+    yield s
+    s.close()
+
+
+def _require_gpu():
+    from demodel_amd.gpu import have_gpu
+
+    assert have_gpu()
+
+
+def test_gpu_land_and_verify(tmp_path):
+    _require_gpu()
+    from demodel_amd.engine.pipeline import Lander
+
+    data = os.urandom(7 << 20)
+    pos = [0]
+
+    def fill(view):
+        n = min(len(view), len(data) - pos[0])
+        view[:n] = data[pos[0]:pos[0] + n]
+        pos[0] += n
+        return n
+
+    lander = Lander(slab_bytes=1 << 20, n_slabs=4, verify_chunk=64 << 10)
+    blob = lander.land(fill, len(data), gpu_chain=True)
+    # exact whole-blob digest from the GPU chain kernel
+    assert blob.sha256 == hashlib.sha256(data).hexdigest()
+    # chunk digests from the batch kernel
+    vc = 64 << 10
+    assert blob.chunk_digests == [
+        hashlib.sha256(data[o:o + vc]).hexdigest()
+        for o in range(0, len(data), vc)]
+    # round-trip: the landed bytes really are in HBM
+    t = blob.torch_u8()
+    assert bytes(t.cpu().numpy().tobytes()) == data
+
+
+def test_gpu_pull_hf_views(stack, tmp_path):
+    _require_gpu()
+    import torch
+
+    from demodel_amd.engine import pull as pull_mod
+    from demodel_amd.engine.formats import safetensors as st
+
+    a = (np.random.rand(256, 128) * 100).astype(np.float32)
+    spec = {"weight": ("F32", a.shape, a.nbytes)}
+    head, _ = st.build_header(spec)
+    p = tmp_path / "m.safetensors"
+    p.write_bytes(head + a.tobytes())
+    stack.origin.add_hf_repo("org/g", {"m.safetensors": str(p)})
+
+    res = pull_mod.pull_hf("org/g", endpoint=stack.origin_base,
+                           verify="digest", workers=1)
+    assert res.files[0].digest_ok is True
+    assert res.device.startswith("cuda")
+    t = res.tensors()["weight"]
+    assert t.is_cuda and t.dtype == torch.float32
+    assert torch.equal(t.cpu(), torch.from_numpy(a))
+
+
+def test_gpu_pull_ollama_dequant(stack, tmp_path):
+    _require_gpu()
+    import torch
+
+    from demodel_amd.engine import pull as pull_mod
+    from demodel_amd.engine.formats import gguf
+
+    gg_path = tmp_path / "m.gguf"
+    gguf.build_file(str(gg_path), [
+        ("blk.0.w.weight", (256, 8), 12),     # q4_K
+        ("blk.0.n.weight", (64,), 0),         # f32
+    ])
+    stack.origin.add_ollama_model("library/g", "latest", [
+        ("application/vnd.ollama.image.model", str(gg_path)),
+    ])
+    res = pull_mod.pull_ollama("g", "latest", endpoint=stack.origin_base,
+                               verify="digest", workers=1)
+    assert all(f.digest_ok for f in res.files)
+    gg = res.meta["gguf_model"]
+    t = gg.tensor("blk.0.w.weight")
+    out = gguf.dequant_tensor_gpu(gg, t)
+    assert out.dtype == torch.bfloat16 and out.shape == (8, 256)
+    raw = gg_path.read_bytes()[gg.data_offset + t.offset:]
+    want = gguf.dequant_cpu(12, raw[:t.nbytes], t.n_elems)
+    wf = torch.from_numpy(want).to(torch.bfloat16).float().view(8, 256)
+    gf = out.float().cpu()
+    mask = torch.isfinite(wf)
+    assert torch.allclose(gf[mask], wf[mask], rtol=1 / 64, atol=1e-3)
