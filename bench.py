@@ -107,19 +107,27 @@ def main():
         if dist:
             dist.barrier()
 
-    def one_step():
+    digest_map: dict = {}
+
+    def one_step(record_digests=False):
         res = pull_mod.pull_hf(
             "bench/model", endpoint=endpoint, workers=args.workers,
-            verify=args.verify, landers=landers)
+            verify=args.verify, landers=landers,
+            digest_map=digest_map or None)
         assert res.total_bytes == total_bytes, res.total_bytes
+        if record_digests:
+            for f in res.files:
+                digest_map[f.name] = f.blob.digest_blob
         # model-ready: materialize the tensor views
-        n_t = len(res.tensors()) if args.model != "tiny" or True else 0
+        n_t = len(res.tensors())
         if have_gpu:
             torch.cuda.synchronize()
         return res, n_t
 
-    for i in range(args.warmup):
-        res, _ = one_step()
+    # warmup; the first pull records chunk digests so every TIMED pull is a
+    # fully verified re-pull (GPU sha256_batch compared against the record)
+    for i in range(max(args.warmup, 1 if args.verify != "off" else 0)):
+        res, _ = one_step(record_digests=(i == 0))
         log(f"warmup {i}: {res.seconds_to_ready:.2f}s "
             f"({res.gbps:.2f} GB/s)")
 

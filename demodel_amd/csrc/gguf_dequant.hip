@@ -107,7 +107,7 @@ __global__ void dequant_q4_K(const uint8_t* __restrict__ src,
     uint16_t* o = dst + sb * 256;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int j = lane * 4 + r * 64;   // element index 0..255
+      int j = r * 64 + lane;       // element index 0..255, lane-contiguous
       int sub = j >> 5;            // 32-elem sub-block 0..7
       uint8_t sc, mn;
       scale_min_k4(sub, scales, &sc, &mn);
@@ -137,7 +137,7 @@ __global__ void dequant_q6_K(const uint8_t* __restrict__ src,
     uint16_t* o = dst + sb * 256;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int j = lane * 4 + r * 64;   // element 0..255
+      int j = r * 64 + lane;       // element 0..255, lane-contiguous
       int n = j >> 7;              // 128-half 0..1
       int rr = j & 127;
       int half = rr >> 5;          // 0..3

@@ -257,6 +257,11 @@ PYBIND11_MODULE(_hip, m) {
     hipError_t e = hipGetDeviceCount(&n);
     return e == hipSuccess ? n : 0;
   });
+  m.def("device_probe", [] {
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    return py::make_tuple(n, std::string(hipGetErrorString(e)));
+  });
   m.def("set_device", [](int d) { HIP_CHECK(hipSetDevice(d)); });
   m.def("device_sync", [] { HIP_CHECK(hipDeviceSynchronize()); },
         py::call_guard<py::gil_scoped_release>());

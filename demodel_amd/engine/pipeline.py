@@ -35,11 +35,16 @@ class LandedBlob:
     nbytes: int
     device: str
     buffer: object                    # _hip.DeviceBuffer | bytearray
-    chunk_digests: list[str] = field(default_factory=list)
+    digest_blob: bytes = b""          # n_chunks x 32B raw sha256 digests
     verify_chunk: int = VERIFY_CHUNK
     sha256: str | None = None         # exact whole-blob digest if computed
     head: bytes = b""                 # first bytes (for header parsing)
     timings: dict = field(default_factory=dict)
+
+    @property
+    def chunk_digests(self) -> list[str]:
+        return [self.digest_blob[i:i + 32].hex()
+                for i in range(0, len(self.digest_blob), 32)]
 
     def torch_u8(self):
         """The landed bytes as a zero-copy 1-D torch.uint8 tensor."""
@@ -48,6 +53,26 @@ class LandedBlob:
         if self.device == "cpu":
             return torch.frombuffer(self.buffer, dtype=torch.uint8)
         return torch.from_dlpack(self.buffer.to_dlpack())
+
+
+class DigestMismatch(IOError):
+    def __init__(self, chunk_index: int, verify_chunk: int):
+        self.chunk_index = chunk_index
+        super().__init__(
+            f"chunk digest mismatch at chunk {chunk_index} "
+            f"(byte offset {chunk_index * verify_chunk})")
+
+
+def check_digests(got: bytes, expected: bytes, verify_chunk: int) -> None:
+    if got == expected:
+        return
+    if len(got) != len(expected):
+        raise IOError(
+            f"digest count mismatch: {len(got) // 32} vs "
+            f"{len(expected) // 32} chunks")
+    for i in range(0, len(got), 32):
+        if got[i:i + 32] != expected[i:i + 32]:
+            raise DigestMismatch(i // 32, verify_chunk)
 
 
 class Lander:
@@ -70,11 +95,14 @@ class Lander:
 
     def land(self, fill, nbytes: int, verify: bool = True,
              host_chain: bool = False, gpu_chain: bool = False,
-             keep_head: bool = True) -> LandedBlob:
+             keep_head: bool = True,
+             expected_digests: bytes | None = None) -> LandedBlob:
         """Land exactly `nbytes` from `fill` into a fresh HBM buffer.
 
         fill(view: memoryview) -> int: write up to len(view) bytes into the
         pinned slab view, return bytes written (0 = EOF/underrun -> error).
+        expected_digests: raw 32B-per-chunk sha256s to verify against
+        (raises DigestMismatch).
         """
         h = self._h
         t0 = time.perf_counter()
@@ -130,8 +158,11 @@ class Lander:
                           verify_chunk=self.verify_chunk, head=bytes(head))
         t_fill_done = time.perf_counter()
 
-        if verify and nbytes > 0:
-            blob.chunk_digests = self._gpu_chunk_digests(buf, nbytes)
+        if (verify or expected_digests is not None) and nbytes > 0:
+            blob.digest_blob = self._gpu_chunk_digests(buf, nbytes)
+            if expected_digests is not None:
+                check_digests(blob.digest_blob, expected_digests,
+                              self.verify_chunk)
         else:
             self.copy_stream.sync()
 
@@ -150,7 +181,7 @@ class Lander:
         }
         return blob
 
-    def _gpu_chunk_digests(self, buf, nbytes: int) -> list[str]:
+    def _gpu_chunk_digests(self, buf, nbytes: int) -> bytes:
         h = self._h
         n_chunks = (nbytes + self.verify_chunk - 1) // self.verify_chunk
         dig_dev = h.DeviceBuffer(n_chunks * 32)
@@ -168,11 +199,11 @@ class Lander:
         h.d2h_async(addr, dig_dev.ptr, n_chunks * 32,
                     self.verify_stream.handle)
         self.verify_stream.sync()
-        out = []
-        for c in range(n_chunks):
-            words = struct.unpack_from(">8I", host, c * 32)
-            out.append("".join(f"{w:08x}" for w in words))
-        return out
+        # kernel writes big-endian words as native u32 -> swap to the
+        # canonical sha256 byte order
+        import numpy as np
+
+        return np.frombuffer(bytes(host), dtype="<u4").byteswap().tobytes()
 
     def _finalize_gpu_chain(self, state_buf, buf, nbytes: int) -> str:
         """Fold the ragged tail + padding into the device chain state."""
@@ -260,7 +291,8 @@ class HostLander:
 
     def land(self, fill, nbytes: int, verify: bool = True,
              host_chain: bool = False, gpu_chain: bool = False,
-             keep_head: bool = True) -> LandedBlob:
+             keep_head: bool = True,
+             expected_digests: bytes | None = None) -> LandedBlob:
         t0 = time.perf_counter()
         buf = bytearray(nbytes)
         mv = memoryview(buf)
@@ -282,12 +314,13 @@ class HostLander:
                           verify_chunk=self.verify_chunk,
                           head=bytes(buf[:min(nbytes, self.head_bytes)]))
         t_land = time.perf_counter()
-        if verify:
+        if verify or expected_digests is not None:
             vc = self.verify_chunk
-            blob.chunk_digests = [
-                hashlib.sha256(mv[o:o + vc]).hexdigest()
-                for o in range(0, nbytes, vc)
-            ]
+            blob.digest_blob = b"".join(
+                hashlib.sha256(mv[o:o + vc]).digest()
+                for o in range(0, nbytes, vc))
+            if expected_digests is not None:
+                check_digests(blob.digest_blob, expected_digests, vc)
         if chain is not None:
             blob.sha256 = chain.hexdigest()
         t1 = time.perf_counter()

@@ -137,7 +137,8 @@ def _etag_sha256(etag: str | None) -> str | None:
 
 def _pull_blob(landers: LanderPool, name: str, url: str,
                expected_sha: str | None, verify: str,
-               cafile, insecure, headers=None) -> PulledFile:
+               cafile, insecure, headers=None,
+               expected_digests: bytes | None = None) -> PulledFile:
     t0 = time.perf_counter()
     src = fetch.http_get(url, cafile=cafile, insecure=insecure,
                          headers=headers)
@@ -154,6 +155,7 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
             verify=verify in ("chunked", "digest", "gpu-digest"),
             host_chain=(verify == "digest"),
             gpu_chain=(verify == "gpu-digest"),
+            expected_digests=expected_digests,
         )
     finally:
         src.close()
@@ -181,7 +183,8 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             cafile=None, insecure: bool = False,
             patterns: list[str] | None = None,
             landers: LanderPool | None = None,
-            slab_bytes: int = 32 << 20) -> PullResult:
+            slab_bytes: int = 32 << 20,
+            digest_map: dict[str, bytes] | None = None) -> PullResult:
     endpoint = (endpoint or os.environ.get("HF_ENDPOINT")
                 or HF_DEFAULT_ENDPOINT).rstrip("/")
     t0 = time.perf_counter()
@@ -198,7 +201,8 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             ex.submit(
                 _pull_blob, landers, n,
                 f"{endpoint}/{repo}/resolve/{rev}/{n}",
-                None, verify, cafile, insecure): n
+                None, verify, cafile, insecure, None,
+                (digest_map or {}).get(n)): n
             for n in names
         }
         for fut in cf.as_completed(futs):

@@ -37,7 +37,22 @@ def have_gpu() -> bool:
                 f"{_import_error!r}. Run python -m demodel_amd.build."
             )
         return False
-    return _hip.device_count() > 0
+    if _hip.device_count() > 0:
+        return True
+    # A GPU torch sees but our runtime doesn't is a broken install, not a
+    # CPU box — never fall back silently (the HIP runtime can report
+    # no-device transiently before first context init; poke torch first).
+    if _torch_has_gpu():
+        import torch
+
+        torch.cuda.init()
+        n, err = _hip.device_probe()
+        if n > 0:
+            return True
+        raise RuntimeError(
+            f"torch sees a GPU but demodel_amd._hip does not "
+            f"(hipGetDeviceCount: {err}); refusing CPU fallback")
+    return False
 
 
 def hip():

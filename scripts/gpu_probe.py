@@ -1,0 +1,108 @@
+#!/usr/bin/env python3
+"""Microbenchmarks of the landing-pipeline primitives on one MI355X:
+H2D pinned copy, sha256_batch throughput (by chunk size), sha256 chain
+rate, scatter bandwidth, dequant bandwidth.  Prints JSON lines."""
+
+import ctypes
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+
+from demodel_amd.gpu import hip  # noqa: E402
+
+
+def bench(fn, iters=5, warmup=2):
+    for _ in range(warmup):
+        fn()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    h = hip()
+    h.set_device(0)
+    s = h.Stream(0)
+    N = 1 << 30  # 1 GiB
+
+    pool = h.PinnedPool(N, 1)
+    buf = h.DeviceBuffer(N)
+
+    def h2d():
+        h.h2d_async(buf.ptr, pool.slab_ptr(0), N, s.handle)
+        s.sync()
+
+    t = bench(h2d)
+    print(json.dumps({"op": "h2d_pinned", "gib": 1, "s": round(t, 4),
+                      "GBps": round(N / t / 1e9, 2)}), flush=True)
+
+    for chunk in (16 << 10, 64 << 10, 256 << 10, 1 << 20):
+        n_chunks = N // chunk
+        dig = h.DeviceBuffer(n_chunks * 32)
+
+        def hash_():
+            h.sha256_batch(buf.ptr, N, chunk, dig.ptr, n_chunks, s.handle)
+            s.sync()
+
+        t = bench(hash_, iters=3)
+        print(json.dumps({"op": "sha256_batch", "chunk_kib": chunk >> 10,
+                          "s": round(t, 4),
+                          "GBps": round(N / t / 1e9, 2)}), flush=True)
+
+    # chain rate over 64 MiB
+    M = 64 << 20
+    state = h.DeviceBuffer(32)
+    h.sha256_chain_init(state.ptr, s.handle)
+
+    def chain():
+        h.sha256_chain_update(state.ptr, buf.ptr, M // 64, s.handle)
+        s.sync()
+
+    t = bench(chain, iters=2, warmup=1)
+    print(json.dumps({"op": "sha256_chain", "mib": 64, "s": round(t, 3),
+                      "MBps": round(M / t / 1e6, 1)}), flush=True)
+
+    # scatter: 1 GiB in 4 MiB ranges to a second buffer
+    dst = h.DeviceBuffer(N)
+    n_desc = N // (4 << 20)
+    desc = bytearray()
+    import struct
+
+    for i in range(n_desc):
+        desc += struct.pack("<4Q", i * (4 << 20), dst.ptr + i * (4 << 20),
+                            4 << 20, 0)
+    dbuf = h.DeviceBuffer(len(desc))
+    src_c = (ctypes.c_char * len(desc)).from_buffer(desc)
+    h.h2d_async(dbuf.ptr, ctypes.addressof(src_c), len(desc), s.handle)
+    s.sync()
+
+    def scat():
+        h.scatter_ranges(buf.ptr, dbuf.ptr, n_desc, s.handle)
+        s.sync()
+
+    t = bench(scat)
+    print(json.dumps({"op": "scatter_4mib_ranges", "s": round(t, 4),
+                      "GBps_moved": round(2 * N / t / 1e9, 2)}), flush=True)
+
+    # q4_K dequant bandwidth: 1 GiB of superblocks
+    n_sb = N // 144
+    out = h.DeviceBuffer(n_sb * 256 * 2)
+
+    def dq():
+        h.gguf_dequant(12, buf.ptr, out.ptr, n_sb, s.handle)
+        s.sync()
+
+    t = bench(dq)
+    print(json.dumps({"op": "dequant_q4K", "in_gib": 1, "s": round(t, 4),
+                      "GBps_in": round(N / t / 1e9, 2),
+                      "GBps_out": round(n_sb * 512 / t / 1e9, 2)}),
+          flush=True)
+
+
+if __name__ == "__main__":
+    main()
