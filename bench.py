@@ -34,6 +34,15 @@ def log(msg):
           flush=True)
 
 
+class _FanoutResult:
+    """Minimal result shim for fan-out steps (log fields only)."""
+
+    def __init__(self, nbytes):
+        self.total_bytes = nbytes
+        self.seconds_to_ready = 0.0
+        self.gbps = 0.0
+
+
 def make_model_files(model: str, data_dir: str):
     from demodel_amd.testing import synth
 
@@ -53,6 +62,12 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--model", default="llama3-8b",
                     choices=list(GEOMS))
+    ap.add_argument("--mode", default="dp",
+                    choices=["dp", "shard", "broadcast"],
+                    help="dp: independent per-rank pulls (weak scaling); "
+                         "shard: rank r pulls 1/N of the manifest, "
+                         "file-wise RCCL broadcast reassembly (config 3); "
+                         "broadcast: rank 0 pulls, RCCL fan-out (R1)")
     ap.add_argument("--workers", type=int, default=4)
     ap.add_argument("--verify", default="chunked",
                     choices=["chunked", "digest", "gpu-digest", "off"])
@@ -109,7 +124,7 @@ def main():
 
     digest_map: dict = {}
 
-    def one_step(record_digests=False):
+    def dp_step(record_digests=False):
         res = pull_mod.pull_hf(
             "bench/model", endpoint=endpoint, workers=args.workers,
             verify=args.verify, landers=landers,
@@ -123,6 +138,47 @@ def main():
         if have_gpu:
             torch.cuda.synchronize()
         return res, n_t
+
+    file_sizes = [(n, os.path.getsize(p)) for n, p in sorted(files.items())]
+
+    def fanout_step(record_digests=False):
+        """shard/broadcast modes: every rank ends with the full model."""
+        from demodel_amd.engine.pull import _pull_blob
+        from demodel_amd.parallel.fanout import (shard_assignment,
+                                                 sharded_pull_fanout)
+
+        if args.mode == "broadcast":
+            plan = shard_assignment(file_sizes, 1, 0)  # rank 0 owns all
+            plan.my_files = plan.my_files if rank == 0 else []
+        else:
+            plan = shard_assignment(file_sizes, world, rank)
+
+        def pull_one(name):
+            pf = _pull_blob(
+                landers, name,
+                f"{endpoint}/bench/model/resolve/main/{name}",
+                None, args.verify, None, False,
+                expected_digests=digest_map.get(name))
+            if record_digests:
+                digest_map[name] = pf.blob.digest_blob
+            return pf.blob.torch_u8()
+
+        def alloc(nb):
+            return torch.empty(nb, dtype=torch.uint8,
+                               device="cuda" if have_gpu else "cpu")
+
+        out = sharded_pull_fanout(plan, pull_one, alloc)
+        got = sum(t.numel() for t in out.values())
+        assert got == total_bytes, (got, total_bytes)
+        if have_gpu:
+            torch.cuda.synchronize()
+        return out
+
+    def one_step(record_digests=False):
+        if args.mode == "dp":
+            return dp_step(record_digests)
+        res = fanout_step(record_digests)
+        return _FanoutResult(total_bytes * world), len(res)
 
     # warmup; the first pull records chunk digests so every TIMED pull is a
     # fully verified re-pull (GPU sha256_batch compared against the record)
@@ -161,7 +217,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(ms_per_step, 1),
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": "weak" if args.mode == "dp" else "strong",
             "vs_baseline": None,
             "dtype": "bf16",
             "data": "synthetic",
@@ -172,7 +228,9 @@ def main():
                 "files": len(files),
                 "verify": args.verify,
                 "seconds_to_ready": round(ms_per_step / 1000.0, 3),
-                "parallelism": f"independent-pull dp{world}",
+                "parallelism": (
+                    f"independent-pull dp{world}" if args.mode == "dp"
+                    else f"{args.mode}-fanout rccl x{world}"),
                 "device": "cuda" if have_gpu else "cpu",
             },
         }

@@ -107,52 +107,14 @@ class Lander:
         h = self._h
         t0 = time.perf_counter()
         buf = h.DeviceBuffer(max(nbytes, 1))
-        chain = hashlib.sha256() if host_chain else None
         gpu_state = None
         if gpu_chain:
             gpu_state = h.DeviceBuffer(32)
             h.sha256_chain_init(gpu_state.ptr, self.verify_stream.handle)
-        head = bytearray()
-        n_slabs = self.pool.n_slabs
-        off = 0
-        i = 0
-        fill_s = 0.0
-        while off < nbytes:
-            slab = i % n_slabs
-            if self._slab_busy[slab]:
-                self._slab_events[slab].sync()
-            want = min(self.slab_bytes, nbytes - off)
-            view = self.pool.slab_view(slab)[:want]
-            tf = time.perf_counter()
-            got = 0
-            while got < want:
-                n = fill(view[got:])
-                if n <= 0:
-                    raise IOError(
-                        f"blob underrun at {off + got}/{nbytes} bytes")
-                got += n
-            fill_s += time.perf_counter() - tf
-            if keep_head and len(head) < self.head_bytes:
-                take = min(want, self.head_bytes - len(head))
-                head += bytes(view[:take])
-            if chain is not None:
-                chain.update(view)
-            h.h2d_async(buf.ptr + off, self.pool.slab_ptr(slab), want,
-                        self.copy_stream.handle)
-            self._slab_events[slab].record(self.copy_stream.handle)
-            self._slab_busy[slab] = True
-            if gpu_state is not None:
-                # chain over the landed region (whole 64B blocks only; the
-                # ragged tail is folded in at finalize)
-                self._slab_events[slab].wait(self.verify_stream.handle)
-                h.sha256_chain_update(gpu_state.ptr, buf.ptr + off,
-                                      want // 64
-                                      if off + want < nbytes
-                                      else (nbytes - (nbytes % 64) - off)
-                                      // 64,
-                                      self.verify_stream.handle)
-            off += want
-            i += 1
+        chain = hashlib.sha256() if host_chain else None
+        head, fill_s = self.land_into(
+            buf, 0, fill, nbytes, file_size=nbytes, chain=chain,
+            gpu_state=gpu_state, keep_head=keep_head)
         blob = LandedBlob(nbytes=nbytes,
                           device=f"cuda:{self.device_index}", buffer=buf,
                           verify_chunk=self.verify_chunk, head=bytes(head))
@@ -180,6 +142,65 @@ class Lander:
             "gbps": nbytes / max(t1 - t0, 1e-9) / 1e9,
         }
         return blob
+
+    def land_into(self, buf, base_off: int, fill, nbytes: int,
+                  file_size: int | None = None, chain=None, gpu_state=None,
+                  keep_head: bool = False) -> tuple[bytearray, float]:
+        """Land `nbytes` from `fill` into buf at byte offset `base_off`
+        through this lander's pinned ring.  Segment primitive: several
+        landers (one per worker thread) can fill disjoint ranges of one
+        DeviceBuffer concurrently — range-parallel pulls of a single blob.
+
+        Returns (head_bytes, fill_seconds)."""
+        h = self._h
+        head = bytearray()
+        n_slabs = self.pool.n_slabs
+        file_size = file_size if file_size is not None else base_off + nbytes
+        off = 0
+        i = 0
+        fill_s = 0.0
+        while off < nbytes:
+            slab = i % n_slabs
+            if self._slab_busy[slab]:
+                self._slab_events[slab].sync()
+            want = min(self.slab_bytes, nbytes - off)
+            view = self.pool.slab_view(slab)[:want]
+            tf = time.perf_counter()
+            got = 0
+            while got < want:
+                n = fill(view[got:])
+                if n <= 0:
+                    raise IOError(
+                        f"blob underrun at {base_off + off + got}"
+                        f"/{file_size} bytes")
+                got += n
+            fill_s += time.perf_counter() - tf
+            if keep_head and len(head) < self.head_bytes:
+                take = min(want, self.head_bytes - len(head))
+                head += bytes(view[:take])
+            if chain is not None:
+                chain.update(view)
+            h.h2d_async(buf.ptr + base_off + off, self.pool.slab_ptr(slab),
+                        want, self.copy_stream.handle)
+            self._slab_events[slab].record(self.copy_stream.handle)
+            self._slab_busy[slab] = True
+            if gpu_state is not None:
+                # chain over the landed region (whole 64B blocks only; the
+                # ragged tail is folded in at finalize)
+                self._slab_events[slab].wait(self.verify_stream.handle)
+                abs_off = base_off + off
+                nblk = (want if abs_off + want < file_size
+                        else file_size - (file_size % 64) - abs_off) // 64
+                h.sha256_chain_update(gpu_state.ptr, buf.ptr + abs_off,
+                                      max(nblk, 0),
+                                      self.verify_stream.handle)
+            off += want
+            i += 1
+        return head, fill_s
+
+    def sync(self) -> None:
+        self.copy_stream.sync()
+        self.verify_stream.sync()
 
     def _gpu_chunk_digests(self, buf, nbytes: int) -> bytes:
         h = self._h

@@ -135,28 +135,56 @@ def _etag_sha256(etag: str | None) -> str | None:
     return None
 
 
+SEGMENT_MIN = 512 << 20     # segment blobs bigger than this
+MAX_SEGMENTS = 8
+
+
 def _pull_blob(landers: LanderPool, name: str, url: str,
                expected_sha: str | None, verify: str,
                cafile, insecure, headers=None,
-               expected_digests: bytes | None = None) -> PulledFile:
+               expected_digests: bytes | None = None,
+               seg_executor: cf.ThreadPoolExecutor | None = None
+               ) -> PulledFile:
     t0 = time.perf_counter()
+    do_verify = verify in ("chunked", "digest", "gpu-digest")
+    want_segments = (seg_executor is not None
+                     and verify in ("chunked", "off"))
+    req_headers = dict(headers or {})
+    if want_segments:
+        # probe range support by asking for the first segment outright; a
+        # 200 means no range support and we just stream the whole body
+        req_headers["Range"] = f"bytes=0-{SEGMENT_MIN - 1}"
     src = fetch.http_get(url, cafile=cafile, insecure=insecure,
-                         headers=headers)
+                         headers=req_headers)
     try:
-        if src.status != 200:
-            raise fetch.FetchError(
-                f"GET {url} -> {src.status} {src.resp.reason}")
-        nbytes = src.length
-        if nbytes < 0:
-            raise fetch.FetchError(f"no content-length for blob {url}")
-        lander = landers.get()
-        blob = lander.land(
-            src.fill, nbytes,
-            verify=verify in ("chunked", "digest", "gpu-digest"),
-            host_chain=(verify == "digest"),
-            gpu_chain=(verify == "gpu-digest"),
-            expected_digests=expected_digests,
-        )
+        if want_segments and src.status == 206:
+            cr = src.resp.get("content-range", "")
+            total = int(cr.rsplit("/", 1)[1])
+            if total > SEGMENT_MIN:
+                blob = _pull_segmented(landers, url, total, src,
+                                       expected_digests, cafile, insecure,
+                                       headers, seg_executor)
+            else:
+                blob = landers.get().land(
+                    src.fill, min(total, SEGMENT_MIN),
+                    verify=do_verify,
+                    expected_digests=expected_digests)
+        else:
+            if src.status != 200:
+                raise fetch.FetchError(
+                    f"GET {url} -> {src.status} {src.resp.reason}")
+            nbytes = src.length
+            if nbytes < 0:
+                raise fetch.FetchError(f"no content-length for blob {url}")
+            lander = landers.get()
+            blob = lander.land(
+                src.fill, nbytes,
+                verify=do_verify,
+                host_chain=(verify == "digest"),
+                gpu_chain=(verify == "gpu-digest"),
+                expected_digests=expected_digests,
+            )
+        nbytes = blob.nbytes
     finally:
         src.close()
     etag = src.resp.get("x-linked-etag") or src.resp.get("etag")
@@ -175,6 +203,63 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
              name, nbytes, pf.seconds, nbytes / max(pf.seconds, 1e-9) / 1e9,
              blob.device)
     return pf
+
+
+def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
+                    expected_digests, cafile, insecure, headers,
+                    seg_executor) -> "object":
+    """Range-parallel landing of one blob: segment 0 comes from the
+    already-open 206 stream, the rest are parallel range GETs, all landing
+    into disjoint ranges of one HBM buffer through per-thread pinned
+    rings.  Chunk verification runs once over the whole buffer."""
+    from .pipeline import LandedBlob
+
+    lander0 = landers.get()
+    h = lander0._h
+    buf = h.DeviceBuffer(total)
+    n_segs = min(MAX_SEGMENTS, (total + SEGMENT_MIN - 1) // SEGMENT_MIN)
+    bounds = [total * i // n_segs for i in range(n_segs + 1)]
+    # segment 0 must cover exactly what the open stream is serving
+    bounds[1] = max(bounds[1], min(SEGMENT_MIN, total))
+
+    def land_range(i):
+        lo, hi = bounds[i], bounds[i + 1]
+        if hi <= lo:
+            return 0.0
+        rng = {"Range": f"bytes={lo}-{hi - 1}"}
+        if headers:
+            rng.update(headers)
+        s = fetch.http_get(url, cafile=cafile, insecure=insecure,
+                           headers=rng)
+        try:
+            if s.status != 206:
+                raise fetch.FetchError(
+                    f"range GET {url} [{lo},{hi}) -> {s.status}")
+            lander = landers.get()
+            lander.land_into(buf, lo, s.fill, hi - lo, file_size=total)
+            lander.sync()
+        finally:
+            s.close()
+        return 0.0
+
+    futs = [seg_executor.submit(land_range, i) for i in range(1, n_segs)]
+    # this thread lands segment 0 from the open stream (it was asked for
+    # [0, SEGMENT_MIN) which equals bounds[1] when total >= SEGMENT_MIN)
+    head, _ = lander0.land_into(buf, 0, src0.fill, bounds[1],
+                                file_size=total, keep_head=True)
+    lander0.sync()
+    for f in futs:
+        f.result()
+    blob = LandedBlob(nbytes=total, device=f"cuda:{lander0.device_index}",
+                      buffer=buf, verify_chunk=lander0.verify_chunk,
+                      head=bytes(head))
+    blob.digest_blob = lander0._gpu_chunk_digests(buf, total)
+    if expected_digests is not None:
+        from .pipeline import check_digests
+
+        check_digests(blob.digest_blob, expected_digests,
+                      lander0.verify_chunk)
+    return blob
 
 
 def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
@@ -196,17 +281,23 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
                  if any(fnmatch.fnmatch(n, p) for p in patterns)]
     landers = landers or LanderPool(device_index, slab_bytes=slab_bytes)
     result = PullResult(spec=f"hf://{repo}@{rev}")
+    from ..gpu import have_gpu
+
+    seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
+              if have_gpu() else None)
     with cf.ThreadPoolExecutor(max_workers=workers) as ex:
         futs = {
             ex.submit(
                 _pull_blob, landers, n,
                 f"{endpoint}/{repo}/resolve/{rev}/{n}",
                 None, verify, cafile, insecure, None,
-                (digest_map or {}).get(n)): n
+                (digest_map or {}).get(n), seg_ex): n
             for n in names
         }
         for fut in cf.as_completed(futs):
             result.files.append(fut.result())
+    if seg_ex:
+        seg_ex.shutdown()
     result.files.sort(key=lambda f: f.name)
     result.total_bytes = sum(f.nbytes for f in result.files)
     result.seconds_to_ready = time.perf_counter() - t0
@@ -239,6 +330,10 @@ def pull_ollama(name: str, tag: str = "latest",
         layers.append(manifest["config"])
     landers = landers or LanderPool(device_index)
     result = PullResult(spec=f"ollama://{name}:{tag}")
+    from ..gpu import have_gpu
+
+    seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
+              if have_gpu() and verify in ("chunked", "off") else None)
     with cf.ThreadPoolExecutor(max_workers=workers) as ex:
         futs = {}
         for layer in layers:
@@ -247,11 +342,14 @@ def pull_ollama(name: str, tag: str = "latest",
             expected = digest.split(":", 1)[1] \
                 if digest.startswith("sha256:") else None
             futs[ex.submit(_pull_blob, landers, digest, url, expected,
-                           verify, cafile, insecure)] = layer
+                           verify, cafile, insecure, None, None,
+                           seg_ex)] = layer
         for fut in cf.as_completed(futs):
             pf = fut.result()
             pf.name = futs[fut].get("mediaType", pf.name)
             result.files.append(pf)
+    if seg_ex:
+        seg_ex.shutdown()
     result.total_bytes = sum(f.nbytes for f in result.files)
     result.seconds_to_ready = time.perf_counter() - t0
     result.device = result.files[0].blob.device if result.files else "cpu"

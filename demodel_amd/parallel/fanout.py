@@ -1,0 +1,135 @@
+"""RCCL-over-xGMI fan-out for pulled models.
+
+The reference's "syncing, distributing" story is aspirational
+(README.md:6-8); here it is concrete: one process per GPU
+(torch.distributed, backend "nccl" == RCCL on ROCm), and a pulled model
+fans out across the node over xGMI.
+
+Design for the topology: MI355X xGMI is point-to-point (7 links/GPU), so
+per-link bandwidth bounds ring collectives.  We therefore:
+
+* bucket broadcasts at BUCKET_BYTES so several transfers pipeline instead
+  of one serialized monster collective;
+* reassemble sharded pulls at *file* granularity — rank r owns ~1/N of
+  the manifest bytes, every file is broadcast from its owner in a fixed
+  manifest order, enqueued asynchronously as soon as the owner's landing
+  completes, so downloads overlap the xGMI traffic of already-landed
+  files (SURVEY.md §7 hard part (d)).
+
+Everything here works identically under gloo on CPU (world_size>1 tests
+run without GPUs).
+"""
+
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+
+from ..utils.log import get_logger
+
+log = get_logger("fanout")
+
+BUCKET_BYTES = 256 << 20
+
+
+def init_distributed(backend: str | None = None):
+    """Idempotent process-group init from torchrun env; returns
+    (rank, world_size)."""
+    import torch
+    import torch.distributed as dist
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world <= 1:
+        return 0, 1
+    if not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        dist.init_process_group(backend=backend)
+    return dist.get_rank(), dist.get_world_size()
+
+
+def broadcast_blob(tensor_u8, src: int, bucket_bytes: int = BUCKET_BYTES,
+                   async_op: bool = False, group=None):
+    """Bucketed broadcast of a 1-D u8 tensor (a landed blob).  Returns a
+    list of work handles when async_op, else completes synchronously."""
+    import torch.distributed as dist
+
+    works = []
+    n = tensor_u8.numel()
+    for off in range(0, n, bucket_bytes):
+        part = tensor_u8.narrow(0, off, min(bucket_bytes, n - off))
+        works.append(dist.broadcast(part, src, group=group, async_op=True))
+    if async_op:
+        return works
+    for w in works:
+        w.wait()
+    return []
+
+
+@dataclass
+class ShardPlan:
+    # file name -> (owner_rank, nbytes); iteration order = manifest order
+    owners: dict
+    my_files: list
+
+    def total_bytes(self) -> int:
+        return sum(nb for _, nb in self.owners.values())
+
+
+def shard_assignment(files: list[tuple[str, int]], world: int,
+                     rank: int) -> ShardPlan:
+    """Greedy byte-balanced assignment of manifest files to ranks.
+
+    files: [(name, nbytes)] in manifest order.
+    """
+    loads = [0] * world
+    owners = {}
+    for name, nbytes in sorted(files, key=lambda f: -f[1]):
+        r = loads.index(min(loads))
+        owners[name] = (r, nbytes)
+        loads[r] += nbytes
+    ordered = {name: owners[name] for name, _ in files}
+    my = [name for name, (r, _) in ordered.items() if r == rank]
+    return ShardPlan(owners=ordered, my_files=my)
+
+
+def sharded_pull_fanout(plan: ShardPlan, pull_one, alloc_u8,
+                        bucket_bytes: int = BUCKET_BYTES, group=None):
+    """Each rank pulls its owned files (pull_one(name) -> 1-D u8 tensor of
+    the landed blob), then every file is broadcast from its owner in
+    manifest order; download of later files overlaps the broadcast of
+    earlier ones.
+
+    pull_one: called only for files this rank owns; must return the
+        landed tensor (blocking).
+    alloc_u8(nbytes): allocate a receive tensor on this rank's device.
+
+    Returns {name: tensor_u8} with EVERY manifest file resident locally.
+    """
+    import concurrent.futures as cf
+
+    import torch.distributed as dist
+
+    rank = dist.get_rank(group)
+
+    # kick off this rank's downloads in manifest order on worker threads
+    ex = cf.ThreadPoolExecutor(max_workers=4)
+    pulls = {}
+    for name in plan.my_files:
+        pulls[name] = ex.submit(pull_one, name)
+
+    out = {}
+    works = []
+    for name, (owner, nbytes) in plan.owners.items():
+        if rank == owner:
+            tensor = pulls[name].result()  # wait for my landing
+            assert tensor.numel() == nbytes, (name, tensor.numel(), nbytes)
+        else:
+            tensor = alloc_u8(nbytes)
+        out[name] = tensor
+        works += broadcast_blob(tensor, owner, bucket_bytes,
+                                async_op=True, group=group)
+    for w in works:
+        w.wait()
+    ex.shutdown()
+    return out

@@ -1,0 +1,92 @@
+"""RCCL/gloo fan-out tests — multi-process (world_size=2, gloo backend)
+on CPU, per SURVEY.md §4's distributed-test plan."""
+
+import hashlib
+import os
+import socket
+
+import pytest
+import torch
+
+from demodel_amd.parallel.fanout import shard_assignment
+
+
+def test_shard_assignment_balanced():
+    files = [(f"f{i}", (i + 1) * 1000) for i in range(10)]
+    plans = [shard_assignment(files, 4, r) for r in range(4)]
+    # every file owned exactly once, order preserved
+    assert list(plans[0].owners) == [f"f{i}" for i in range(10)]
+    loads = [0] * 4
+    for name, (owner, nb) in plans[0].owners.items():
+        loads[owner] += nb
+    assert max(loads) - min(loads) <= max(nb for _, nb in files)
+    # my_files consistent with owners
+    for r, p in enumerate(plans):
+        assert all(p.owners[n][0] == r for n in p.my_files)
+        assert p.owners == plans[0].owners
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _worker(rank, world, port, tmpdir):
+    import torch.distributed as dist
+
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world)
+    try:
+        from demodel_amd.parallel.fanout import (broadcast_blob,
+                                                 shard_assignment,
+                                                 sharded_pull_fanout)
+
+        # --- bucketed broadcast ---
+        n = 3 << 20  # 3 MiB, several 1 MiB buckets
+        if rank == 0:
+            t = torch.arange(n, dtype=torch.int64) % 251
+            t = t.to(torch.uint8)
+        else:
+            t = torch.zeros(n, dtype=torch.uint8)
+        broadcast_blob(t, src=0, bucket_bytes=1 << 20)
+        want = (torch.arange(n, dtype=torch.int64) % 251).to(torch.uint8)
+        assert torch.equal(t, want), "broadcast content mismatch"
+
+        # --- sharded pull + fan-out ---
+        def content(name, nbytes):
+            seed = int.from_bytes(
+                hashlib.sha256(name.encode()).digest()[:4], "little")
+            g = torch.Generator().manual_seed(seed)
+            return torch.randint(0, 256, (nbytes,), generator=g,
+                                 dtype=torch.uint8)
+
+        files = [("a.bin", 100_000), ("b.bin", 400_000),
+                 ("c.bin", 50_000), ("d.bin", 250_000)]
+        plan = shard_assignment(files, world, rank)
+
+        def pull_one(name):
+            nb = dict(files)[name]
+            return content(name, nb)
+
+        out = sharded_pull_fanout(
+            plan, pull_one, lambda nb: torch.zeros(nb, dtype=torch.uint8),
+            bucket_bytes=128 << 10)
+        assert set(out) == {n for n, _ in files}
+        for name, nb in files:
+            assert torch.equal(out[name], content(name, nb)), name
+        # write a success marker per rank
+        with open(os.path.join(tmpdir, f"ok{rank}"), "w") as f:
+            f.write("ok")
+    finally:
+        dist.destroy_process_group()
+
+
+def test_sharded_fanout_gloo_world2(tmp_path):
+    port = _free_port()
+    torch.multiprocessing.spawn(
+        _worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+    assert (tmp_path / "ok0").exists() and (tmp_path / "ok1").exists()
