@@ -218,9 +218,14 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     h = lander0._h
     buf = h.DeviceBuffer(total)
     n_segs = min(MAX_SEGMENTS, (total + SEGMENT_MIN - 1) // SEGMENT_MIN)
-    bounds = [total * i // n_segs for i in range(n_segs + 1)]
-    # segment 0 must cover exactly what the open stream is serving
-    bounds[1] = max(bounds[1], min(SEGMENT_MIN, total))
+    # segment 0 covers exactly what the already-open 206 stream serves
+    # ([0, SEGMENT_MIN)); the rest of the file splits evenly
+    seg0 = min(SEGMENT_MIN, total)
+    rest = total - seg0
+    bounds = [0, seg0] + [
+        seg0 + rest * i // (n_segs - 1) for i in range(1, n_segs)
+    ] if n_segs > 1 else [0, total]
+    assert bounds[-1] == total, bounds
 
     def land_range(i):
         lo, hi = bounds[i], bounds[i + 1]

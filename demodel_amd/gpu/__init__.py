@@ -49,9 +49,13 @@ def have_gpu() -> bool:
         n, err = _hip.device_probe()
         if n > 0:
             return True
+        env = {k: v for k, v in os.environ.items()
+               if "VISIBLE" in k or k.startswith(("HSA", "ROCR", "HIP",
+                                                  "GPU"))}
         raise RuntimeError(
             f"torch sees a GPU but demodel_amd._hip does not "
-            f"(hipGetDeviceCount: {err}); refusing CPU fallback")
+            f"(hipGetDeviceCount: {err}); refusing CPU fallback. "
+            f"env={env} kfd={os.path.exists('/dev/kfd')}")
     return False
 
 
