@@ -451,12 +451,14 @@ class ProxyServer:
         out.replace("X-Demodel-Cache", "HIT")
         writer.write(http1.serialize_response(out))
         if req.method != "HEAD":
-            loop = asyncio.get_running_loop()
             await writer.drain()
             with hit.open_body() as f:
-                # zero-copy page-cache -> socket where possible
+                # zero-copy page-cache -> socket on a worker thread
+                # (parallel cache hits don't serialize on the event loop)
+                from ..utils.netio import sendfile_threaded
+
                 try:
-                    await loop.sendfile(writer.transport, f, fallback=True)
+                    await sendfile_threaded(writer, f, 0, hit.body_size)
                 except (NotImplementedError, RuntimeError, OSError):
                     f.seek(0)
                     while True:

@@ -253,6 +253,11 @@ class FakeOrigin:
         self._etag_cache[key] = h.hexdigest()
         return self._etag_cache[key]
 
+    async def _sendfile_threaded(self, writer, f, start: int, length: int):
+        from ..utils.netio import sendfile_threaded
+
+        await sendfile_threaded(writer, f, start, length)
+
     async def _error(self, writer, status: int):
         body = json.dumps({"error": status}).encode()
         head = ResponseHead("HTTP/1.1", status, "Error",
@@ -304,8 +309,11 @@ class FakeOrigin:
             f.seek(start)
             if self.tls_ctx is None:
                 try:
-                    await loop.sendfile(writer.transport, f, offset=start,
-                                        count=length, fallback=True)
+                    # blocking sendfile on a worker thread: the asyncio
+                    # loop.sendfile path serializes every stream through
+                    # the event loop (~3x slower at 4+ parallel blobs —
+                    # scripts/net_probe.py)
+                    await self._sendfile_threaded(writer, f, start, length)
                     return
                 except (NotImplementedError, OSError):
                     f.seek(start)

@@ -1,0 +1,66 @@
+"""Async-friendly high-throughput socket IO helpers.
+
+sendfile_threaded: a blocking os.sendfile loop on a worker thread, used
+for blob bodies by both the proxy's cache-hit serving and the test/bench
+origin.  asyncio's loop.sendfile serializes all streams through the event
+loop thread and tops out ~3x lower at 4+ parallel blob streams
+(scripts/net_probe.py); a thread per in-flight body keeps every byte in
+kernel space and scales with cores.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import concurrent.futures as cf
+import os
+import select as sel
+
+_POOL: cf.ThreadPoolExecutor | None = None
+
+
+def _pool() -> cf.ThreadPoolExecutor:
+    global _POOL
+    if _POOL is None:
+        _POOL = cf.ThreadPoolExecutor(max_workers=32,
+                                      thread_name_prefix="sendfile")
+    return _POOL
+
+
+async def sendfile_threaded(writer: asyncio.StreamWriter, f,
+                            start: int, length: int) -> None:
+    """Send [start, start+length) of file f on writer's socket.
+
+    Raises NotImplementedError when the transport has no plain socket
+    (TLS) — caller falls back to chunked writes.
+    """
+    sock = writer.transport.get_extra_info("socket")
+    tls = writer.transport.get_extra_info("sslcontext")
+    if sock is None or tls is not None:
+        raise NotImplementedError
+    await writer.drain()
+    writer.transport.pause_reading()
+    fd = sock.fileno()
+    infd = f.fileno()
+
+    def run():
+        off = start
+        remaining = length
+        while remaining > 0:
+            try:
+                sent = os.sendfile(fd, infd, off, remaining)
+            except BlockingIOError:
+                sel.select([], [fd], [], 10)
+                continue
+            if sent == 0:
+                raise ConnectionResetError("peer went away")
+            off += sent
+            remaining -= sent
+
+    loop = asyncio.get_running_loop()
+    try:
+        await loop.run_in_executor(_pool(), run)
+    finally:
+        try:
+            writer.transport.resume_reading()
+        except Exception:
+            pass

@@ -6,6 +6,18 @@ import pytest
 # make the in-tree package importable regardless of cwd
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+# Initialize HIP device enumeration BEFORE any test module imports: some
+# import during collection otherwise breaks hipGetDeviceCount in this
+# process (observed on MI355X boxes: full-suite collection -> "no
+# ROCm-capable device"; single-file runs fine).  hipGetDeviceCount caches
+# its result at first call, so probing here pins the good state.
+try:
+    from demodel_amd import _hip as _early_hip
+
+    _EARLY_PROBE = _early_hip.device_probe()
+except Exception as _e:  # extension not built yet — build fixture handles it
+    _EARLY_PROBE = (0, repr(_e))
+
 
 def pytest_configure(config):
     config.addinivalue_line(
