@@ -14,6 +14,17 @@ import os
 _hip = None
 _import_error: Exception | None = None
 
+# torch MUST load before our extension: torch-rocm wheels bundle their own
+# libamdhip64, and whichever HIP runtime initializes second in a process
+# sees no devices (observed on MI355X: _hip-first broke torch.cuda,
+# torch-first broke _hip when a different copy got mapped).  Importing
+# torch first means the dynamic loader resolves our DT_NEEDED libamdhip64
+# soname to torch's already-loaded copy — one runtime, shared state.
+try:
+    import torch  # noqa: F401
+except Exception:
+    pass
+
 try:
     from demodel_amd import _hip as _hip  # type: ignore
 except Exception as e:  # pragma: no cover - exercised only on broken builds

@@ -6,17 +6,14 @@ import pytest
 # make the in-tree package importable regardless of cwd
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-# Initialize HIP device enumeration BEFORE any test module imports: some
-# import during collection otherwise breaks hipGetDeviceCount in this
-# process (observed on MI355X boxes: full-suite collection -> "no
-# ROCm-capable device"; single-file runs fine).  hipGetDeviceCount caches
-# its result at first call, so probing here pins the good state.
+# Load torch + our HIP extension in the right order BEFORE any test module
+# imports: torch-rocm bundles its own libamdhip64 and whichever HIP runtime
+# initializes second in a process sees no devices (demodel_amd/gpu's import
+# order comment).  demodel_amd.gpu imports torch first, then _hip.
 try:
-    from demodel_amd import _hip as _early_hip
-
-    _EARLY_PROBE = _early_hip.device_probe()
-except Exception as _e:  # extension not built yet — build fixture handles it
-    _EARLY_PROBE = (0, repr(_e))
+    import demodel_amd.gpu as _early_gpu  # noqa: F401
+except Exception:
+    pass
 
 
 def pytest_configure(config):
