@@ -53,7 +53,9 @@ def test_sha256_batch_matches_hashlib(hipmod, nbytes, chunk):
     h.sha256_batch(buf.ptr, nbytes, chunk, dig.ptr, n_chunks, s.handle)
     raw = _download(h, dig, n_chunks * 32, s)
     for c in range(n_chunks):
-        words = struct.unpack_from(">8I", raw, c * 32)
+        # kernel stores digest words as native u32; canonical sha256 bytes
+        # are those words big-endian
+        words = struct.unpack_from("<8I", raw, c * 32)
         got = "".join(f"{w:08x}" for w in words)
         want = hashlib.sha256(data[c * chunk:(c + 1) * chunk]).hexdigest()
         assert got == want, f"chunk {c}"
@@ -82,8 +84,9 @@ def test_scatter_ranges(hipmod):
     s = h.Stream(0)
     src_data = os.urandom(1 << 20)
     src = _upload(h, src_data, s)
-    # three destination buffers with assorted offsets/lengths
-    specs = [(0, 1000), (1000, 65536), (66536 + 3, 999999)]
+    # three destination buffers with assorted offsets/lengths (all inside
+    # the 1 MiB source)
+    specs = [(0, 1000), (1000, 65536), (66536 + 3, 900000)]
     dsts = [h.DeviceBuffer(ln + 64) for _, ln in specs]
     desc = b""
     for (off, ln), d in zip(specs, dsts):
