@@ -25,9 +25,11 @@ __device__ __forceinline__ float f16_to_f32(const uint8_t* p) {
     if (man == 0) {
       f = sign;
     } else {
-      int e = 0;
-      while (!(man & 0x400)) { man <<= 1; ++e; }
-      f = sign | ((uint32_t)(127 - 15 - e) << 23) | ((man & 0x3FF) << 13);
+      // f16 subnormal: value = man * 2^-24; after k shifts bit10 is set
+      // and value = (1 + frac) * 2^(-14-k) -> f32 biased exp = 113 - k
+      int k = 0;
+      while (!(man & 0x400)) { man <<= 1; ++k; }
+      f = sign | ((uint32_t)(113 - k) << 23) | ((man & 0x3FF) << 13);
     }
   } else if (exp == 31) {
     f = sign | 0x7F800000u | (man << 13);

@@ -371,4 +371,12 @@ PYBIND11_MODULE(_hip, m) {
                               n_blocks, (hipStream_t)stream);
         },
         py::call_guard<py::gil_scoped_release>());
+  m.def("inflate_streams",
+        [](uintptr_t desc, int n_streams, uintptr_t stream) {
+          launch_inflate_streams((const uint64_t*)desc, n_streams, nullptr,
+                                 (hipStream_t)stream);
+        },
+        py::call_guard<py::gil_scoped_release>(),
+        "DEFLATE-inflate n_streams descriptors (8 u64 each: src, src_len, "
+        "dst, dst_cap, written, status, consumed, pad), wave per stream");
 }

@@ -1,0 +1,125 @@
+"""GPU decompression front-end: gzip containers + raw DEFLATE streams.
+
+Drives csrc/inflate.hip (wave-per-stream DEFLATE) over batches of
+compressed regions already resident in HBM: cached response bodies in
+their original Content-Encoding (reference CONTRIBUTING.md:116), gzip
+members of dataset streams, parquet page payloads.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import struct
+from dataclasses import dataclass
+
+DESC_WORDS = 8  # u64s per InflateDesc
+
+ERR = {0: "ok", -2: "format", -3: "overflow", -4: "underrun"}
+
+
+def gzip_deflate_offset(head: bytes) -> int:
+    """Offset of the raw DEFLATE stream inside a gzip member (RFC 1952)."""
+    if len(head) < 10 or head[:2] != b"\x1f\x8b":
+        raise ValueError("not gzip (bad magic)")
+    if head[2] != 8:
+        raise ValueError(f"unsupported gzip method {head[2]}")
+    flg = head[3]
+    off = 10
+    if flg & 0x04:  # FEXTRA
+        xlen = struct.unpack_from("<H", head, off)[0]
+        off += 2 + xlen
+    if flg & 0x08:  # FNAME
+        off = head.index(b"\0", off) + 1
+    if flg & 0x10:  # FCOMMENT
+        off = head.index(b"\0", off) + 1
+    if flg & 0x02:  # FHCRC
+        off += 2
+    return off
+
+
+@dataclass
+class InflateResult:
+    written: int
+    status: int
+    consumed: int
+
+    @property
+    def ok(self) -> bool:
+        return self.status == 0
+
+    @property
+    def error(self) -> str | None:
+        return None if self.ok else ERR.get(self.status, str(self.status))
+
+
+def inflate_gpu(streams: list[tuple[int, int, int, int]],
+                stream_handle=None) -> list[InflateResult]:
+    """Inflate raw DEFLATE streams on the GPU.
+
+    streams: (src_ptr, src_len, dst_ptr, dst_cap) device addresses.
+    Returns per-stream InflateResult; raises on any failed stream only if
+    the caller doesn't inspect (callers should check .ok).
+    """
+    from ...gpu import hip
+
+    h = hip()
+    n = len(streams)
+    if n == 0:
+        return []
+    own = stream_handle is None
+    s = h.Stream(0) if own else None
+    handle = s.handle if own else stream_handle
+    desc = bytearray(n * DESC_WORDS * 8)
+    for i, (src, slen, dst, cap) in enumerate(streams):
+        struct.pack_into("<8Q", desc, i * DESC_WORDS * 8,
+                         src, slen, dst, cap, 0, 0, 0, 0)
+    dbuf = h.DeviceBuffer(len(desc))
+    carr = (ctypes.c_char * len(desc)).from_buffer(desc)
+    h.h2d_async(dbuf.ptr, ctypes.addressof(carr), len(desc), handle)
+    h.inflate_streams(dbuf.ptr, n, handle)
+    h.d2h_async(ctypes.addressof(carr), dbuf.ptr, len(desc), handle)
+    if own:
+        s.sync()
+    else:
+        h.device_sync()
+    out = []
+    for i in range(n):
+        vals = struct.unpack_from("<8Q", desc, i * DESC_WORDS * 8)
+        written, status_u, consumed = vals[4], vals[5], vals[6]
+        status = status_u - (1 << 64) if status_u >= (1 << 63) else status_u
+        out.append(InflateResult(written=written, status=int(status),
+                                 consumed=consumed))
+    return out
+
+
+def gunzip_blob_gpu(blob, out_size: int | None = None):
+    """Decompress a gzip blob landed in HBM -> new DeviceBuffer.
+
+    Uses the gzip ISIZE trailer for sizing (mod 2^32; capacity doubles on
+    overflow).  Single-member fast path; multi-member walks `consumed`.
+    """
+    from ...gpu import hip
+
+    h = hip()
+    off = gzip_deflate_offset(blob.head)
+    if out_size is None:
+        tail = bytearray(8)
+        addr = ctypes.addressof((ctypes.c_char * 8).from_buffer(tail))
+        s = h.Stream(0)
+        h.d2h_async(addr, blob.buffer.ptr + blob.nbytes - 8, 8, s.handle)
+        s.sync()
+        out_size = struct.unpack("<I", bytes(tail[4:8]))[0]
+        if out_size == 0:
+            out_size = 1
+    cap = out_size
+    for _ in range(3):
+        dst = h.DeviceBuffer(cap)
+        res = inflate_gpu([(blob.buffer.ptr + off,
+                            blob.nbytes - off - 8, dst.ptr, cap)])[0]
+        if res.ok:
+            return dst, res
+        if res.status == -3:  # overflow: ISIZE wrapped (>4 GiB payloads)
+            cap *= 4
+            continue
+        raise IOError(f"GPU inflate failed: {res.error}")
+    raise IOError("GPU inflate: capacity growth exhausted")

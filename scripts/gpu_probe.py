@@ -104,5 +104,39 @@ def main():
           flush=True)
 
 
+def inflate_bench():
+    import zlib
+
+    from demodel_amd.engine.formats.compress import inflate_gpu
+    from demodel_amd.gpu import hip
+
+    h = hip()
+    s = h.Stream(0)
+    # 256 streams x 4 MiB of mixed text/random payload
+    base = (b"some plainly compressible text payload " * 1000
+            + os.urandom(1 << 20))
+    data = (base * ((4 << 20) // len(base) + 1))[:(4 << 20)]
+    comp = zlib.compressobj(6, zlib.DEFLATED, -15)
+    blob = comp.compress(data) + comp.flush()
+    import ctypes
+
+    n_streams = 256
+    src = h.DeviceBuffer(len(blob))
+    carr = (ctypes.c_char * len(blob)).from_buffer_copy(blob)
+    h.h2d_async(src.ptr, ctypes.addressof(carr), len(blob), s.handle)
+    s.sync()
+    dsts = [h.DeviceBuffer(len(data)) for _ in range(n_streams)]
+    streams = [(src.ptr, len(blob), d.ptr, len(data)) for d in dsts]
+
+    t = bench(lambda: inflate_gpu(streams), iters=2, warmup=1)
+    out_bytes = len(data) * n_streams
+    print(json.dumps({"op": "inflate_deflate", "streams": n_streams,
+                      "mib_out_each": len(data) >> 20, "s": round(t, 3),
+                      "GBps_out": round(out_bytes / t / 1e9, 2),
+                      "ratio": round(len(data) / len(blob), 2)}),
+          flush=True)
+
+
 if __name__ == "__main__":
     main()
+    inflate_bench()
