@@ -379,4 +379,13 @@ PYBIND11_MODULE(_hip, m) {
         py::call_guard<py::gil_scoped_release>(),
         "DEFLATE-inflate n_streams descriptors (8 u64 each: src, src_len, "
         "dst, dst_cap, written, status, consumed, pad), wave per stream");
+  m.def("zstd_frames",
+        [](uintptr_t desc, int n_frames, uintptr_t stream) {
+          launch_zstd_frames((const uint64_t*)desc, n_frames, nullptr,
+                             (hipStream_t)stream);
+        },
+        py::call_guard<py::gil_scoped_release>(),
+        "zstd-decompress n_frames descriptors (8 u64 each: src, src_len, "
+        "dst, dst_cap, written, status, consumed, ws), wave per frame; ws "
+        "needs >= 144 KiB per frame");
 }

@@ -14,7 +14,8 @@ from dataclasses import dataclass
 
 DESC_WORDS = 8  # u64s per InflateDesc
 
-ERR = {0: "ok", -2: "format", -3: "overflow", -4: "underrun"}
+ERR = {0: "ok", -1: "magic", -2: "format", -3: "overflow",
+       -4: "underrun", -5: "dictionary-unsupported"}
 
 
 def gzip_deflate_offset(head: bytes) -> int:
@@ -77,6 +78,50 @@ def inflate_gpu(streams: list[tuple[int, int, int, int]],
     carr = (ctypes.c_char * len(desc)).from_buffer(desc)
     h.h2d_async(dbuf.ptr, ctypes.addressof(carr), len(desc), handle)
     h.inflate_streams(dbuf.ptr, n, handle)
+    h.d2h_async(ctypes.addressof(carr), dbuf.ptr, len(desc), handle)
+    if own:
+        s.sync()
+    else:
+        h.device_sync()
+    out = []
+    for i in range(n):
+        vals = struct.unpack_from("<8Q", desc, i * DESC_WORDS * 8)
+        written, status_u, consumed = vals[4], vals[5], vals[6]
+        status = status_u - (1 << 64) if status_u >= (1 << 63) else status_u
+        out.append(InflateResult(written=written, status=int(status),
+                                 consumed=consumed))
+    return out
+
+
+ZSTD_WS_BYTES = 144 << 10
+
+
+def zstd_gpu(frames: list[tuple[int, int, int, int]],
+             stream_handle=None) -> list[InflateResult]:
+    """Decompress zstd frames on the GPU (csrc/zstd_kernel.hip).
+
+    frames: (src_ptr, src_len, dst_ptr, dst_cap) device addresses.
+    A 144 KiB workspace per frame is allocated here.
+    """
+    from ...gpu import hip
+
+    h = hip()
+    n = len(frames)
+    if n == 0:
+        return []
+    own = stream_handle is None
+    s = h.Stream(0) if own else None
+    handle = s.handle if own else stream_handle
+    ws = h.DeviceBuffer(n * ZSTD_WS_BYTES)
+    desc = bytearray(n * DESC_WORDS * 8)
+    for i, (src, slen, dst, cap) in enumerate(frames):
+        struct.pack_into("<8Q", desc, i * DESC_WORDS * 8,
+                         src, slen, dst, cap, 0, 0, 0,
+                         ws.ptr + i * ZSTD_WS_BYTES)
+    dbuf = h.DeviceBuffer(len(desc))
+    carr = (ctypes.c_char * len(desc)).from_buffer(desc)
+    h.h2d_async(dbuf.ptr, ctypes.addressof(carr), len(desc), handle)
+    h.zstd_frames(dbuf.ptr, n, handle)
     h.d2h_async(ctypes.addressof(carr), dbuf.ptr, len(desc), handle)
     if own:
         s.sync()
