@@ -48,7 +48,8 @@ def _run(cmd: list[str]) -> None:
 
 def build_native(force: bool = False) -> str:
     out = os.path.join(PKG_DIR, "_native.so")
-    srcs = [os.path.join(CSRC, "certs.cpp")]
+    srcs = [os.path.join(CSRC, "certs.cpp"),
+            os.path.join(CSRC, "zstd_host.cpp")]
     if not force and _newer(out, srcs):
         return out
     cmd = (

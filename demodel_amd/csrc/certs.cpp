@@ -172,8 +172,12 @@ std::pair<std::string, std::string> leaf_create(const std::string& ca_cert_pem,
 
 }  // namespace
 
+void register_zstd_host(py::module_& m);
+
 PYBIND11_MODULE(_native, m) {
-  m.doc() = "demodel-amd native helpers (libcrypto cert minting)";
+  m.doc() = "demodel-amd native helpers (libcrypto cert minting, "
+            "host zstd reference decoder)";
+  register_zstd_host(m);
   m.def("ca_create", &ca_create, py::arg("ecdsa") = false,
         py::call_guard<py::gil_scoped_release>(),
         "Generate a self-signed demodel CA -> (cert_pem, key_pem)");
