@@ -26,6 +26,9 @@ GEOMS = {
     "llama3-8b": ("LLAMA3_8B", 4),
     "llama3-70b": ("LLAMA3_70B", 16),
     "tiny": (None, 2),
+    "gguf-8b": ("LLAMA3_8B", 0),     # Ollama q4_K pull + GPU dequant
+    "gguf-tiny": (None, 0),
+    "dataset": (None, 0),            # zstd streaming into HBM ring
 }
 
 
@@ -43,13 +46,23 @@ class _FanoutResult:
         self.gbps = 0.0
 
 
+TINY_GEOM = {"hidden": 256, "inter": 688, "layers": 4, "heads": 8,
+             "kv_heads": 4, "vocab": 32000}
+
+
 def make_model_files(model: str, data_dir: str):
     from demodel_amd.testing import synth
 
     if model == "tiny":
-        geom = {"hidden": 256, "inter": 688, "layers": 4, "heads": 8,
-                "kv_heads": 4, "vocab": 32000}
-        return synth.write_shards(data_dir, geom, 2)
+        return synth.write_shards(data_dir, TINY_GEOM, 2)
+    if model == "dataset":
+        return synth.write_dataset_shards(data_dir)
+    if model in ("gguf-8b", "gguf-tiny"):
+        geom = synth.LLAMA3_8B if model == "gguf-8b" else TINY_GEOM
+        path = os.path.join(data_dir, "model.gguf")
+        os.makedirs(data_dir, exist_ok=True)
+        synth.write_gguf_model(path, geom, qtype=12)
+        return {"model.gguf": path}
     geom_name, n_shards = GEOMS[model]
     geom = getattr(synth, geom_name)
     return synth.write_shards(data_dir, geom, n_shards)
@@ -139,6 +152,95 @@ def main():
             torch.cuda.synchronize()
         return res, n_t
 
+    def gguf_step(record_digests=False):
+        """Ollama-style pull (config 4): land the q4_K GGUF blob, then
+        dequant EVERY quant tensor to bf16 on-GPU; ready = bf16 weights
+        resident."""
+        from demodel_amd.engine.formats import gguf
+        from demodel_amd.gpu import have_gpu as _hg, hip
+
+        res = pull_mod.pull_hf(
+            "bench/model", endpoint=endpoint, workers=args.workers,
+            verify=args.verify, landers=landers,
+            digest_map=digest_map or None)
+        assert res.total_bytes == total_bytes
+        if record_digests:
+            for f in res.files:
+                digest_map[f.name] = f.blob.digest_blob
+        blob = res.files[0].blob
+        gg = gguf.parse(blob)
+        out_bytes = 0
+        if _hg():
+            h = hip()
+            stream = h.Stream(0)
+            outs = []
+            for t in gg.tensors:
+                if t.type_id in (2, 8, 12, 14):
+                    dst = h.DeviceBuffer(t.n_elems * 2)
+                    h.gguf_dequant(t.type_id,
+                                   blob.buffer.ptr + gg.data_offset
+                                   + t.offset,
+                                   dst.ptr, t.n_blocks, stream.handle)
+                    outs.append(dst)
+                    out_bytes += t.n_elems * 2
+            stream.sync()
+            del outs
+        if have_gpu:
+            torch.cuda.synchronize()
+        res.meta["dequant_bf16_bytes"] = out_bytes
+        return res, len(gg.tensors)
+
+    def dataset_step(record_digests=False):
+        """Dataset streaming (config 5): pull zstd shards, decompress all
+        frames wave-parallel into an HBM ring; counts DECOMPRESSED bytes
+        landed."""
+        import json as _json
+
+        from demodel_amd.engine.formats.compress import zstd_gpu
+        from demodel_amd.gpu import have_gpu as _hg, hip
+
+        res = pull_mod.pull_hf(
+            "bench/model", endpoint=endpoint, workers=args.workers,
+            verify=args.verify, landers=landers,
+            digest_map=digest_map or None)
+        if record_digests:
+            for f in res.files:
+                digest_map[f.name] = f.blob.digest_blob
+        byname = {f.name: f for f in res.files}
+        decompressed = 0
+        if _hg():
+            h = hip()
+            frames = []
+            total_out = 0
+            plan = []
+            for name, f in byname.items():
+                if not name.endswith(".zst"):
+                    continue
+                idx = _json.loads(
+                    bytes(byname[name + ".idx.json"].blob.head[
+                        :byname[name + ".idx.json"].blob.nbytes]))
+                for fr in idx["frames"]:
+                    plan.append((f.blob.buffer.ptr + fr["offset"],
+                                 fr["compressed"], fr["decompressed"]))
+                    total_out += fr["decompressed"]
+            ring = h.DeviceBuffer(total_out)
+            off = 0
+            for src, clen, dlen in plan:
+                frames.append((src, clen, ring.ptr + off, dlen))
+                off += dlen
+            results = zstd_gpu(frames)
+            for i, r in enumerate(results):
+                assert r.ok and r.written == plan[i][2], (i, r)
+            decompressed = total_out
+            del ring
+        else:
+            decompressed = res.total_bytes  # CPU plumbing mode
+        if have_gpu:
+            torch.cuda.synchronize()
+        res.meta["decompressed_bytes"] = decompressed
+        res.total_bytes = decompressed  # the bytes that LAND in the ring
+        return res, len(byname)
+
     file_sizes = [(n, os.path.getsize(p)) for n, p in sorted(files.items())]
 
     def fanout_step(record_digests=False):
@@ -174,7 +276,16 @@ def main():
             torch.cuda.synchronize()
         return out
 
+    step_bytes = total_bytes  # may be overridden by workloads below
+
     def one_step(record_digests=False):
+        nonlocal step_bytes
+        if args.model in ("gguf-8b", "gguf-tiny"):
+            return gguf_step(record_digests)
+        if args.model == "dataset":
+            res, n = dataset_step(record_digests)
+            step_bytes = res.total_bytes
+            return res, n
         if args.mode == "dp":
             return dp_step(record_digests)
         res = fanout_step(record_digests)
@@ -203,7 +314,7 @@ def main():
         dist.all_reduce(et, op=dist.ReduceOp.MAX)
         elapsed = float(et.item())
 
-    steps_bytes = total_bytes * args.steps * world
+    steps_bytes = step_bytes * args.steps * world
     value = steps_bytes / elapsed / 1e9
     ms_per_step = elapsed / args.steps * 1000.0
 

@@ -191,14 +191,15 @@ def build_file(path: str, tensors: list[tuple[str, tuple[int, ...], int]],
             head += struct.pack("<Q", d)
         head += struct.pack("<IQ", t.type_id, t.offset)
     data_offset = (len(head) + align - 1) // align * align
+    # payloads tile a fixed random block so multi-GB synthesis is IO-bound,
+    # not RNG-bound
+    tile = rng.integers(0, 256, size=16 << 20, dtype=np.uint8).tobytes()
     with open(path, "wb") as f:
         f.write(head)
         f.write(b"\0" * (data_offset - len(head)))
         for t in infos:
-            payload = rng.integers(0, 256, size=t.nbytes,
-                                   dtype=np.uint8).tobytes()
-            # keep f16/f32 payloads finite so dequant comparisons are exact
             if t.type_id in (0, 1, 30):
+                # keep float payloads finite so dequant comparisons work
                 vals = rng.standard_normal(t.n_elems)
                 if t.type_id == 0:
                     payload = vals.astype(np.float32).tobytes()
@@ -207,7 +208,13 @@ def build_file(path: str, tensors: list[tuple[str, tuple[int, ...], int]],
                 else:
                     payload = (vals.astype(np.float32).view(np.uint32)
                                >> 16).astype(np.uint16).tobytes()
-            f.write(payload)
+                f.write(payload)
+            else:
+                left = t.nbytes
+                while left > 0:
+                    take = min(left, len(tile))
+                    f.write(tile[:take])
+                    left -= take
             pad = (t.nbytes + align - 1) // align * align - t.nbytes
             f.write(b"\0" * pad)
     gg = parse_bytes(open(path, "rb").read(min(

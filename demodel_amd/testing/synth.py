@@ -101,6 +101,53 @@ def write_shards(out_dir: str, geom: dict, n_shards: int,
     return files
 
 
+def write_dataset_shards(out_dir: str, n_shards: int = 4,
+                         frames_per_shard: int = 48,
+                         frame_bytes: int = 8 << 20,
+                         level: int = 1, reuse: bool = True
+                         ) -> dict[str, str]:
+    """Synthetic c4-en-like text dataset: shards of concatenated zstd
+    frames (one frame ~= one record batch), plus a .idx.json sidecar per
+    shard with frame offsets/sizes (plays the role parquet metadata plays
+    for page locations).  Returns {rfilename: path} incl. sidecars."""
+    import pyarrow as pa
+
+    os.makedirs(out_dir, exist_ok=True)
+    rng = np.random.default_rng(42)
+    words = [f"w{i:04d}" for i in range(30000)]
+    codec = pa.Codec("zstd", compression_level=level)
+
+    # one representative text block, tiled with per-frame mutation
+    idx = rng.integers(0, len(words), size=frame_bytes // 6)
+    base_text = " ".join(words[i] for i in idx).encode()[:frame_bytes]
+
+    files: dict[str, str] = {}
+    for s in range(n_shards):
+        fname = f"c4-train.{s:05d}-of-{n_shards:05d}.zst"
+        path = os.path.join(out_dir, fname)
+        idx_path = path + ".idx.json"
+        if reuse and os.path.exists(path) and os.path.exists(idx_path):
+            files[fname] = path
+            files[fname + ".idx.json"] = idx_path
+            continue
+        frames = []
+        off = 0
+        with open(path, "wb") as f:
+            for k in range(frames_per_shard):
+                stamp = f"[shard {s} frame {k}] ".encode()
+                raw = stamp + base_text[len(stamp):]
+                comp = bytes(codec.compress(raw))
+                f.write(comp)
+                frames.append({"offset": off, "compressed": len(comp),
+                               "decompressed": len(raw)})
+                off += len(comp)
+        with open(idx_path, "w") as f:
+            json.dump({"frames": frames}, f)
+        files[fname] = path
+        files[fname + ".idx.json"] = idx_path
+    return files
+
+
 def write_gguf_model(path: str, geom: dict, qtype: int = 12,
                      reuse: bool = True):
     """Synthetic GGUF (default q4_K) with llama-shaped 2-D tensors."""
