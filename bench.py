@@ -169,22 +169,22 @@ def main():
                 digest_map[f.name] = f.blob.digest_blob
         blob = res.files[0].blob
         gg = gguf.parse(blob)
-        out_bytes = 0
-        if _hg():
+        quants = [t for t in gg.tensors if t.type_id in (2, 8, 12, 14)]
+        out_bytes = sum(t.n_elems * 2 for t in quants)
+        if _hg() and quants:
             h = hip()
             stream = h.Stream(0)
-            outs = []
-            for t in gg.tensors:
-                if t.type_id in (2, 8, 12, 14):
-                    dst = h.DeviceBuffer(t.n_elems * 2)
-                    h.gguf_dequant(t.type_id,
-                                   blob.buffer.ptr + gg.data_offset
-                                   + t.offset,
-                                   dst.ptr, t.n_blocks, stream.handle)
-                    outs.append(dst)
-                    out_bytes += t.n_elems * 2
+            # one arena for all bf16 outputs: per-tensor hipMalloc would
+            # serialize ~300 allocations between the async launches
+            arena = h.DeviceBuffer(out_bytes)
+            off = 0
+            for t in quants:
+                h.gguf_dequant(t.type_id,
+                               blob.buffer.ptr + gg.data_offset + t.offset,
+                               arena.ptr + off, t.n_blocks, stream.handle)
+                off += t.n_elems * 2
             stream.sync()
-            del outs
+            del arena
         if have_gpu:
             torch.cuda.synchronize()
         res.meta["dequant_bf16_bytes"] = out_bytes

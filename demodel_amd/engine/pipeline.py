@@ -55,6 +55,18 @@ class LandedBlob:
         return torch.from_dlpack(self.buffer.to_dlpack())
 
 
+class LandingError(IOError):
+    """A fill source died mid-landing; `landed` bytes (from the segment's
+    start) are safely in HBM — a resume can continue from there with a
+    Range request."""
+
+    def __init__(self, landed: int, cause: BaseException | None = None):
+        self.landed = landed
+        self.cause = cause
+        super().__init__(f"landing interrupted after {landed} bytes"
+                         + (f": {cause!r}" if cause else ""))
+
+
 class DigestMismatch(IOError):
     def __init__(self, chunk_index: int, verify_chunk: int):
         self.chunk_index = chunk_index
@@ -168,11 +180,12 @@ class Lander:
             tf = time.perf_counter()
             got = 0
             while got < want:
-                n = fill(view[got:])
+                try:
+                    n = fill(view[got:])
+                except Exception as e:
+                    raise LandingError(off, e) from e
                 if n <= 0:
-                    raise IOError(
-                        f"blob underrun at {base_off + off + got}"
-                        f"/{file_size} bytes")
+                    raise LandingError(off)
                 got += n
             fill_s += time.perf_counter() - tf
             if keep_head and len(head) < self.head_bytes:
@@ -201,6 +214,12 @@ class Lander:
     def sync(self) -> None:
         self.copy_stream.sync()
         self.verify_stream.sync()
+
+    def alloc(self, nbytes: int):
+        return self._h.DeviceBuffer(max(nbytes, 1))
+
+    def finish_verify(self, buf, nbytes: int) -> bytes:
+        return self._gpu_chunk_digests(buf, nbytes)
 
     def _gpu_chunk_digests(self, buf, nbytes: int) -> bytes:
         h = self._h
@@ -309,6 +328,44 @@ class HostLander:
         self.slab_bytes = slab_bytes
         self.verify_chunk = verify_chunk
         self.head_bytes = head_bytes
+
+    def alloc(self, nbytes: int):
+        return bytearray(nbytes)
+
+    def land_into(self, buf, base_off: int, fill, nbytes: int,
+                  file_size: int | None = None, chain=None, gpu_state=None,
+                  keep_head: bool = False) -> tuple[bytearray, float]:
+        mv = memoryview(buf)
+        off = 0
+        t0 = time.perf_counter()
+        while off < nbytes:
+            want = min(self.slab_bytes, nbytes - off)
+            got = 0
+            while got < want:
+                lo = base_off + off + got
+                try:
+                    n = fill(mv[lo:base_off + off + want])
+                except Exception as e:
+                    raise LandingError(off, e) from e
+                if n <= 0:
+                    raise LandingError(off)
+                got += n
+            if chain is not None:
+                chain.update(mv[base_off + off:base_off + off + want])
+            off += want
+        head = bytearray(
+            mv[base_off:base_off + min(nbytes, self.head_bytes)]) \
+            if keep_head else bytearray()
+        return head, time.perf_counter() - t0
+
+    def finish_verify(self, buf, nbytes: int) -> bytes:
+        mv = memoryview(buf)
+        return b"".join(
+            hashlib.sha256(mv[o:o + self.verify_chunk]).digest()
+            for o in range(0, nbytes, self.verify_chunk))
+
+    def sync(self) -> None:
+        pass
 
     def land(self, fill, nbytes: int, verify: bool = True,
              host_chain: bool = False, gpu_chain: bool = False,

@@ -137,6 +137,61 @@ def _etag_sha256(etag: str | None) -> str | None:
 
 SEGMENT_MIN = 512 << 20     # segment blobs bigger than this
 MAX_SEGMENTS = 8
+RESUME_RETRIES = 4
+
+
+def _land_with_resume(lander, open_fn, nbytes: int, verify: bool,
+                      expected_digests: bytes | None,
+                      host_chain: bool = False,
+                      retries: int | None = None):
+    """Land a blob with automatic Range-resume on mid-stream failures.
+
+    open_fn(offset) -> fill callable streaming bytes [offset, nbytes).
+    Works with both Lander (HBM) and HostLander (RAM).
+    """
+    import hashlib
+
+    from .pipeline import (LandedBlob, LandingError, check_digests)
+
+    if retries is None:
+        retries = RESUME_RETRIES
+    buf = lander.alloc(nbytes)
+    chain = hashlib.sha256() if host_chain else None
+    head = bytearray()
+    landed = 0
+    attempt = 0
+    fill = open_fn(0)
+    while landed < nbytes:
+        try:
+            hd, _ = lander.land_into(
+                buf, landed, fill, nbytes - landed, file_size=nbytes,
+                chain=chain, keep_head=landed < lander.head_bytes)
+            if hd:
+                head += hd[:max(0, lander.head_bytes - len(head))]
+            landed = nbytes
+        except LandingError as e:
+            landed += e.landed
+            attempt += 1
+            if attempt > retries:
+                raise IOError(
+                    f"pull failed after {attempt} attempts at byte "
+                    f"{landed}/{nbytes}") from e
+            log.info("resuming pull at byte %d/%d (attempt %d)",
+                     landed, nbytes, attempt)
+            fill = open_fn(landed)
+    lander.sync()
+    device = ("cpu" if isinstance(buf, bytearray)
+              else f"cuda:{lander.device_index}")
+    blob = LandedBlob(nbytes=nbytes, device=device, buffer=buf,
+                      verify_chunk=lander.verify_chunk, head=bytes(head))
+    if verify or expected_digests is not None:
+        blob.digest_blob = lander.finish_verify(buf, nbytes)
+        if expected_digests is not None:
+            check_digests(blob.digest_blob, expected_digests,
+                          lander.verify_chunk)
+    if chain is not None:
+        blob.sha256 = chain.hexdigest()
+    return blob
 
 
 def _pull_blob(landers: LanderPool, name: str, url: str,
@@ -177,13 +232,41 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
             if nbytes < 0:
                 raise fetch.FetchError(f"no content-length for blob {url}")
             lander = landers.get()
-            blob = lander.land(
-                src.fill, nbytes,
-                verify=do_verify,
-                host_chain=(verify == "digest"),
-                gpu_chain=(verify == "gpu-digest"),
-                expected_digests=expected_digests,
-            )
+            if verify == "gpu-digest":
+                # the sequential GPU chain can't resume across re-fetches
+                blob = lander.land(
+                    src.fill, nbytes, verify=do_verify, gpu_chain=True,
+                    expected_digests=expected_digests)
+            else:
+                extra_sources = []
+                first = [True]
+
+                def open_fn(offset):
+                    if offset == 0 and first[0]:
+                        first[0] = False
+                        return src.fill
+                    rng_headers = dict(headers or {})
+                    if offset:
+                        rng_headers["Range"] = f"bytes={offset}-"
+                    s2 = fetch.http_get(src.resp.url, cafile=cafile,
+                                        insecure=insecure,
+                                        headers=rng_headers)
+                    want = 206 if offset else 200
+                    if s2.status != want:
+                        s2.close()
+                        raise fetch.FetchError(
+                            f"resume GET -> {s2.status} (want {want})")
+                    extra_sources.append(s2)
+                    return s2.fill
+
+                try:
+                    blob = _land_with_resume(
+                        lander, open_fn, nbytes, verify=do_verify,
+                        expected_digests=expected_digests,
+                        host_chain=(verify == "digest"))
+                finally:
+                    for s2 in extra_sources:
+                        s2.close()
         nbytes = blob.nbytes
     finally:
         src.close()

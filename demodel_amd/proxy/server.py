@@ -440,13 +440,44 @@ class ProxyServer:
         log.info("MISS %s %s -> %d (%d bytes)", req.method, uri,
                  resp.status, total)
 
+    @staticmethod
+    def _parse_range(spec: str | None, size: int):
+        """Single-range parse -> (start, length) or None for full body."""
+        if not spec or not spec.startswith("bytes=") or size == 0:
+            return None
+        part = spec[len("bytes="):].split(",")[0].strip()
+        s, _, e = part.partition("-")
+        try:
+            if s:
+                start = int(s)
+                end = int(e) if e else size - 1
+            else:
+                start = max(0, size - int(e))
+                end = size - 1
+        except ValueError:
+            return None
+        if start >= size or end < start:
+            return None
+        return start, min(end, size - 1) - start + 1
+
     async def _serve_cached(self, hit, writer, req: RequestHead) -> None:
         out = ResponseHead("HTTP/1.1", hit.status, hit.reason or "OK",
                            [(k, v) for k, v in hit.headers
                             if k.lower() not in ("connection", "keep-alive",
                                                  "transfer-encoding",
-                                                 "content-length")])
-        out.replace("Content-Length", str(hit.body_size))
+                                                 "content-length",
+                                                 "content-range")])
+        start, length = 0, hit.body_size
+        rng = self._parse_range(req.get("range"), hit.body_size) \
+            if hit.status == 200 else None
+        if rng is not None:
+            start, length = rng
+            out.status, out.reason = 206, "Partial Content"
+            out.replace("Content-Range",
+                        f"bytes {start}-{start + length - 1}"
+                        f"/{hit.body_size}")
+        out.replace("Content-Length", str(length))
+        out.replace("Accept-Ranges", "bytes")
         out.replace("Connection", "keep-alive")
         out.replace("X-Demodel-Cache", "HIT")
         writer.write(http1.serialize_response(out))
@@ -458,13 +489,15 @@ class ProxyServer:
                 from ..utils.netio import sendfile_threaded
 
                 try:
-                    await sendfile_threaded(writer, f, 0, hit.body_size)
+                    await sendfile_threaded(writer, f, start, length)
                 except (NotImplementedError, RuntimeError, OSError):
-                    f.seek(0)
-                    while True:
-                        data = f.read(http1.CHUNK)
+                    f.seek(start)
+                    left = length
+                    while left > 0:
+                        data = f.read(min(http1.CHUNK, left))
                         if not data:
                             break
+                        left -= len(data)
                         writer.write(data)
                         await writer.drain()
         await writer.drain()

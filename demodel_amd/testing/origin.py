@@ -47,6 +47,10 @@ class FakeOrigin:
         self.port: int | None = None
         self._server = None
         self.requests: list[str] = []  # log of "<METHOD> <path>"
+        # fault injection: path-substring -> serve N body bytes then drop
+        # the connection (consumed on first match) — exercises the
+        # engine's Range-resume path
+        self.drop_once: dict[str, int] = {}
 
     # ------------------------------------------------------------------ #
     # content registration
@@ -279,6 +283,11 @@ class FakeOrigin:
 
     async def _serve_file(self, writer, req: RequestHead, path: str,
                           etag: str, extra: list[tuple[str, str]] | None = None):
+        drop_at = None
+        for key in list(self.drop_once):
+            if key in req.target:
+                drop_at = self.drop_once.pop(key)
+                break
         size = os.path.getsize(path)
         start, end = 0, size - 1
         status, reason = 200, "OK"
@@ -303,6 +312,14 @@ class FakeOrigin:
             ResponseHead("HTTP/1.1", status, reason, headers)))
         await writer.drain()
         if req.method == "HEAD":
+            return
+        if drop_at is not None:
+            # fault injection: truncate the body and kill the connection
+            with open(path, "rb") as f:
+                f.seek(start)
+                writer.write(f.read(min(drop_at, length)))
+                await writer.drain()
+            writer.transport.abort()
             return
         loop = asyncio.get_running_loop()
         with open(path, "rb") as f:

@@ -101,9 +101,9 @@ def write_shards(out_dir: str, geom: dict, n_shards: int,
     return files
 
 
-def write_dataset_shards(out_dir: str, n_shards: int = 4,
-                         frames_per_shard: int = 48,
-                         frame_bytes: int = 8 << 20,
+def write_dataset_shards(out_dir: str, n_shards: int = 8,
+                         frames_per_shard: int = 256,
+                         frame_bytes: int = 2 << 20,
                          level: int = 1, reuse: bool = True
                          ) -> dict[str, str]:
     """Synthetic c4-en-like text dataset: shards of concatenated zstd

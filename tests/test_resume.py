@@ -1,0 +1,83 @@
+"""Range-resume tests: the engine survives mid-stream connection drops
+(Range continuation), and the proxy serves Range slices from cache."""
+
+import hashlib
+import os
+import urllib.request
+
+import pytest
+
+from demodel_amd.engine import pull as pull_mod
+from helpers import Stack
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    s = Stack(tmp_path)
+    yield s
+    s.close()
+
+
+def test_pull_resumes_after_drop(stack, tmp_path):
+    data = os.urandom(2 << 20)
+    p = tmp_path / "big.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/r", {"big.bin": str(p)})
+    # the first GET of the blob dies after 300 KiB
+    stack.origin.drop_once["big.bin"] = 300 << 10
+
+    # small slabs so the drop at 300 KiB leaves a completed-slab prefix
+    # and the resume continues WITH a Range request
+    res = pull_mod.pull_hf("org/r", endpoint=stack.origin_base,
+                           verify="digest", workers=1,
+                           slab_bytes=256 << 10)
+    f = res.files[0]
+    assert f.nbytes == len(data)
+    assert f.blob.sha256 == hashlib.sha256(data).hexdigest()
+    assert f.digest_ok is True
+    assert bytes(f.blob.buffer) == data
+    assert not stack.origin.drop_once  # fault fired
+    # origin saw the blob requested twice (original + resume)
+    blob_gets = [r for r in stack.origin.requests
+                 if r.startswith("GET") and "big.bin" in r
+                 and "/cdn/" in r]
+    assert len(blob_gets) == 2
+
+
+def test_pull_fails_when_retries_exhausted(stack, tmp_path, monkeypatch):
+    data = os.urandom(1 << 20)
+    p = tmp_path / "cursed.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/c", {"cursed.bin": str(p)})
+    stack.origin.drop_once["cursed.bin"] = 100
+    import demodel_amd.engine.pull as pm
+
+    monkeypatch.setattr(pm, "RESUME_RETRIES", 0)
+    with pytest.raises(IOError):
+        pull_mod.pull_hf("org/c", endpoint=stack.origin_base,
+                         verify="chunked", workers=1)
+
+
+def test_proxy_serves_range_from_cache(stack, tmp_path):
+    data = os.urandom(500_000)
+    p = tmp_path / "r.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/rr", {"r.bin": str(p)})
+    url = f"{stack.endpoint}/org/rr/resolve/main/r.bin"
+    # prime the cache
+    with urllib.request.urlopen(url, timeout=20) as r:
+        assert r.read() == data
+
+    req = urllib.request.Request(url, headers={"Range": "bytes=1000-4999"})
+    with urllib.request.urlopen(req, timeout=20) as r:
+        assert r.status == 206
+        assert r.headers["X-Demodel-Cache"] == "HIT"
+        assert r.headers["Content-Range"] == \
+            f"bytes 1000-4999/{len(data)}"
+        assert r.read() == data[1000:5000]
+
+    # suffix range
+    req = urllib.request.Request(url, headers={"Range": "bytes=-100"})
+    with urllib.request.urlopen(req, timeout=20) as r:
+        assert r.status == 206
+        assert r.read() == data[-100:]
