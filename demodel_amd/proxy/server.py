@@ -47,11 +47,29 @@ class ProxyServer:
         self.transfers = TransferLog()
         self._server: asyncio.AbstractServer | None = None
         self.port: int | None = None
-        # reverse-mode routing table: path-prefix -> upstream base
-        self.reverse_routes: list[tuple[str, str]] = [
-            ("/v2/", "https://registry.ollama.ai"),
-            ("/", "https://huggingface.co"),
-        ]
+        # reverse-mode routing table: path-prefix -> upstream base.
+        # Overridable via DEMODEL_REVERSE_ROUTES="/v2/=https://host,/=..."
+        # or DEMODEL_REVERSE_UPSTREAM (catch-all) — pointing the catch-all
+        # at another demodel node makes this node a caching peer of it
+        # (the reference's "syncing, distributing" story, README.md:6-8).
+        import os as _os
+
+        routes_env = _os.environ.get("DEMODEL_REVERSE_ROUTES")
+        if routes_env:
+            self.reverse_routes = []
+            for item in routes_env.split(","):
+                prefix, _, base = item.partition("=")
+                if prefix and base:
+                    self.reverse_routes.append((prefix.strip(),
+                                                base.strip()))
+        else:
+            catch_all = _os.environ.get("DEMODEL_REVERSE_UPSTREAM",
+                                        "https://huggingface.co")
+            self.reverse_routes = [
+                ("/v2/", _os.environ.get("DEMODEL_OLLAMA_UPSTREAM",
+                                         "https://registry.ollama.ai")),
+                ("/", catch_all),
+            ]
         self._upstream_ssl: ssl.SSLContext | None = None
 
     # ------------------------------------------------------------------ #
@@ -203,6 +221,32 @@ class ProxyServer:
                     u.hostname, port, is_tls, t
         raise ProtocolError(f"no reverse route for {t!r}")
 
+    async def _serve_stats(self, head: RequestHead, writer) -> bool:
+        """GET /__demodel/stats — transfer metrics (observability; the
+        reference had only two println hooks)."""
+        import json as _json
+
+        recs = self.transfers.records
+        hits = [r for r in recs if r.get("event") == "hit"]
+        misses = [r for r in recs if r.get("event") == "miss"]
+        body = _json.dumps({
+            "requests": len(recs),
+            "cache_hits": len(hits),
+            "cache_misses": len(misses),
+            "hit_bytes": sum(r.get("bytes", 0) for r in hits),
+            "miss_bytes": sum(r.get("bytes", 0) for r in misses),
+            "mitm_hosts": self.cfg.mitm_hosts,
+        }, indent=1).encode()
+        out = ResponseHead("HTTP/1.1", 200, "OK",
+                           [("Content-Type", "application/json"),
+                            ("Content-Length", str(len(body))),
+                            ("Connection", "keep-alive")])
+        writer.write(http1.serialize_response(out))
+        if head.method != "HEAD":
+            writer.write(body)
+        await writer.drain()
+        return head.get("connection", "").lower() == "close"
+
     @staticmethod
     def _absolute_uri(location: str, base_uri: str) -> str:
         """Resolve a Location header against the request URI."""
@@ -222,6 +266,8 @@ class ProxyServer:
 
     async def _handle_request(self, head: RequestHead, reader, writer,
                               tls_host: str | None) -> bool:
+        if head.target.startswith("/__demodel/"):
+            return await self._serve_stats(head, writer)
         uri, host, port, is_tls, path = self._canonical_uri(head, tls_host)
         reverse_mode = tls_host is None and "://" not in head.target
 
