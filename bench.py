@@ -50,7 +50,7 @@ TINY_GEOM = {"hidden": 256, "inter": 688, "layers": 4, "heads": 8,
              "kv_heads": 4, "vocab": 32000}
 
 
-def make_model_files(model: str, data_dir: str):
+def make_model_files(model: str, data_dir: str, n_shards_override=None):
     from demodel_amd.testing import synth
 
     if model == "tiny":
@@ -65,7 +65,8 @@ def make_model_files(model: str, data_dir: str):
         return {"model.gguf": path}
     geom_name, n_shards = GEOMS[model]
     geom = getattr(synth, geom_name)
-    return synth.write_shards(data_dir, geom, n_shards)
+    return synth.write_shards(data_dir, geom,
+                              n_shards_override or n_shards)
 
 
 def main():
@@ -86,6 +87,8 @@ def main():
                     choices=["chunked", "digest", "gpu-digest", "off"])
     ap.add_argument("--slab-mib", type=int, default=64)
     ap.add_argument("--data-dir", default=None)
+    ap.add_argument("--shards", type=int, default=None,
+                    help="override safetensors shard count")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -113,9 +116,11 @@ def main():
 
     # ---- setup (untimed): synth model + per-rank loopback origin -------
     data_dir = args.data_dir or os.path.join(
-        os.environ.get("TMPDIR", "/tmp"), f"demodel_bench_{args.model}")
+        os.environ.get("TMPDIR", "/tmp"),
+        f"demodel_bench_{args.model}"
+        + (f"_s{args.shards}" if args.shards else ""))
     t = time.time()
-    files = make_model_files(args.model, data_dir)
+    files = make_model_files(args.model, data_dir, args.shards)
     log(f"model files ready in {time.time() - t:.1f}s "
         f"({sum(os.path.getsize(p) for p in files.values()) / 1e9:.2f} GB)")
 

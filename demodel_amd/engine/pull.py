@@ -135,8 +135,8 @@ def _etag_sha256(etag: str | None) -> str | None:
     return None
 
 
-SEGMENT_MIN = 512 << 20     # segment blobs bigger than this
-MAX_SEGMENTS = 8
+SEGMENT_MIN = int(os.environ.get("DEMODEL_SEGMENT_MIN_MB", "512")) << 20
+MAX_SEGMENTS = int(os.environ.get("DEMODEL_MAX_SEGMENTS", "8"))
 RESUME_RETRIES = 4
 
 
