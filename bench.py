@@ -150,7 +150,8 @@ def main():
         assert res.total_bytes == total_bytes, res.total_bytes
         if record_digests:
             for f in res.files:
-                digest_map[f.name] = f.blob.digest_blob
+                if f.blob.digest_blob:
+                    digest_map[f.name] = f.blob.digest_blob
         # model-ready: materialize the tensor views
         n_t = len(res.tensors())
         if have_gpu:

@@ -282,9 +282,12 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
     pf = PulledFile(name=name, url=url, nbytes=nbytes, blob=blob,
                     etag=etag, digest_ok=ok,
                     seconds=time.perf_counter() - t0)
-    log.info("landed %s: %d bytes in %.3fs (%.2f GB/s) on %s",
+    t = blob.timings
+    log.info("landed %s: %d bytes in %.3fs (%.2f GB/s) on %s%s",
              name, nbytes, pf.seconds, nbytes / max(pf.seconds, 1e-9) / 1e9,
-             blob.device)
+             blob.device,
+             (f" [fill {t['fill_s']:.3f}s land {t['land_s']:.3f}s "
+              f"verify {t['verify_s']:.3f}s]") if t else "")
     return pf
 
 
