@@ -82,10 +82,10 @@ def main():
                          "shard: rank r pulls 1/N of the manifest, "
                          "file-wise RCCL broadcast reassembly (config 3); "
                          "broadcast: rank 0 pulls, RCCL fan-out (R1)")
-    ap.add_argument("--workers", type=int, default=4)
+    ap.add_argument("--workers", type=int, default=8)
     ap.add_argument("--verify", default="chunked",
                     choices=["chunked", "digest", "gpu-digest", "off"])
-    ap.add_argument("--slab-mib", type=int, default=64)
+    ap.add_argument("--slab-mib", type=int, default=128)
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--shards", type=int, default=None,
                     help="override safetensors shard count")
