@@ -120,7 +120,13 @@ def main():
         f"demodel_bench_{args.model}"
         + (f"_s{args.shards}" if args.shards else ""))
     t = time.time()
-    files = make_model_files(args.model, data_dir, args.shards)
+    # rank 0 generates the shared synthetic files; others wait
+    if rank == 0:
+        files = make_model_files(args.model, data_dir, args.shards)
+    if dist:
+        dist.barrier()
+    if rank != 0:
+        files = make_model_files(args.model, data_dir, args.shards)
     log(f"model files ready in {time.time() - t:.1f}s "
         f"({sum(os.path.getsize(p) for p in files.values()) / 1e9:.2f} GB)")
 
@@ -303,6 +309,7 @@ def main():
         res, _ = one_step(record_digests=(i == 0))
         log(f"warmup {i}: {res.seconds_to_ready:.2f}s "
             f"({res.gbps:.2f} GB/s)")
+        del res  # free HBM before the next pull (70B barely fits twice)
 
     sync()
     t0 = time.perf_counter()
@@ -310,6 +317,7 @@ def main():
     for i in range(args.steps):
         res, _ = one_step()
         per_step.append(res.seconds_to_ready)
+        del res
     sync()
     elapsed = time.perf_counter() - t0
 

@@ -1,0 +1,98 @@
+"""Model loading: scatter a landed safetensors blob into existing
+tensors (SURVEY.md §2.3 K4's real job).
+
+Zero-copy views (safetensors.torch_views) cover the "give me tensors"
+case; this module covers loading into a model whose parameters already
+live at allocator-chosen addresses — one scatter_ranges launch moves
+every matching tensor's bytes HBM->HBM (3.2 TB/s class), with an
+f32->bf16 cast path when the checkpoint dtype is wider than the model's.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import struct
+
+from ..gpu import hip
+from .formats.safetensors import SafetensorsHeader
+
+
+def load_into(blob, header: SafetensorsHeader, targets: dict,
+              strict: bool = True, stream=None) -> list[str]:
+    """Scatter tensors from `blob` into `targets` (name -> torch tensor
+    on the same device).  Returns the names loaded.
+
+    Same-dtype tensors batch into one scatter_ranges launch; f32->bf16
+    casts launch per tensor.  Raises on shape/dtype mismatches (strict)
+    or missing tensors (strict).
+    """
+    import torch
+
+    h = hip()
+    own = stream is None
+    s = h.Stream(0) if own else stream
+    handle = s.handle
+
+    desc = bytearray()
+    casts = []
+    loaded = []
+    names = {t.name: t for t in header.tensors}
+    for name, dst in targets.items():
+        info = names.get(name)
+        if info is None:
+            if strict:
+                raise KeyError(f"tensor {name!r} not in checkpoint")
+            continue
+        if tuple(dst.shape) != info.shape:
+            raise ValueError(
+                f"{name}: shape {tuple(dst.shape)} != {info.shape}")
+        if not dst.is_contiguous():
+            raise ValueError(f"{name}: target must be contiguous")
+        src_off = header.data_offset + info.begin
+        if dst.dtype == getattr(torch, info.torch_dtype):
+            desc += struct.pack("<4Q", src_off, dst.data_ptr(),
+                                info.nbytes, 0)
+        elif info.torch_dtype == "float32" and dst.dtype == torch.bfloat16:
+            casts.append((src_off, dst.data_ptr(), info.nbytes // 4))
+        else:
+            raise ValueError(
+                f"{name}: cannot load {info.torch_dtype} into {dst.dtype}")
+        loaded.append(name)
+    if strict:
+        missing = set(names) - set(targets)
+        if missing:
+            raise KeyError(f"model is missing tensors: {sorted(missing)[:5]}"
+                           f"{'...' if len(missing) > 5 else ''}")
+
+    if desc:
+        dbuf = h.DeviceBuffer(len(desc))
+        carr = (ctypes.c_char * len(desc)).from_buffer(desc)
+        h.h2d_async(dbuf.ptr, ctypes.addressof(carr), len(desc), handle)
+        h.scatter_ranges(blob.buffer.ptr, dbuf.ptr, len(desc) // 32,
+                         handle)
+    for src_off, dst_ptr, n in casts:
+        h.cast_f32_to_bf16(blob.buffer.ptr + src_off, dst_ptr, n, handle)
+    # synchronous by contract: desc buffer and host staging must outlive
+    # the launches
+    s.sync()
+    return loaded
+
+
+def load_model_from_pull(result, model, strict: bool = False) -> int:
+    """Load a pull result's safetensors shards into a torch module's
+    named parameters/buffers.  Returns tensors loaded."""
+    from .formats import safetensors as st
+
+    targets = dict(model.named_parameters())
+    targets.update(dict(model.named_buffers()))
+    n = 0
+    for f in result.files:
+        if not f.name.endswith(".safetensors"):
+            continue
+        hdr = st.parse_header(f.blob.head)
+        present = {t.name for t in hdr.tensors}
+        n += len(load_into(f.blob, hdr,
+                           {k: v for k, v in targets.items()
+                            if k in present},
+                           strict=strict))
+    return n
