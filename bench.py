@@ -89,6 +89,11 @@ def main():
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--shards", type=int, default=None,
                     help="override safetensors shard count")
+    ap.add_argument("--virtual", action="store_true",
+                    help="serve blobs from memory (no disk) — for models "
+                         "bigger than the box's disk, e.g. llama3-70b; "
+                         "payloads are patterned bytes, so tensor-view "
+                         "materialization is skipped")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -120,22 +125,35 @@ def main():
         f"demodel_bench_{args.model}"
         + (f"_s{args.shards}" if args.shards else ""))
     t = time.time()
-    # rank 0 generates the shared synthetic files; others wait
-    if rank == 0:
-        files = make_model_files(args.model, data_dir, args.shards)
-    if dist:
-        dist.barrier()
-    if rank != 0:
-        files = make_model_files(args.model, data_dir, args.shards)
-    log(f"model files ready in {time.time() - t:.1f}s "
-        f"({sum(os.path.getsize(p) for p in files.values()) / 1e9:.2f} GB)")
-
     lt = LoopThread()
-    origin = FakeOrigin(data_dir, redirect_blobs=True)
-    origin.add_hf_repo("bench/model", files)
+    if args.virtual:
+        from demodel_amd.testing import synth
+
+        geom_name, n_shards = GEOMS[args.model]
+        assert geom_name, f"--virtual unsupported for {args.model}"
+        geom = getattr(synth, geom_name)
+        sizes = synth.shard_sizes(geom, args.shards or n_shards)
+        files = {n: None for n in sizes}
+        os.makedirs(data_dir, exist_ok=True)
+        origin = FakeOrigin(data_dir, redirect_blobs=True)
+        origin.add_hf_repo_virtual("bench/model", sizes)
+        total_bytes = sum(sizes.values())
+    else:
+        # rank 0 generates the shared synthetic files; others wait
+        if rank == 0:
+            files = make_model_files(args.model, data_dir, args.shards)
+        if dist:
+            dist.barrier()
+        if rank != 0:
+            files = make_model_files(args.model, data_dir, args.shards)
+        origin = FakeOrigin(data_dir, redirect_blobs=True)
+        origin.add_hf_repo("bench/model", files)
+        total_bytes = sum(os.path.getsize(p) for p in files.values())
+    log(f"model files ready in {time.time() - t:.1f}s "
+        f"({total_bytes / 1e9:.2f} GB{' virtual' if args.virtual else ''})")
+
     port = lt.call(origin.start())
     endpoint = f"http://127.0.0.1:{port}"
-    total_bytes = sum(os.path.getsize(p) for p in files.values())
 
     landers = LanderPool(local_rank if have_gpu else 0,
                          slab_bytes=args.slab_mib << 20)
@@ -158,8 +176,9 @@ def main():
             for f in res.files:
                 if f.blob.digest_blob:
                     digest_map[f.name] = f.blob.digest_blob
-        # model-ready: materialize the tensor views
-        n_t = len(res.tensors())
+        # model-ready: materialize the tensor views (virtual payloads are
+        # patterned bytes, not parseable safetensors)
+        n_t = 0 if args.virtual else len(res.tensors())
         if have_gpu:
             torch.cuda.synchronize()
         return res, n_t
@@ -253,7 +272,8 @@ def main():
         res.total_bytes = decompressed  # the bytes that LAND in the ring
         return res, len(byname)
 
-    file_sizes = [(n, os.path.getsize(p)) for n, p in sorted(files.items())]
+    file_sizes = [(n, os.path.getsize(p) if p else sizes[n])
+                  for n, p in sorted(files.items())]
 
     def fanout_step(record_digests=False):
         """shard/broadcast modes: every rank ends with the full model."""

@@ -61,6 +61,27 @@ class FakeOrigin:
         self.hf_repos[repo_id] = {"sha": commit, "files": dict(files)}
         return commit
 
+    def add_hf_repo_virtual(self, repo_id: str, sizes: dict[str, int],
+                            commit: str | None = None) -> str:
+        """Register a repo whose blobs are served from memory (tiled
+        deterministic pattern) — no disk, no page cache; for benchmarks
+        bigger than the box's disk (e.g. the 141 GB Llama-3-70B set)."""
+        commit = commit or hashlib.sha1(repo_id.encode()).hexdigest()
+        self.hf_repos[repo_id] = {"sha": commit, "files": {},
+                                  "virtual": dict(sizes)}
+        return commit
+
+    _pattern: bytes | None = None
+
+    @classmethod
+    def _virtual_pattern(cls) -> bytes:
+        if cls._pattern is None:
+            import numpy as np
+
+            cls._pattern = np.random.default_rng(7).integers(
+                0, 256, size=64 << 20, dtype=np.uint8).tobytes()
+        return cls._pattern
+
     def add_ollama_model(self, name: str, tag: str,
                          layers: list[tuple[str, str]]) -> dict:
         """layers: list of (media_type, blob_path). Returns the manifest."""
@@ -166,10 +187,8 @@ class FakeOrigin:
             if repo is None:
                 return await self._error(writer, 404)
             items = []
-            for name in sorted(repo["files"]):
-                fpath = repo["files"][name]
-                size = os.path.getsize(fpath)
-                etag = self._file_etag(fpath)
+            for name in sorted(self._repo_names(repo)):
+                size, etag, _ = self._file_info(repo, name)
                 items.append({
                     "type": "file", "path": name, "size": size,
                     "oid": etag[:40],
@@ -199,7 +218,7 @@ class FakeOrigin:
                 "private": False, "gated": False, "disabled": False,
                 "downloads": 0, "likes": 0, "tags": [],
                 "siblings": [{"rfilename": name}
-                             for name in sorted(repo["files"])],
+                             for name in sorted(self._repo_names(repo))],
             }
             body = json.dumps(info).encode()
             return await self._reply(
@@ -213,13 +232,12 @@ class FakeOrigin:
             rev = parts[i + 1] if i + 1 < len(parts) else "main"
             fname = "/".join(parts[i + 2:])
             repo = self.hf_repos.get(repo_id)
-            if repo is None or fname not in repo["files"]:
+            if repo is None or fname not in self._repo_names(repo):
                 return await self._error(writer, 404)
-            fpath = repo["files"][fname]
-            etag = self._file_etag(fpath)
+            size, etag, fpath = self._file_info(repo, fname)
             extra = [("X-Repo-Commit", repo["sha"]),
                      ("X-Linked-Etag", f'"{etag}"'),
-                     ("X-Linked-Size", str(os.path.getsize(fpath)))]
+                     ("X-Linked-Size", str(size))]
             if self.redirect_blobs:
                 loc = f"/cdn/{repo_id}/{repo['sha']}/{fname}"
                 return await self._reply(
@@ -227,6 +245,9 @@ class FakeOrigin:
                     extra + [("Location", loc),
                              ("Content-Type", "text/plain")],
                     b"redirect")
+            if fpath is None:
+                return await self._serve_virtual(writer, head, size,
+                                                 etag, extra=extra)
             return await self._serve_file(writer, head, fpath, etag=etag,
                                           extra=extra)
 
@@ -235,13 +256,59 @@ class FakeOrigin:
             repo_id = "/".join(parts[1:3])
             fname = "/".join(parts[4:])
             repo = self.hf_repos.get(repo_id)
-            if repo is None or fname not in repo["files"]:
+            if repo is None or fname not in self._repo_names(repo):
                 return await self._error(writer, 404)
-            fpath = repo["files"][fname]
-            return await self._serve_file(writer, head, fpath,
-                                          etag=self._file_etag(fpath))
+            size, etag, fpath = self._file_info(repo, fname)
+            if fpath is None:
+                return await self._serve_virtual(writer, head, size, etag)
+            return await self._serve_file(writer, head, fpath, etag=etag)
 
         return await self._error(writer, 404)
+
+    def _repo_names(self, repo: dict) -> set[str]:
+        return set(repo["files"]) | set(repo.get("virtual", {}))
+
+    def _file_info(self, repo: dict, name: str):
+        """-> (size, etag, path_or_None)."""
+        if name in repo["files"]:
+            p = repo["files"][name]
+            return os.path.getsize(p), self._file_etag(p), p
+        size = repo["virtual"][name]
+        # deliberately 40-hex (sha1-shaped): content is synthetic, so no
+        # sha256 etag the engine might try to verify against
+        return size, hashlib.sha1(
+            f"virtual:{name}:{size}".encode()).hexdigest(), None
+
+    async def _serve_virtual(self, writer, req: RequestHead, size: int,
+                             etag: str, extra=None):
+        from ..utils.netio import send_pattern_threaded
+
+        start, end = 0, size - 1
+        status, reason = 200, "OK"
+        rng = req.get("range")
+        if rng and rng.startswith("bytes="):
+            spec = rng[len("bytes="):].split(",")[0]
+            s, _, e = spec.partition("-")
+            if s:
+                start = int(s)
+                end = int(e) if e else size - 1
+            else:
+                start = max(0, size - int(e))
+            status, reason = 206, "Partial Content"
+        length = end - start + 1
+        headers = [("Content-Type", "application/octet-stream"),
+                   ("Content-Length", str(length)),
+                   ("Accept-Ranges", "bytes"),
+                   ("ETag", f'"{etag}"')] + (extra or [])
+        if status == 206:
+            headers.append(("Content-Range", f"bytes {start}-{end}/{size}"))
+        writer.write(http1.serialize_response(
+            ResponseHead("HTTP/1.1", status, reason, headers)))
+        await writer.drain()
+        if req.method == "HEAD":
+            return
+        await send_pattern_threaded(writer, self._virtual_pattern(),
+                                    start, length)
 
     _etag_cache: dict[tuple[str, float], str] = {}
 

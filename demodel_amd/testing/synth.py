@@ -51,14 +51,9 @@ def llama_tensor_table(geom: dict) -> list[tuple[str, tuple[int, ...]]]:
     return out
 
 
-def write_shards(out_dir: str, geom: dict, n_shards: int,
-                 seed: int = 0, reuse: bool = True) -> dict[str, str]:
-    """Write llama-shaped bf16 safetensors shards; returns
-    {rfilename: path}.  Payload is a repeated random block (content is
-    irrelevant to the pipeline; generation must not dominate setup)."""
-    os.makedirs(out_dir, exist_ok=True)
+def plan_shards(geom: dict, n_shards: int) -> list[dict]:
+    """Byte-balanced assignment of llama tensors to shard specs."""
     table = llama_tensor_table(geom)
-    # round-robin tensors into shards, balancing bytes
     shard_specs: list[dict] = [dict() for _ in range(n_shards)]
     shard_bytes = [0] * n_shards
     for name, shape in table:
@@ -66,6 +61,27 @@ def write_shards(out_dir: str, geom: dict, n_shards: int,
         i = shard_bytes.index(min(shard_bytes))
         shard_specs[i][name] = ("BF16", shape, nbytes)
         shard_bytes[i] += nbytes
+    return shard_specs
+
+
+def shard_sizes(geom: dict, n_shards: int) -> dict[str, int]:
+    """File sizes write_shards would produce, without writing anything
+    (virtual-origin benchmarks bigger than the disk)."""
+    out = {}
+    for i, spec in enumerate(plan_shards(geom, n_shards)):
+        head, data_bytes = st.build_header(spec)
+        fname = f"model-{i + 1:05d}-of-{n_shards:05d}.safetensors"
+        out[fname] = len(head) + data_bytes
+    return out
+
+
+def write_shards(out_dir: str, geom: dict, n_shards: int,
+                 seed: int = 0, reuse: bool = True) -> dict[str, str]:
+    """Write llama-shaped bf16 safetensors shards; returns
+    {rfilename: path}.  Payload is a repeated random block (content is
+    irrelevant to the pipeline; generation must not dominate setup)."""
+    os.makedirs(out_dir, exist_ok=True)
+    shard_specs = plan_shards(geom, n_shards)
 
     rng = np.random.default_rng(seed)
     rand_block = rng.integers(0, 256, size=64 << 20, dtype=np.uint8)

@@ -26,6 +26,48 @@ def _pool() -> cf.ThreadPoolExecutor:
     return _POOL
 
 
+async def send_pattern_threaded(writer: asyncio.StreamWriter,
+                                pattern: bytes, start: int,
+                                length: int) -> None:
+    """Send `length` bytes of an infinitely-tiled `pattern` beginning at
+    absolute offset `start`, straight from memory on a worker thread
+    (virtual blobs: benchmarks larger than the disk)."""
+    sock = writer.transport.get_extra_info("socket")
+    tls = writer.transport.get_extra_info("sslcontext")
+    if sock is None or tls is not None:
+        raise NotImplementedError
+    await writer.drain()
+    writer.transport.pause_reading()
+    mv = memoryview(pattern)
+    plen = len(pattern)
+
+    def run():
+        sent_total = 0
+        while sent_total < length:
+            phase = (start + sent_total) % plen
+            piece = mv[phase:min(plen, phase + (length - sent_total))]
+            off = 0
+            while off < len(piece):
+                try:
+                    n = sock.send(piece[off:])
+                except BlockingIOError:
+                    sel.select([], [sock.fileno()], [], 10)
+                    continue
+                if n == 0:
+                    raise ConnectionResetError("peer went away")
+                off += n
+            sent_total += len(piece)
+
+    loop = asyncio.get_running_loop()
+    try:
+        await loop.run_in_executor(_pool(), run)
+    finally:
+        try:
+            writer.transport.resume_reading()
+        except Exception:
+            pass
+
+
 async def sendfile_threaded(writer: asyncio.StreamWriter, f,
                             start: int, length: int) -> None:
     """Send [start, start+length) of file f on writer's socket.
