@@ -1,0 +1,75 @@
+"""bench.py contract tests: the driver launches it directly (N=1) and
+under torch.distributed.run (N>1, one rank per GPU) — keep both paths
+green on CPU so round-end GPU runs don't hit launcher bugs."""
+
+import json
+import os
+import socket
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _check_line(line: str, world: int):
+    out = json.loads(line)
+    assert out["metric"] == "pull_gbps_into_hbm"
+    assert out["n_gpus"] == world
+    assert out["unit"] == "GB/s"
+    assert out["higher_is_better"] is True
+    assert out["value"] > 0
+    assert out["data"] == "synthetic"
+    assert "ms_per_step" in out and out["ms_per_step"] > 0
+    assert "config" in out and "model" in out["config"]
+    return out
+
+
+def test_bench_single_process(tmp_path):
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--model", "tiny", "--steps", "2",
+         "--warmup", "1", "--data-dir", str(tmp_path / "d")],
+        cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+    assert len(line) == 1, r.stdout  # EXACTLY one JSON line
+    _check_line(line[0], world=1)
+
+
+def test_bench_torchrun_world2(tmp_path):
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(_free_port()), "bench.py", "--gpus", "2",
+         "--model", "tiny", "--steps", "2", "--warmup", "1",
+         "--data-dir", str(tmp_path / "d")],
+        cwd=REPO, capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+    assert len(line) == 1, r.stdout
+    _check_line(line[0], world=2)
+
+
+def test_bench_torchrun_shard_mode(tmp_path):
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(_free_port()), "bench.py", "--gpus", "2",
+         "--model", "tiny", "--mode", "shard", "--steps", "1",
+         "--warmup", "1", "--data-dir", str(tmp_path / "d")],
+        cwd=REPO, capture_output=True, text=True, timeout=300, env=env)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+    out = _check_line(line[0], world=2)
+    assert out["scaling"] == "strong"
