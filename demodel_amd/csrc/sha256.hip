@@ -206,15 +206,21 @@ sha256_chain_update_kernel(uint32_t* __restrict__ state,
       }
     }
     __syncthreads();
-    // phase B: all lanes redundantly run the rounds (LDS broadcast reads)
+    // phase B: all lanes redundantly run the rounds.  Stage the block's
+    // whole schedule into registers first: 64 independent ds_reads issue
+    // back-to-back (one latency total) instead of one blocking read per
+    // round.
     for (int b = 0; b < group; ++b) {
+      uint32_t kwreg[64];
+#pragma unroll
+      for (int t = 0; t < 64; ++t) kwreg[t] = kw[b][t];
       uint32_t a = s[0], bb = s[1], c = s[2], d = s[3];
       uint32_t e = s[4], f = s[5], g = s[6], h = s[7];
 #pragma unroll
       for (int t = 0; t < 64; ++t) {
         uint32_t S1 = rotr(e, 6) ^ rotr(e, 11) ^ rotr(e, 25);
         uint32_t ch = (e & f) ^ (~e & g);
-        uint32_t t1 = h + S1 + ch + kw[b][t];
+        uint32_t t1 = h + S1 + ch + kwreg[t];
         uint32_t S0 = rotr(a, 2) ^ rotr(a, 13) ^ rotr(a, 22);
         uint32_t maj = (a & bb) ^ (a & c) ^ (bb & c);
         uint32_t t2 = S0 + maj;

@@ -1,0 +1,230 @@
+"""Parquet page extraction for GPU decompression (BASELINE.json config 5:
+c4-en parquet streaming).
+
+A parquet file's column chunks are sequences of pages, each preceded by a
+Thrift compact-protocol PageHeader giving the compressed/uncompressed
+sizes.  The heavy bytes are the page payloads; the headers are tiny.  So:
+CPU walks the headers (this module — a minimal Thrift compact reader, no
+parquet library in the data path), and the GPU decompresses every page
+payload in one wave-parallel zstd/deflate launch into the HBM ring.
+
+Scope: v1 data pages + dictionary pages (what pyarrow writes by
+default); compressed codecs ZSTD and GZIP-less SNAPPY are NOT all
+supported — ZSTD pages go to the GPU, UNCOMPRESSED pass through, anything
+else fails loudly.  Data page v2 is rejected loudly (levels would need
+splitting).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+# Thrift compact protocol type ids
+_CT_STOP = 0
+_CT_TRUE = 1
+_CT_FALSE = 2
+_CT_BYTE = 3
+_CT_I16 = 4
+_CT_I32 = 5
+_CT_I64 = 6
+_CT_DOUBLE = 7
+_CT_BINARY = 8
+_CT_LIST = 9
+_CT_SET = 10
+_CT_MAP = 11
+_CT_STRUCT = 12
+
+
+class _Compact:
+    def __init__(self, data, pos=0):
+        self.d = data
+        self.p = pos
+
+    def byte(self):
+        b = self.d[self.p]
+        self.p += 1
+        return b
+
+    def varint(self):
+        out = 0
+        shift = 0
+        while True:
+            b = self.byte()
+            out |= (b & 0x7F) << shift
+            if not b & 0x80:
+                return out
+            shift += 7
+
+    def zigzag(self):
+        v = self.varint()
+        return (v >> 1) ^ -(v & 1)
+
+    def skip(self, ctype):
+        if ctype in (_CT_TRUE, _CT_FALSE):
+            return
+        if ctype == _CT_BYTE:
+            self.byte()
+        elif ctype in (_CT_I16, _CT_I32, _CT_I64):
+            self.varint()
+        elif ctype == _CT_DOUBLE:
+            self.p += 8
+        elif ctype == _CT_BINARY:
+            n = self.varint()  # NB: varint() moves p; don't fold into +=
+            self.p += n
+        elif ctype in (_CT_LIST, _CT_SET):
+            b = self.byte()
+            n = b >> 4
+            et = b & 0xF
+            if n == 15:
+                n = self.varint()
+            for _ in range(n):
+                self.skip(et)
+        elif ctype == _CT_MAP:
+            n = self.varint()
+            if n:
+                kv = self.byte()
+                for _ in range(n):
+                    self.skip(kv >> 4)
+                    self.skip(kv & 0xF)
+        elif ctype == _CT_STRUCT:
+            last = 0
+            while True:
+                b = self.byte()
+                if b == _CT_STOP:
+                    return
+                delta = b >> 4
+                ft = b & 0xF
+                last = last + delta if delta else self.zigzag()
+                self.skip(ft)
+        else:
+            raise ValueError(f"unknown thrift compact type {ctype}")
+
+    def read_struct_fields(self):
+        """Yield (field_id, ctype) and leave .p at each value start;
+        caller must consume or .skip(ctype)."""
+        last = 0
+        while True:
+            b = self.byte()
+            if b == _CT_STOP:
+                return
+            delta = b >> 4
+            ft = b & 0xF
+            if delta:
+                last = last + delta
+            else:
+                last = self.zigzag()
+            yield last, ft
+
+
+@dataclass
+class PageInfo:
+    page_type: int          # 0 data, 2 dictionary, 3 data_v2
+    comp_offset: int        # absolute offset of the compressed payload
+    comp_size: int
+    uncomp_size: int
+
+
+def parse_page_header(data, pos: int) -> tuple[PageInfo, int]:
+    """Parse one PageHeader at `pos`; returns (info, payload_offset)."""
+    c = _Compact(data, pos)
+    page_type = None
+    comp = uncomp = None
+    for fid, ft in c.read_struct_fields():
+        if fid == 1 and ft in (_CT_I32, _CT_BYTE, _CT_I16):
+            page_type = c.zigzag() if ft != _CT_BYTE else c.byte()
+        elif fid == 2:
+            uncomp = c.zigzag()
+        elif fid == 3:
+            comp = c.zigzag()
+        else:
+            c.skip(ft)
+    if page_type is None or comp is None or uncomp is None:
+        raise ValueError("malformed parquet PageHeader")
+    if page_type == 3:
+        raise ValueError("data page v2 not supported (levels split)")
+    return PageInfo(page_type, c.p, comp, uncomp), c.p
+
+
+CODEC_UNCOMPRESSED = 0
+CODEC_SNAPPY = 1
+CODEC_GZIP = 2
+CODEC_ZSTD = 6
+
+
+def column_chunk_pages(data, start: int, total_compressed: int
+                       ) -> list[PageInfo]:
+    """Walk all pages of one column chunk (headers live in `data`, a
+    buffer covering [start, start+total_compressed))."""
+    pages = []
+    pos = start
+    end = start + total_compressed
+    while pos < end:
+        info, payload = parse_page_header(data, pos)
+        pages.append(info)
+        pos = payload + info.comp_size
+    if pos != end:
+        raise ValueError(f"column chunk walk overran: {pos} != {end}")
+    return pages
+
+
+def file_pages(path: str):
+    """All (column-chunk codec, PageInfo) of a parquet file, using
+    pyarrow only for the FOOTER metadata (offsets/codecs), never for page
+    payloads.  Returns (file_bytes, [(codec, PageInfo)])."""
+    import pyarrow.parquet as pq
+
+    meta = pq.ParquetFile(path).metadata
+    raw = open(path, "rb").read()
+    out = []
+    codec_names = {"UNCOMPRESSED": CODEC_UNCOMPRESSED,
+                   "SNAPPY": CODEC_SNAPPY, "GZIP": CODEC_GZIP,
+                   "ZSTD": CODEC_ZSTD}
+    for rg in range(meta.num_row_groups):
+        for col in range(meta.num_columns):
+            cc = meta.row_group(rg).column(col)
+            codec = codec_names.get(cc.compression.upper())
+            if codec is None:
+                raise ValueError(f"unsupported codec {cc.compression}")
+            start = cc.dictionary_page_offset
+            if start is None or start <= 0:
+                start = cc.data_page_offset
+            for info in column_chunk_pages(raw, start,
+                                           cc.total_compressed_size):
+                out.append((codec, info))
+    return raw, out
+
+
+def decompress_pages_gpu(blob, pages, ring=None):
+    """GPU-decompress a landed parquet blob's ZSTD pages into an HBM
+    ring.  pages: [(codec, PageInfo)] with offsets relative to the blob.
+    Returns (ring_buffer, [(out_offset, size)]) covering every page
+    (UNCOMPRESSED pages are device-copied)."""
+    from ...gpu import hip
+    from .compress import zstd_gpu
+
+    h = hip()
+    total = sum(p.uncomp_size for _, p in pages)
+    ring = ring or h.DeviceBuffer(max(total, 1))
+    s = h.Stream(0)
+    frames = []
+    spans = []
+    off = 0
+    for codec, p in pages:
+        if codec == CODEC_ZSTD:
+            frames.append((blob.buffer.ptr + p.comp_offset, p.comp_size,
+                           ring.ptr + off, p.uncomp_size))
+        elif codec == CODEC_UNCOMPRESSED:
+            h.d2d_async(ring.ptr + off, blob.buffer.ptr + p.comp_offset,
+                        p.comp_size, s.handle)
+        else:
+            raise ValueError(f"GPU path supports ZSTD/UNCOMPRESSED, "
+                             f"got codec {codec}")
+        spans.append((off, p.uncomp_size))
+        off += p.uncomp_size
+    if frames:
+        results = zstd_gpu(frames, stream_handle=s.handle)
+        bad = [(i, r) for i, r in enumerate(results) if not r.ok]
+        if bad:
+            raise IOError(f"GPU page decompress failed: {bad[:3]}")
+    s.sync()
+    return ring, spans
