@@ -28,7 +28,8 @@ GEOMS = {
     "tiny": (None, 2),
     "gguf-8b": ("LLAMA3_8B", 0),     # Ollama q4_K pull + GPU dequant
     "gguf-tiny": (None, 0),
-    "dataset": (None, 0),            # zstd streaming into HBM ring
+    "dataset": (None, 0),            # zstd frame shards into HBM ring
+    "parquet": (None, 0),            # real c4-like parquet, ZSTD pages
 }
 
 
@@ -57,6 +58,8 @@ def make_model_files(model: str, data_dir: str, n_shards_override=None):
         return synth.write_shards(data_dir, TINY_GEOM, 2)
     if model == "dataset":
         return synth.write_dataset_shards(data_dir)
+    if model == "parquet":
+        return synth.write_parquet_shards(data_dir)
     if model in ("gguf-8b", "gguf-tiny"):
         geom = synth.LLAMA3_8B if model == "gguf-8b" else TINY_GEOM
         path = os.path.join(data_dir, "model.gguf")
@@ -221,6 +224,53 @@ def main():
         res.meta["dequant_bf16_bytes"] = out_bytes
         return res, len(gg.tensors)
 
+    def parquet_step(record_digests=False):
+        """Real parquet streaming (config 5): land shards, read footers +
+        page headers back from HBM, GPU-decompress every ZSTD page into
+        the ring."""
+        from demodel_amd.engine.formats import parquet as pqf
+        from demodel_amd.engine.formats.compress import zstd_gpu
+        from demodel_amd.gpu import have_gpu as _hg, hip
+
+        res = pull_mod.pull_hf(
+            "bench/model", endpoint=endpoint, workers=args.workers,
+            verify=args.verify, landers=landers,
+            digest_map=digest_map or None)
+        if record_digests:
+            for f in res.files:
+                if f.blob.digest_blob:
+                    digest_map[f.name] = f.blob.digest_blob
+        decompressed = 0
+        if _hg():
+            h = hip()
+            frames = []
+            plan = []
+            total_out = 0
+            for f in res.files:
+                pages = pqf.blob_pages(f.blob)
+                for codec, p in pages:
+                    assert codec == pqf.CODEC_ZSTD, codec
+                    plan.append((f.blob, p))
+                    total_out += p.uncomp_size
+            ring = h.DeviceBuffer(total_out)
+            off = 0
+            for blob, p in plan:
+                frames.append((blob.buffer.ptr + p.comp_offset,
+                               p.comp_size, ring.ptr + off, p.uncomp_size))
+                off += p.uncomp_size
+            results = zstd_gpu(frames)
+            for i, r in enumerate(results):
+                assert r.ok and r.written == plan[i][1].uncomp_size, (i, r)
+            decompressed = total_out
+            del ring
+        else:
+            decompressed = res.total_bytes
+        if have_gpu:
+            torch.cuda.synchronize()
+        res.meta["decompressed_bytes"] = decompressed
+        res.total_bytes = decompressed
+        return res, len(res.files)
+
     def dataset_step(record_digests=False):
         """Dataset streaming (config 5): pull zstd shards, decompress all
         frames wave-parallel into an HBM ring; counts DECOMPRESSED bytes
@@ -316,6 +366,10 @@ def main():
             return gguf_step(record_digests)
         if args.model == "dataset":
             res, n = dataset_step(record_digests)
+            step_bytes = res.total_bytes
+            return res, n
+        if args.model == "parquet":
+            res, n = parquet_step(record_digests)
             step_bytes = res.total_bytes
             return res, n
         if args.mode == "dp":

@@ -167,6 +167,133 @@ def column_chunk_pages(data, start: int, total_compressed: int
     return pages
 
 
+@dataclass
+class ChunkMeta:
+    codec: int
+    start: int               # first page offset (dictionary if present)
+    total_compressed: int
+
+
+def parse_footer(footer: bytes) -> list[ChunkMeta]:
+    """Parse a parquet FileMetaData (Thrift compact) for column-chunk
+    locations and codecs — self-contained, no parquet library.
+
+    `footer` is the thrift blob (file[-8-len:-8])."""
+    chunks: list[ChunkMeta] = []
+    c = _Compact(footer, 0)
+
+    def parse_column_meta(cc: _Compact):
+        codec = start = total = dict_off = data_off = None
+        for fid, ft in cc.read_struct_fields():
+            if fid == 4:
+                codec = cc.zigzag()
+            elif fid == 7:
+                total = cc.zigzag()
+            elif fid == 9:
+                data_off = cc.zigzag()
+            elif fid == 11:
+                dict_off = cc.zigzag()
+            else:
+                cc.skip(ft)
+        start = dict_off if dict_off not in (None, 0) else data_off
+        if codec is None or total is None or start is None:
+            raise ValueError("incomplete ColumnMetaData")
+        return ChunkMeta(codec, start, total)
+
+    def parse_column_chunk(cc: _Compact):
+        meta = None
+        for fid, ft in cc.read_struct_fields():
+            if fid == 3 and ft == _CT_STRUCT:
+                meta = parse_column_meta(cc)
+            else:
+                cc.skip(ft)
+        if meta is None:
+            raise ValueError("ColumnChunk without meta_data")
+        return meta
+
+    def parse_row_group(cc: _Compact):
+        out = []
+        for fid, ft in cc.read_struct_fields():
+            if fid == 1 and ft == _CT_LIST:
+                b = cc.byte()
+                n = b >> 4
+                et = b & 0xF
+                if n == 15:
+                    n = cc.varint()
+                assert et == _CT_STRUCT
+                for _ in range(n):
+                    out.append(parse_column_chunk(cc))
+            else:
+                cc.skip(ft)
+        return out
+
+    for fid, ft in c.read_struct_fields():
+        if fid == 4 and ft == _CT_LIST:
+            b = c.byte()
+            n = b >> 4
+            et = b & 0xF
+            if n == 15:
+                n = c.varint()
+            assert et == _CT_STRUCT
+            for _ in range(n):
+                chunks.extend(parse_row_group(c))
+        else:
+            c.skip(ft)
+    return chunks
+
+
+def footer_span(tail8: bytes) -> int:
+    """Footer thrift length from a file's last 8 bytes
+    (u32 LE length + b'PAR1')."""
+    import struct as _struct
+
+    if tail8[4:] != b"PAR1":
+        raise ValueError("not a parquet file (missing PAR1 trailer)")
+    return _struct.unpack("<I", tail8[:4])[0]
+
+
+def blob_pages(blob) -> list[tuple[int, "PageInfo"]]:
+    """Page locations of a parquet blob LANDED IN HBM: footer + page
+    headers are read back via small staged D2H copies (payload bytes
+    never leave the device).  Returns [(codec, PageInfo)]."""
+    import ctypes
+
+    from ...gpu import hip
+
+    h = hip()
+    s = h.Stream(0)
+
+    def d2h(off: int, n: int) -> bytes:
+        n = min(n, blob.nbytes - off)
+        out = bytearray(n)
+        addr = ctypes.addressof((ctypes.c_char * n).from_buffer(out))
+        h.d2h_async(addr, blob.buffer.ptr + off, n, s.handle)
+        s.sync()
+        return bytes(out)
+
+    flen = footer_span(d2h(blob.nbytes - 8, 8))
+    footer = d2h(blob.nbytes - 8 - flen, flen)
+    out = []
+    for cm in parse_footer(footer):
+        # walk headers with a sliding 64 KiB host window
+        pos = cm.start
+        end = cm.start + cm.total_compressed
+        win_off = -1
+        win = b""
+        while pos < end:
+            if win_off < 0 or pos + 512 > win_off + len(win):
+                win_off = pos
+                win = d2h(pos, 64 << 10)
+            info, payload = parse_page_header(win, pos - win_off)
+            info = PageInfo(info.page_type, payload + win_off,
+                            info.comp_size, info.uncomp_size)
+            out.append((cm.codec, info))
+            pos = info.comp_offset + info.comp_size
+        if pos != end:
+            raise ValueError("page walk overran column chunk")
+    return out
+
+
 def file_pages(path: str):
     """All (column-chunk codec, PageInfo) of a parquet file, using
     pyarrow only for the FOOTER metadata (offsets/codecs), never for page

@@ -164,6 +164,43 @@ def write_dataset_shards(out_dir: str, n_shards: int = 8,
     return files
 
 
+def write_parquet_shards(out_dir: str, n_shards: int = 8,
+                         rows_per_shard: int = 300_000,
+                         reuse: bool = True) -> dict[str, str]:
+    """c4-en-like parquet shards (text/url/timestamp columns, ZSTD
+    pages) — BASELINE.json config 5's dataset."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    os.makedirs(out_dir, exist_ok=True)
+    rng = np.random.default_rng(11)
+    vocab = np.array([f"w{i:04d}" for i in range(20000)])
+
+    base_n = 30_000
+    docs = [" ".join(vocab[rng.integers(0, len(vocab), size=120)])
+            for _ in range(base_n)]
+    reps = (rows_per_shard + base_n - 1) // base_n
+
+    files = {}
+    for s in range(n_shards):
+        fname = f"c4-train.{s:05d}-of-{n_shards:05d}.parquet"
+        path = os.path.join(out_dir, fname)
+        if reuse and os.path.exists(path) and os.path.getsize(path) > 0:
+            files[fname] = path
+            continue
+        n = rows_per_shard
+        text = (docs * reps)[:n]
+        table = pa.table({
+            "text": text,
+            "url": [f"https://example.com/{s}/{i}" for i in range(n)],
+            "timestamp": np.arange(n, dtype=np.int64) + s,
+        })
+        pq.write_table(table, path, compression="zstd",
+                       data_page_version="1.0")
+        files[fname] = path
+    return files
+
+
 def write_gguf_model(path: str, geom: dict, qtype: int = 12,
                      reuse: bool = True):
     """Synthetic GGUF (default q4_K) with llama-shaped 2-D tensors."""

@@ -61,6 +61,31 @@ def test_zstd_pages_decode_with_both_decoders(tmp_path):
         assert status == 0 and got == want
 
 
+def test_parse_footer_matches_pyarrow(tmp_path):
+    p = tmp_path / "f.parquet"
+    _write_parquet(str(p))
+    raw = open(str(p), "rb").read()
+    flen = pqf.footer_span(raw[-8:])
+    chunks = pqf.parse_footer(raw[-8 - flen:-8])
+    meta = pq.ParquetFile(str(p)).metadata
+    want = []
+    for rg in range(meta.num_row_groups):
+        for col in range(meta.num_columns):
+            cc = meta.row_group(rg).column(col)
+            start = cc.dictionary_page_offset or cc.data_page_offset
+            want.append((start, cc.total_compressed_size))
+    assert [(c.start, c.total_compressed) for c in chunks] == want
+    assert all(c.codec == pqf.CODEC_ZSTD for c in chunks)
+    # and the self-contained walk agrees with the pyarrow-footer walk
+    _, via_pyarrow = pqf.file_pages(str(p))
+    pages_self = []
+    for cm in chunks:
+        for info in pqf.column_chunk_pages(raw, cm.start,
+                                           cm.total_compressed):
+            pages_self.append((cm.codec, info))
+    assert pages_self == via_pyarrow
+
+
 def test_uncompressed_parquet_pages(tmp_path):
     p = tmp_path / "u.parquet"
     _write_parquet(str(p), codec="none")
