@@ -80,11 +80,13 @@ def main():
     ap.add_argument("--model", default="llama3-8b",
                     choices=list(GEOMS))
     ap.add_argument("--mode", default="dp",
-                    choices=["dp", "shard", "broadcast"],
+                    choices=["dp", "shard", "broadcast", "allgather"],
                     help="dp: independent per-rank pulls (weak scaling); "
-                         "shard: rank r pulls 1/N of the manifest, "
+                         "shard: rank r pulls 1/N of the manifest files, "
                          "file-wise RCCL broadcast reassembly (config 3); "
-                         "broadcast: rank 0 pulls, RCCL fan-out (R1)")
+                         "broadcast: rank 0 pulls, RCCL fan-out (R1); "
+                         "allgather: byte-range shards + bucketed "
+                         "all-gather overlapped with download (R2)")
     ap.add_argument("--workers", type=int, default=8)
     ap.add_argument("--verify", default="chunked",
                     choices=["chunked", "digest", "gpu-digest", "off"])
@@ -360,6 +362,68 @@ def main():
 
     step_bytes = total_bytes  # may be overridden by workloads below
 
+    def allgather_step(record_digests=False):
+        """R2: every rank pulls 1/N of the concatenated manifest bytes
+        via HTTP Range, reassembly by bucketed all-gather overlapped
+        with the remaining download."""
+        from demodel_amd.engine import fetch
+        from demodel_amd.parallel.fanout import range_sharded_allgather
+
+        names = [n for n, _ in file_sizes]
+        sizes = [sz for _, sz in file_sizes]
+        prefix = [0]
+        for sz in sizes:
+            prefix.append(prefix[-1] + sz)
+
+        class _Shim:
+            def __init__(self, ptr):
+                self.ptr = ptr
+
+        bucket = 256 << 20
+
+        def pull_range(lo, want, dest, bucket_done):
+            lander = landers.get()
+            buf = (_Shim(dest.data_ptr()) if have_gpu
+                   else memoryview(dest.numpy()))
+            done = 0
+            flushed = 0
+            while done < want:
+                g = lo + done  # global offset
+                fi = next(i for i in range(len(names))
+                          if prefix[i + 1] > g)
+                f_lo = g - prefix[fi]
+                take = min(prefix[fi + 1] - g, want - done,
+                           bucket - (done % bucket) or bucket)
+                url = (f"{endpoint}/bench/model/resolve/main/"
+                       f"{names[fi]}")
+                src = fetch.http_get(url, headers={
+                    "Range": f"bytes={f_lo}-{f_lo + take - 1}"})
+                try:
+                    assert src.status == 206, src.status
+                    lander.land_into(buf, done, src.fill, take,
+                                     file_size=want)
+                finally:
+                    src.close()
+                done += take
+                lander.sync()
+                while (flushed + 1) * bucket <= done:
+                    bucket_done(flushed)
+                    flushed += 1
+            while flushed * bucket < want:
+                bucket_done(flushed)
+                flushed += 1
+
+        def alloc(nb):
+            return torch.empty(nb, dtype=torch.uint8,
+                               device="cuda" if have_gpu else "cpu")
+
+        full = range_sharded_allgather(total_bytes, pull_range, alloc,
+                                       bucket_bytes=bucket)
+        assert full.numel() >= total_bytes
+        if have_gpu:
+            torch.cuda.synchronize()
+        return _FanoutResult(total_bytes * world), 0
+
     def one_step(record_digests=False):
         nonlocal step_bytes
         if args.model in ("gguf-8b", "gguf-tiny"):
@@ -374,6 +438,8 @@ def main():
             return res, n
         if args.mode == "dp":
             return dp_step(record_digests)
+        if args.mode == "allgather":
+            return allgather_step(record_digests)
         res = fanout_step(record_digests)
         return _FanoutResult(total_bytes * world), len(res)
 

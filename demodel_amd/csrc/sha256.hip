@@ -162,7 +162,7 @@ __global__ void sha256_chain_init_kernel(uint32_t* state) {
   if (threadIdx.x == 0 && blockIdx.x == 0) sha256_init_state(state);
 }
 
-__global__ void __launch_bounds__(64)
+__global__ void __launch_bounds__(64, 1)
 sha256_chain_update_kernel(uint32_t* __restrict__ state,
                            const uint8_t* __restrict__ data,
                            uint64_t nblocks) {

@@ -66,6 +66,53 @@ def broadcast_blob(tensor_u8, src: int, bucket_bytes: int = BUCKET_BYTES,
     return []
 
 
+def range_sharded_allgather(total_bytes: int, pull_range, alloc_u8,
+                            bucket_bytes: int = BUCKET_BYTES, group=None):
+    """R2 proper: the blob's byte range splits evenly across ranks, rank
+    r pulls [r*shard, (r+1)*shard) with an HTTP Range request, and equal
+    shards reassemble with bucketed all-gather — each bucket's collective
+    enqueues as soon as that bucket has landed locally, so xGMI traffic
+    overlaps the remaining download (SURVEY.md §7 hard part (d)).
+
+    pull_range(offset, nbytes, dest_u8, bucket_done): stream the global
+        byte range [offset, offset+nbytes) into dest_u8 (this rank's
+        shard view), calling bucket_done(i) once bucket i's bytes are
+        visible on-device.
+    alloc_u8(n): device tensor allocator.
+
+    Returns the assembled tensor (padded to world*shard; caller slices
+    [:total_bytes])."""
+    import torch.distributed as dist
+
+    rank = dist.get_rank(group)
+    world = dist.get_world_size(group)
+    shard = (total_bytes + world - 1) // world
+    full = alloc_u8(shard * world)
+    mine = full.narrow(0, rank * shard, shard)
+
+    n_buckets = (shard + bucket_bytes - 1) // bucket_bytes
+    works = []
+
+    def gather_bucket(b: int):
+        off = b * bucket_bytes
+        n = min(bucket_bytes, shard - off)
+        outs = [full.narrow(0, r * shard + off, n) for r in range(world)]
+        works.append(dist.all_gather(outs, mine.narrow(0, off, n),
+                                     group=group, async_op=True))
+
+    lo = rank * shard
+    want = max(0, min(shard, total_bytes - lo))
+    if want < shard:
+        mine.narrow(0, want, shard - want).zero_()
+    pull_range(lo, want, mine, gather_bucket)
+    # buckets the pull callback didn't flush (tail / zero-want ranks)
+    for b in range(len(works), n_buckets):
+        gather_bucket(b)
+    for w in works:
+        w.wait()
+    return full
+
+
 @dataclass
 class ShardPlan:
     # file name -> (owner_rank, nbytes); iteration order = manifest order

@@ -78,6 +78,34 @@ def _worker(rank, world, port, tmpdir):
         assert set(out) == {n for n, _ in files}
         for name, nb in files:
             assert torch.equal(out[name], content(name, nb)), name
+        # --- byte-range shards + bucketed all-gather (R2) ---
+        from demodel_amd.parallel.fanout import range_sharded_allgather
+
+        total = 1_000_000  # not divisible by bucket or world
+        blob = (torch.arange(total, dtype=torch.int64) * 7 % 253
+                ).to(torch.uint8)
+
+        def pull_range(lo, want, dest, bucket_done):
+            bucket = 100_000
+            done = 0
+            flushed = 0
+            while done < want:
+                take = min(bucket, want - done)
+                dest[done:done + take] = blob[lo + done:lo + done + take]
+                done += take
+                while (flushed + 1) * bucket <= done:
+                    bucket_done(flushed)
+                    flushed += 1
+            while flushed * bucket < want:
+                bucket_done(flushed)
+                flushed += 1
+
+        full = range_sharded_allgather(
+            total, pull_range,
+            lambda n2: torch.zeros(n2, dtype=torch.uint8),
+            bucket_bytes=100_000)
+        assert torch.equal(full[:total], blob), "allgather reassembly"
+
         # write a success marker per rank
         with open(os.path.join(tmpdir, f"ok{rank}"), "w") as f:
             f.write("ok")
