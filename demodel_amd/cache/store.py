@@ -69,7 +69,15 @@ class CachedResponse:
 
 
 class CacheWriter:
-    """Streaming cache fill; finalize() commits, abort() discards."""
+    """Streaming cache fill; finalize() commits, abort() discards.
+
+    Digesting modes (store.digest_mode):
+      * "sync"  — hash inline while writing (tests, small bodies)
+      * "async" — commit immediately, compute whole+chunk sha256 on a
+        background worker and patch the meta when done (the proxy's blob
+        path: a 10+ GB fill must not run at single-thread hashlib speed)
+      * "off"   — no digests
+    """
 
     def __init__(self, store: "CacheStore", uri: str, status: int,
                  reason: str, headers: list[tuple[str, str]]):
@@ -86,7 +94,8 @@ class CacheWriter:
         )
         self._f = os.fdopen(fd, "wb")
         self._size = 0
-        self._whole = hashlib.sha256()
+        self._sync = store.digest_mode == "sync"
+        self._whole = hashlib.sha256() if self._sync else None
         self._chunk = hashlib.sha256()
         self._chunk_fill = 0
         self._chunk_digests: list[str] = []
@@ -95,6 +104,8 @@ class CacheWriter:
     def write(self, data: bytes) -> None:
         self._f.write(data)
         self._size += len(data)
+        if not self._sync:
+            return
         self._whole.update(data)
         view = memoryview(data)
         while view:
@@ -108,7 +119,7 @@ class CacheWriter:
                 self._chunk_fill = 0
 
     def finalize(self) -> CachedResponse:
-        if self._chunk_fill:
+        if self._sync and self._chunk_fill:
             self._chunk_digests.append(self._chunk.hexdigest())
         self._f.close()
         body_path = os.path.join(self._store.root, self.key)
@@ -116,24 +127,13 @@ class CacheWriter:
         entry = CachedResponse(
             uri=self.uri, status=self.status, reason=self.reason,
             headers=self.headers, body_path=body_path, body_size=self._size,
-            sha256=self._whole.hexdigest(), chunk_bytes=self.chunk_bytes,
+            sha256=self._whole.hexdigest() if self._sync else None,
+            chunk_bytes=self.chunk_bytes,
             chunk_sha256=self._chunk_digests, created=time.time(),
         )
-        meta = {
-            "uri": entry.uri,
-            "status": entry.status,
-            "reason": entry.reason,
-            "headers": entry.headers,
-            "body_size": entry.body_size,
-            "sha256": entry.sha256,
-            "chunk_bytes": entry.chunk_bytes,
-            "chunk_sha256": entry.chunk_sha256,
-            "created": entry.created,
-        }
-        tmp_meta = body_path + ".meta.part"
-        with open(tmp_meta, "w") as f:
-            json.dump(meta, f, indent=1)
-        os.replace(tmp_meta, body_path + ".meta")
+        self._store._write_meta(entry)
+        if self._store.digest_mode == "async":
+            self._store._digest_async(entry)
         return entry
 
     def abort(self) -> None:
@@ -145,10 +145,57 @@ class CacheWriter:
 
 
 class CacheStore:
-    def __init__(self, root: str = ".cache", chunk_bytes: int = 32 << 20):
+    _digest_pool = None
+
+    def __init__(self, root: str = ".cache", chunk_bytes: int = 1 << 20,
+                 digest_mode: str = "sync"):
         self.root = root
         self.chunk_bytes = chunk_bytes
+        self.digest_mode = digest_mode
         os.makedirs(root, exist_ok=True)
+
+    def _write_meta(self, entry: CachedResponse) -> None:
+        meta = {
+            "uri": entry.uri,
+            "status": entry.status,
+            "reason": entry.reason,
+            "headers": entry.headers,
+            "body_size": entry.body_size,
+            "sha256": entry.sha256,
+            "chunk_bytes": entry.chunk_bytes,
+            "chunk_sha256": entry.chunk_sha256,
+            "created": entry.created or time.time(),
+        }
+        tmp_meta = entry.body_path + ".meta.part"
+        with open(tmp_meta, "w") as f:
+            json.dump(meta, f, indent=1)
+        os.replace(tmp_meta, entry.body_path + ".meta")
+
+    def _digest_async(self, entry: CachedResponse) -> None:
+        import concurrent.futures as cf
+
+        if CacheStore._digest_pool is None:
+            CacheStore._digest_pool = cf.ThreadPoolExecutor(
+                max_workers=2, thread_name_prefix="cache-digest")
+
+        def job():
+            try:
+                whole = hashlib.sha256()
+                chunks = []
+                with open(entry.body_path, "rb") as f:
+                    while True:
+                        data = f.read(entry.chunk_bytes)
+                        if not data:
+                            break
+                        whole.update(data)
+                        chunks.append(hashlib.sha256(data).hexdigest())
+                entry.sha256 = whole.hexdigest()
+                entry.chunk_sha256 = chunks
+                self._write_meta(entry)
+            except OSError:
+                pass  # entry purged/overwritten meanwhile
+
+        CacheStore._digest_pool.submit(job)
 
     def lookup(self, uri: str) -> CachedResponse | None:
         key = cache_key(uri)

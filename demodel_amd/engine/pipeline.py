@@ -108,7 +108,8 @@ class Lander:
     def land(self, fill, nbytes: int, verify: bool = True,
              host_chain: bool = False, gpu_chain: bool = False,
              keep_head: bool = True,
-             expected_digests: bytes | None = None) -> LandedBlob:
+             expected_digests: bytes | None = None,
+             verify_chunk: int | None = None) -> LandedBlob:
         """Land exactly `nbytes` from `fill` into a fresh HBM buffer.
 
         fill(view: memoryview) -> int: write up to len(view) bytes into the
@@ -127,16 +128,16 @@ class Lander:
         head, fill_s = self.land_into(
             buf, 0, fill, nbytes, file_size=nbytes, chain=chain,
             gpu_state=gpu_state, keep_head=keep_head)
+        vc = verify_chunk or self.verify_chunk
         blob = LandedBlob(nbytes=nbytes,
                           device=f"cuda:{self.device_index}", buffer=buf,
-                          verify_chunk=self.verify_chunk, head=bytes(head))
+                          verify_chunk=vc, head=bytes(head))
         t_fill_done = time.perf_counter()
 
         if (verify or expected_digests is not None) and nbytes > 0:
-            blob.digest_blob = self._gpu_chunk_digests(buf, nbytes)
+            blob.digest_blob = self._gpu_chunk_digests(buf, nbytes, vc)
             if expected_digests is not None:
-                check_digests(blob.digest_blob, expected_digests,
-                              self.verify_chunk)
+                check_digests(blob.digest_blob, expected_digests, vc)
         else:
             self.copy_stream.sync()
 
@@ -218,18 +219,21 @@ class Lander:
     def alloc(self, nbytes: int):
         return self._h.DeviceBuffer(max(nbytes, 1))
 
-    def finish_verify(self, buf, nbytes: int) -> bytes:
-        return self._gpu_chunk_digests(buf, nbytes)
+    def finish_verify(self, buf, nbytes: int,
+                      verify_chunk: int | None = None) -> bytes:
+        return self._gpu_chunk_digests(buf, nbytes, verify_chunk)
 
-    def _gpu_chunk_digests(self, buf, nbytes: int) -> bytes:
+    def _gpu_chunk_digests(self, buf, nbytes: int,
+                           verify_chunk: int | None = None) -> bytes:
         h = self._h
-        n_chunks = (nbytes + self.verify_chunk - 1) // self.verify_chunk
+        vc = verify_chunk or self.verify_chunk
+        n_chunks = (nbytes + vc - 1) // vc
         dig_dev = h.DeviceBuffer(n_chunks * 32)
         # hash must see completed copies
         done = h.Event()
         done.record(self.copy_stream.handle)
         done.wait(self.verify_stream.handle)
-        h.sha256_batch(buf.ptr, nbytes, self.verify_chunk, dig_dev.ptr,
+        h.sha256_batch(buf.ptr, nbytes, vc, dig_dev.ptr,
                        n_chunks, self.verify_stream.handle)
         host = bytearray(n_chunks * 32)
         import ctypes
@@ -358,11 +362,13 @@ class HostLander:
             if keep_head else bytearray()
         return head, time.perf_counter() - t0
 
-    def finish_verify(self, buf, nbytes: int) -> bytes:
+    def finish_verify(self, buf, nbytes: int,
+                      verify_chunk: int | None = None) -> bytes:
+        vc = verify_chunk or self.verify_chunk
         mv = memoryview(buf)
         return b"".join(
-            hashlib.sha256(mv[o:o + self.verify_chunk]).digest()
-            for o in range(0, nbytes, self.verify_chunk))
+            hashlib.sha256(mv[o:o + vc]).digest()
+            for o in range(0, nbytes, vc))
 
     def sync(self) -> None:
         pass
@@ -370,7 +376,8 @@ class HostLander:
     def land(self, fill, nbytes: int, verify: bool = True,
              host_chain: bool = False, gpu_chain: bool = False,
              keep_head: bool = True,
-             expected_digests: bytes | None = None) -> LandedBlob:
+             expected_digests: bytes | None = None,
+             verify_chunk: int | None = None) -> LandedBlob:
         t0 = time.perf_counter()
         buf = bytearray(nbytes)
         mv = memoryview(buf)
@@ -388,12 +395,12 @@ class HostLander:
             if chain is not None:
                 chain.update(mv[off:off + want])
             off += want
+        vc = verify_chunk or self.verify_chunk
         blob = LandedBlob(nbytes=nbytes, device="cpu", buffer=buf,
-                          verify_chunk=self.verify_chunk,
+                          verify_chunk=vc,
                           head=bytes(buf[:min(nbytes, self.head_bytes)]))
         t_land = time.perf_counter()
         if verify or expected_digests is not None:
-            vc = self.verify_chunk
             blob.digest_blob = b"".join(
                 hashlib.sha256(mv[o:o + vc]).digest()
                 for o in range(0, nbytes, vc))

@@ -143,7 +143,8 @@ RESUME_RETRIES = 4
 def _land_with_resume(lander, open_fn, nbytes: int, verify: bool,
                       expected_digests: bytes | None,
                       host_chain: bool = False,
-                      retries: int | None = None):
+                      retries: int | None = None,
+                      verify_chunk: int | None = None):
     """Land a blob with automatic Range-resume on mid-stream failures.
 
     open_fn(offset) -> fill callable streaming bytes [offset, nbytes).
@@ -180,15 +181,15 @@ def _land_with_resume(lander, open_fn, nbytes: int, verify: bool,
                      landed, nbytes, attempt)
             fill = open_fn(landed)
     lander.sync()
+    vc = verify_chunk or lander.verify_chunk
     device = ("cpu" if isinstance(buf, bytearray)
               else f"cuda:{lander.device_index}")
     blob = LandedBlob(nbytes=nbytes, device=device, buffer=buf,
-                      verify_chunk=lander.verify_chunk, head=bytes(head))
+                      verify_chunk=vc, head=bytes(head))
     if verify or expected_digests is not None:
-        blob.digest_blob = lander.finish_verify(buf, nbytes)
+        blob.digest_blob = lander.finish_verify(buf, nbytes, vc)
         if expected_digests is not None:
-            check_digests(blob.digest_blob, expected_digests,
-                          lander.verify_chunk)
+            check_digests(blob.digest_blob, expected_digests, vc)
     if chain is not None:
         blob.sha256 = chain.hexdigest()
     return blob
@@ -198,8 +199,8 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
                expected_sha: str | None, verify: str,
                cafile, insecure, headers=None,
                expected_digests: bytes | None = None,
-               seg_executor: cf.ThreadPoolExecutor | None = None
-               ) -> PulledFile:
+               seg_executor: cf.ThreadPoolExecutor | None = None,
+               verify_chunk: int | None = None) -> PulledFile:
     t0 = time.perf_counter()
     do_verify = verify in ("chunked", "digest", "gpu-digest")
     want_segments = (seg_executor is not None
@@ -218,12 +219,13 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
             if total > SEGMENT_MIN:
                 blob = _pull_segmented(landers, url, total, src,
                                        expected_digests, cafile, insecure,
-                                       headers, seg_executor)
+                                       headers, seg_executor, verify_chunk)
             else:
                 blob = landers.get().land(
                     src.fill, min(total, SEGMENT_MIN),
                     verify=do_verify,
-                    expected_digests=expected_digests)
+                    expected_digests=expected_digests,
+                    verify_chunk=verify_chunk)
         else:
             if src.status != 200:
                 raise fetch.FetchError(
@@ -236,7 +238,8 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
                 # the sequential GPU chain can't resume across re-fetches
                 blob = lander.land(
                     src.fill, nbytes, verify=do_verify, gpu_chain=True,
-                    expected_digests=expected_digests)
+                    expected_digests=expected_digests,
+                    verify_chunk=verify_chunk)
             else:
                 extra_sources = []
                 first = [True]
@@ -263,7 +266,8 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
                     blob = _land_with_resume(
                         lander, open_fn, nbytes, verify=do_verify,
                         expected_digests=expected_digests,
-                        host_chain=(verify == "digest"))
+                        host_chain=(verify == "digest"),
+                        verify_chunk=verify_chunk)
                 finally:
                     for s2 in extra_sources:
                         s2.close()
@@ -293,7 +297,7 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
 
 def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
                     expected_digests, cafile, insecure, headers,
-                    seg_executor) -> "object":
+                    seg_executor, verify_chunk=None) -> "object":
     """Range-parallel landing of one blob: segment 0 comes from the
     already-open 206 stream, the rest are parallel range GETs, all landing
     into disjoint ranges of one HBM buffer through per-thread pinned
@@ -341,16 +345,32 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     lander0.sync()
     for f in futs:
         f.result()
+    vc = verify_chunk or lander0.verify_chunk
     blob = LandedBlob(nbytes=total, device=f"cuda:{lander0.device_index}",
-                      buffer=buf, verify_chunk=lander0.verify_chunk,
-                      head=bytes(head))
-    blob.digest_blob = lander0._gpu_chunk_digests(buf, total)
+                      buffer=buf, verify_chunk=vc, head=bytes(head))
+    blob.digest_blob = lander0._gpu_chunk_digests(buf, total, vc)
     if expected_digests is not None:
         from .pipeline import check_digests
 
-        check_digests(blob.digest_blob, expected_digests,
-                      lander0.verify_chunk)
+        check_digests(blob.digest_blob, expected_digests, vc)
     return blob
+
+
+def fetch_peer_digests(endpoint: str, path: str, cafile=None,
+                       insecure=False) -> tuple[bytes, int] | None:
+    """Ask a demodel peer for its recorded chunk digests of `path`
+    (GET /__demodel/digests/<path>).  Returns (raw_digests, chunk_bytes)
+    or None when the peer has none (not a peer / not cached / digests
+    still computing)."""
+    try:
+        obj = fetch.get_json(f"{endpoint}/__demodel/digests{path}",
+                             cafile=cafile, insecure=insecure)
+    except (fetch.FetchError, OSError, ValueError):
+        return None
+    digs = obj.get("chunk_sha256") or []
+    if not digs or not obj.get("chunk_bytes"):
+        return None
+    return bytes.fromhex("".join(digs)), int(obj["chunk_bytes"])
 
 
 def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
@@ -360,7 +380,11 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             patterns: list[str] | None = None,
             landers: LanderPool | None = None,
             slab_bytes: int = 32 << 20,
-            digest_map: dict[str, bytes] | None = None) -> PullResult:
+            digest_map: dict[str, bytes] | None = None,
+            peer_verify: bool = False) -> PullResult:
+    """peer_verify: when `endpoint` is another demodel node, fetch its
+    recorded chunk digests per blob and GPU-verify the pull against them
+    (verified distribution)."""
     endpoint = (endpoint or os.environ.get("HF_ENDPOINT")
                 or HF_DEFAULT_ENDPOINT).rstrip("/")
     t0 = time.perf_counter()
@@ -376,15 +400,26 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
 
     seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
               if have_gpu() else None)
+
+    def peer_expected(n):
+        if not peer_verify:
+            return None
+        return fetch_peer_digests(endpoint, f"/{repo}/resolve/{rev}/{n}",
+                                  cafile=cafile, insecure=insecure)
+
     with cf.ThreadPoolExecutor(max_workers=workers) as ex:
-        futs = {
-            ex.submit(
+        futs = {}
+        for n in names:
+            pd = peer_expected(n)
+            exp = (digest_map or {}).get(n)
+            vc = None
+            if pd is not None and exp is None:
+                exp, vc = pd
+            futs[ex.submit(
                 _pull_blob, landers, n,
                 f"{endpoint}/{repo}/resolve/{rev}/{n}",
                 None, verify, cafile, insecure, None,
-                (digest_map or {}).get(n), seg_ex): n
-            for n in names
-        }
+                exp, seg_ex, vc)] = n
         for fut in cf.as_completed(futs):
             result.files.append(fut.result())
     if seg_ex:

@@ -42,8 +42,11 @@ class ProxyServer:
                  cache: CacheStore | None = None):
         self.cfg = cfg
         self.leafs = leafs
+        # async digesting: blob fills must not run at hashlib speed; the
+        # digests land in the meta a moment after commit
         self.cache = cache or CacheStore(cfg.cache_dir,
-                                         chunk_bytes=cfg.chunk_bytes)
+                                         chunk_bytes=1 << 20,
+                                         digest_mode="async")
         self.transfers = TransferLog()
         self._server: asyncio.AbstractServer | None = None
         self.port: int | None = None
@@ -221,6 +224,61 @@ class ProxyServer:
                     u.hostname, port, is_tls, t
         raise ProtocolError(f"no reverse route for {t!r}")
 
+    async def _serve_digests(self, head: RequestHead, writer) -> bool:
+        """GET /__demodel/digests/<path> — the cache's recorded per-chunk
+        sha256 digests for the entry <path> resolves to (following the
+        cached redirect chain).  A peer pulling this path from us can
+        GPU-verify every chunk against this record (verified
+        distribution; the cache computes digests at fill time —
+        cache/store.py)."""
+        import json as _json
+
+        path = head.target[len("/__demodel/digests"):]
+        if not path.startswith("/"):
+            return await self._simple(writer, head, 400,
+                                      b'{"error": "bad path"}')
+        try:
+            uri, *_ = self._canonical_uri(
+                RequestHead("GET", path, "HTTP/1.1", []), None)
+        except ProtocolError:
+            return await self._simple(writer, head, 404,
+                                      b'{"error": "no route"}')
+        hit = None
+        lookup_uri = uri
+        for _ in range(6):
+            hit = self.cache.lookup(lookup_uri)
+            if hit is None or not (300 <= hit.status < 400):
+                break
+            loc = dict((k.lower(), v) for k, v in hit.headers
+                       ).get("location")
+            if not loc:
+                break
+            lookup_uri = self._absolute_uri(loc, lookup_uri)
+        if hit is None or 300 <= hit.status < 400:
+            return await self._simple(writer, head, 404,
+                                      b'{"error": "not cached"}')
+        body = _json.dumps({
+            "uri": lookup_uri,
+            "body_size": hit.body_size,
+            "chunk_bytes": hit.chunk_bytes,
+            "sha256": hit.sha256,
+            "chunk_sha256": hit.chunk_sha256,
+        }).encode()
+        return await self._simple(writer, head, 200, body)
+
+    async def _simple(self, writer, head, status: int,
+                      body: bytes) -> bool:
+        out = ResponseHead("HTTP/1.1", status,
+                           "OK" if status == 200 else "Error",
+                           [("Content-Type", "application/json"),
+                            ("Content-Length", str(len(body))),
+                            ("Connection", "keep-alive")])
+        writer.write(http1.serialize_response(out))
+        if head.method != "HEAD":
+            writer.write(body)
+        await writer.drain()
+        return head.get("connection", "").lower() == "close"
+
     async def _serve_stats(self, head: RequestHead, writer) -> bool:
         """GET /__demodel/stats — transfer metrics (observability; the
         reference had only two println hooks)."""
@@ -266,6 +324,8 @@ class ProxyServer:
 
     async def _handle_request(self, head: RequestHead, reader, writer,
                               tls_host: str | None) -> bool:
+        if head.target.startswith("/__demodel/digests/"):
+            return await self._serve_digests(head, writer)
         if head.target.startswith("/__demodel/"):
             return await self._serve_stats(head, writer)
         uri, host, port, is_tls, path = self._canonical_uri(head, tls_host)
