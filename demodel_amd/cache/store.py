@@ -229,6 +229,34 @@ class CacheStore:
     def cacheable(self, method: str, status: int) -> bool:
         return method == "GET" and status in (200, 301, 302, 307, 308)
 
+    def gc(self, max_bytes: int) -> dict:
+        """Evict least-recently-used entries until total body bytes fit
+        under max_bytes.  Recency = body file atime (falls back to
+        mtime).  Returns {kept, evicted, bytes}."""
+        entries = []
+        for fn in os.listdir(self.root):
+            if not fn.endswith(".meta"):
+                continue
+            body = os.path.join(self.root, fn[:-5])
+            if not os.path.exists(body):
+                continue
+            st_ = os.stat(body)
+            entries.append((max(st_.st_atime, st_.st_mtime), body,
+                            st_.st_size))
+        total = sum(sz for _, _, sz in entries)
+        evicted = 0
+        entries.sort()  # oldest first
+        for _, body, sz in entries:
+            if total <= max_bytes:
+                break
+            for p in (body, body + ".meta"):
+                if os.path.exists(p):
+                    os.unlink(p)
+            total -= sz
+            evicted += 1
+        return {"kept": len(entries) - evicted, "evicted": evicted,
+                "bytes": total}
+
     def purge(self, uri: str) -> bool:
         key = cache_key(uri)
         body_path = os.path.join(self.root, key)

@@ -86,6 +86,16 @@ def _cmd_verify(args) -> int:
     return 0 if result.get("ok", False) else 1
 
 
+def _cmd_gc(args) -> int:
+    from .cache import CacheStore
+
+    cfg = load_config()
+    store = CacheStore(cfg.cache_dir)
+    result = store.gc(int(args.max_gb * 1e9))
+    print(json.dumps(result))
+    return 0
+
+
 def main(argv: list[str] | None = None) -> int:
     p = argparse.ArgumentParser(
         prog="demodel",
@@ -119,6 +129,11 @@ def main(argv: list[str] | None = None) -> int:
     vp = sub.add_parser("verify", help="re-verify cached blobs")
     vp.add_argument("--uri", default=None)
     vp.set_defaults(fn=_cmd_verify)
+
+    gp = sub.add_parser("gc", help="evict LRU cache entries over a size "
+                                   "budget")
+    gp.add_argument("--max-gb", type=float, required=True)
+    gp.set_defaults(fn=_cmd_gc)
 
     args = p.parse_args(argv)
     if not hasattr(args, "fn"):

@@ -92,3 +92,44 @@ def test_torn_entry_is_miss(tmp_path):
     with open(entry.body_path, "wb") as f:
         f.write(b"0123")  # truncate the body behind the meta's back
     assert store.lookup(uri) is None
+
+
+def test_gc_lru(tmp_path):
+    import time as _time
+
+    store = CacheStore(str(tmp_path))
+    for i in range(5):
+        w = store.writer(f"https://host/gc{i}", 200, "OK", [])
+        w.write(b"x" * 1000)
+        w.finalize()
+        _time.sleep(0.02)
+    # touch entry 0 so it becomes most-recent
+    e0 = store.lookup("https://host/gc0")
+    now = _time.time()
+    os.utime(e0.body_path, (now + 10, now + 10))
+    res = store.gc(max_bytes=2500)
+    assert res["evicted"] == 3
+    assert store.lookup("https://host/gc0") is not None  # recently used
+    assert store.lookup("https://host/gc4") is not None  # newest
+    assert store.lookup("https://host/gc1") is None
+
+
+def test_verify_cache_cli(tmp_path):
+    from demodel_amd.config import Config
+    from demodel_amd.engine.pull import verify_cache
+
+    store = CacheStore(str(tmp_path))
+    w = store.writer("https://host/v", 200, "OK", [])
+    w.write(os.urandom(5000))
+    w.finalize()
+    cfg = Config(cache_dir=str(tmp_path))
+    res = verify_cache(cfg)
+    assert res["ok"] and res["checked"] == 1
+
+    # corrupt the body -> verify fails
+    e = store.lookup("https://host/v")
+    with open(e.body_path, "r+b") as f:
+        f.seek(100)
+        f.write(b"\x00\x01\x02")
+    res2 = verify_cache(cfg)
+    assert not res2["ok"] and res2["bad"] == ["https://host/v"]
