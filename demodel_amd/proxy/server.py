@@ -346,7 +346,8 @@ class ProxyServer:
         # ---- hook 2: cache lookup --------------------------------------
         # In reverse mode a cached redirect chain is followed inside the
         # cache, so a fully cached pull replays with zero upstream traffic.
-        if self.cache.cacheable(head.method, 200):
+        # HEAD requests serve from cached GET entries (headers only).
+        if head.method in ("GET", "HEAD"):
             lookup_uri = uri
             for _ in range(6):
                 hit = self.cache.lookup(lookup_uri)
