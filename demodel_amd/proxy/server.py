@@ -79,9 +79,18 @@ class ProxyServer:
     # lifecycle
 
     async def start(self) -> int:
+        self._tasks: set = set()
+
+        async def entry(reader, writer):
+            task = asyncio.current_task()
+            self._tasks.add(task)
+            try:
+                await self._handle_client(reader, writer)
+            finally:
+                self._tasks.discard(task)
+
         self._server = await asyncio.start_server(
-            self._handle_client, self.cfg.host, self.cfg.port,
-            limit=http1.MAX_HEAD,
+            entry, self.cfg.host, self.cfg.port, limit=http1.MAX_HEAD,
         )
         self.port = self._server.sockets[0].getsockname()[1]
         log.info("listening on %s:%d", self.cfg.host, self.port)
@@ -91,6 +100,10 @@ class ProxyServer:
         if self._server:
             self._server.close()
             await self._server.wait_closed()
+        for t in list(getattr(self, "_tasks", ())):
+            t.cancel()
+        if getattr(self, "_tasks", None):
+            await asyncio.gather(*self._tasks, return_exceptions=True)
 
     def upstream_ssl(self) -> ssl.SSLContext:
         if self._upstream_ssl is None:

@@ -137,6 +137,44 @@ def inflate_bench():
           flush=True)
 
 
+def zstd_bench(n_frames=2048, frame_mb=2, payload="text"):
+    import ctypes
+
+    import pyarrow as pa
+
+    from demodel_amd.engine.formats.compress import zstd_gpu
+    from demodel_amd.gpu import hip
+
+    h = hip()
+    s = h.Stream(0)
+    n = frame_mb << 20
+    if payload == "text":
+        base = (b"some plainly compressible text payload flows here " * 200
+                + os.urandom(1 << 14))
+    else:
+        base = os.urandom(1 << 20)
+    data = (base * (n // len(base) + 1))[:n]
+    frame = bytes(pa.Codec("zstd", compression_level=1).compress(data))
+    src = h.DeviceBuffer(len(frame))
+    carr = (ctypes.c_char * len(frame)).from_buffer_copy(frame)
+    h.h2d_async(src.ptr, ctypes.addressof(carr), len(frame), s.handle)
+    s.sync()
+    ring = h.DeviceBuffer(n_frames * n)
+    frames = [(src.ptr, len(frame), ring.ptr + i * n, n)
+              for i in range(n_frames)]
+
+    t = bench(lambda: zstd_gpu(frames), iters=2, warmup=1)
+    print(json.dumps({"op": "zstd_decode", "payload": payload,
+                      "frames": n_frames, "mib_each": frame_mb,
+                      "s": round(t, 3),
+                      "GBps_out": round(n_frames * n / t / 1e9, 2),
+                      "MBps_per_wave": round(n / t / 1e6, 2),
+                      "ratio": round(n / len(frame), 2)}), flush=True)
+
+
 if __name__ == "__main__":
     main()
     inflate_bench()
+    zstd_bench()
+    zstd_bench(n_frames=256)
+    zstd_bench(payload="random")
