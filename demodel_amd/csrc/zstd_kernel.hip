@@ -13,13 +13,24 @@ using namespace zstd_core;
 
 namespace {
 
+// LDS output window: match copies read recent output from LDS instead of
+// global memory, which removes the per-sequence vmcnt(0)+barrier ordering
+// that dominated literal/sequence-heavy payloads (measured: matchy frames
+// decode at ~108 MB/s/wave while word-salad crawled at ~1.5 — the
+// difference was per-sequence synchronization, not decode math).  Window
+// slot p & (ZWIN-1) holds output byte p; every producer (literals,
+// matches, raw/RLE blocks) maintains it.  Matches farther back than the
+// window take a rare global-read path behind an explicit vmcnt drain.
+#define ZWIN (64 * 1024)
+#define ZWMASK (ZWIN - 1)
+
 __global__ void __launch_bounds__(64)
 zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
   __shared__ ZShared sh;
   __shared__ SeqRec cur;
-  __shared__ uint64_t lit_used;
   __shared__ int seq_err;
   __shared__ BBits sq;
+  __shared__ uint8_t win[ZWIN];
   int lane = threadIdx.x;
 
   for (int sidx = blockIdx.x; sidx < n_streams; sidx += gridDim.x) {
@@ -273,9 +284,16 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         done = true;
       } else if (op == ZOP_COPY) {
         const uint8_t* s = (const uint8_t*)sh.a;
-        for (uint64_t i = lane; i < sh.c; i += 64) out[sh.b + i] = s[i];
+        for (uint64_t i = lane; i < sh.c; i += 64) {
+          uint8_t v = s[i];
+          out[sh.b + i] = v;
+          win[(sh.b + i) & ZWMASK] = v;
+        }
       } else if (op == ZOP_FILL) {
-        for (uint64_t i = lane; i < sh.c; i += 64) out[sh.b + i] = sh.fillv;
+        for (uint64_t i = lane; i < sh.c; i += 64) {
+          out[sh.b + i] = sh.fillv;
+          win[(sh.b + i) & ZWMASK] = sh.fillv;
+        }
       } else if (op == ZOP_LITS) {
         // ---- 1) literals into ws ------------------------------------
         if (sh.hstreams == 0) {
@@ -303,9 +321,11 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         }
         __syncthreads();
 
-        // ---- 2) sequences: lane 0 decodes, wave executes ------------
+        // ---- 2) sequences: lane 0 decodes, the wave executes --------
+        // Single-wave workgroup: LDS producer->consumer ordering inside
+        // one wave needs only lgkmcnt, so the per-sequence loop runs
+        // with NO barriers; matches read the LDS window.
         if (lane == 0) {
-          lit_used = 0;
           seq_err = sh.herr;
           if (sh.n_seqs > 0 && !seq_err) {
             sq.init((const uint8_t*)sh.a, (int64_t)sh.c);
@@ -320,6 +340,9 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           of_state = sq.get(sh.of_t.log);
           ml_state = sq.get(sh.ml_t.log);
         }
+        uint64_t pos_r = sh.pos;       // per-lane replicated position
+        uint64_t lit_used = 0;
+        const uint8_t* lit_base = (const uint8_t*)sh.lit_ptr;
         for (uint32_t i = 0; i < nseq; ++i) {
           if (lane == 0 && !seq_err) {
             uint8_t ofc = sh.of_t.e[of_state].sym;
@@ -364,43 +387,60 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
               }
               cur.ll = ll; cur.ml = ml; cur.off = offset;
               if (lit_used + ll > sh.lit_len ||
-                  sh.pos + ll + ml > d->dst_cap ||
-                  (uint64_t)offset > sh.pos + ll)
+                  pos_r + ll + ml > d->dst_cap ||
+                  (uint64_t)offset > pos_r + ll)
                 seq_err = 1;
             }
           }
-          __syncthreads();
+          // within-wave LDS publish of cur/seq_err
+          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
           if (seq_err) break;
-          uint64_t p0 = sh.pos;
-          const uint8_t* lsrc = (const uint8_t*)sh.lit_ptr + lit_used;
-          for (uint32_t k = lane; k < cur.ll; k += 64)
-            out[p0 + k] = lsrc[k];
-          __syncthreads();
-          {
-            uint64_t mp = p0 + cur.ll;
-            uint32_t dist = cur.off, len = cur.ml;
+          uint32_t ll = cur.ll, len = cur.ml, dist = cur.off;
+          uint64_t p0 = pos_r;
+          const uint8_t* lsrc = lit_base + lit_used;
+          for (uint32_t k = lane; k < ll; k += 64) {
+            uint8_t v = lsrc[k];
+            out[p0 + k] = v;
+            win[(p0 + k) & ZWMASK] = v;
+          }
+          uint64_t mp = p0 + ll;
+          if (dist <= ZWIN - 128) {
+            // near match: source bytes live in the LDS window
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             if (dist >= len) {
-              for (uint32_t k = lane; k < len; k += 64)
-                out[mp + k] = out[mp + k - dist];
+              for (uint32_t k = lane; k < len; k += 64) {
+                uint8_t v = win[(mp + k - dist) & ZWMASK];
+                out[mp + k] = v;
+                win[(mp + k) & ZWMASK] = v;
+              }
             } else {
               uint64_t copied = 0;
               while (copied < len) {
                 uint32_t n = (uint32_t)min((uint64_t)dist,
                                            (uint64_t)len - copied);
-                for (uint32_t k = lane; k < n; k += 64)
-                  out[mp + copied + k] = out[mp + copied + k - dist];
-                __syncthreads();
+                for (uint32_t k = lane; k < n; k += 64) {
+                  uint8_t v = win[(mp + copied + k - dist) & ZWMASK];
+                  out[mp + copied + k] = v;
+                  win[(mp + copied + k) & ZWMASK] = v;
+                }
+                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
                 copied += n;
               }
             }
+          } else {
+            // far match: read old output from HBM; drain our stores first
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            for (uint32_t k = lane; k < len; k += 64) {
+              uint8_t v = out[mp + k - dist];
+              out[mp + k] = v;
+              win[(mp + k) & ZWMASK] = v;
+            }
           }
-          __syncthreads();
-          if (lane == 0) {
-            lit_used += cur.ll;
-            sh.pos += cur.ll + cur.ml;
-          }
-          __syncthreads();
+          pos_r += ll + len;
+          lit_used += ll;
         }
+        if (lane == 0 && !seq_err) sh.pos = pos_r;
+        __syncthreads();
         // trailing literals
         if (lane == 0 && !seq_err) {
           uint64_t rest = sh.lit_len - lit_used;
@@ -420,7 +460,11 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           done = true;
         } else {
           const uint8_t* s = (const uint8_t*)sh.a;
-          for (uint64_t k = lane; k < sh.c; k += 64) out[sh.b + k] = s[k];
+          for (uint64_t k = lane; k < sh.c; k += 64) {
+            uint8_t v = s[k];
+            out[sh.b + k] = v;
+            win[(sh.b + k) & ZWMASK] = v;
+          }
         }
       }
       __syncthreads();

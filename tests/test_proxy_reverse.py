@@ -103,7 +103,13 @@ def test_cache_entry_chunk_digests_match(stack, tmp_path):
             break
         time.sleep(0.05)
     assert cached
-    hit = stack.proxy.cache.lookup(cached[0]["uri"])
+    # digests are computed asynchronously after commit
+    hit = None
+    for _ in range(100):
+        hit = stack.proxy.cache.lookup(cached[0]["uri"])
+        if hit is not None and hit.sha256:
+            break
+        time.sleep(0.05)
     assert hit is not None
     assert hit.sha256 == hashlib.sha256(data).hexdigest()
     assert hit.read_body() == data
