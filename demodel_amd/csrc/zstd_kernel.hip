@@ -21,15 +21,18 @@ namespace {
 // slot p & (ZWIN-1) holds output byte p; every producer (literals,
 // matches, raw/RLE blocks) maintains it.  Matches farther back than the
 // window take a rare global-read path behind an explicit vmcnt drain.
-#define ZWIN (64 * 1024)
-#define ZWMASK (ZWIN - 1)
-
+// Window size trades per-wave speed (bigger window = fewer far matches)
+// against occupancy (LDS-limited resident workgroups); the launcher picks
+// 16 KiB when many frames provide parallelism, 64 KiB for few frames.
+template <int ZWIN>
 __global__ void __launch_bounds__(64)
 zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
+  constexpr int ZWMASK = ZWIN - 1;
   __shared__ ZShared sh;
   __shared__ SeqRec cur;
   __shared__ int seq_err;
   __shared__ BBits sq;
+  __shared__ uint64_t win_from;   // output pos from which win[] is valid
   __shared__ uint8_t win[ZWIN];
   int lane = threadIdx.x;
 
@@ -53,6 +56,7 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
       sh.status = Z_OK;
       sh.op = ZOP_NONE;
       sh.herr = 0;
+      win_from = 0;
     }
     __syncthreads();
 
@@ -284,16 +288,11 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         done = true;
       } else if (op == ZOP_COPY) {
         const uint8_t* s = (const uint8_t*)sh.a;
-        for (uint64_t i = lane; i < sh.c; i += 64) {
-          uint8_t v = s[i];
-          out[sh.b + i] = v;
-          win[(sh.b + i) & ZWMASK] = v;
-        }
+        for (uint64_t i = lane; i < sh.c; i += 64) out[sh.b + i] = s[i];
+        if (lane == 0) win_from = sh.b + sh.c;  // window gap over this block
       } else if (op == ZOP_FILL) {
-        for (uint64_t i = lane; i < sh.c; i += 64) {
-          out[sh.b + i] = sh.fillv;
-          win[(sh.b + i) & ZWMASK] = sh.fillv;
-        }
+        for (uint64_t i = lane; i < sh.c; i += 64) out[sh.b + i] = sh.fillv;
+        if (lane == 0) win_from = sh.b + sh.c;
       } else if (op == ZOP_LITS) {
         // ---- 1) literals into ws ------------------------------------
         if (sh.hstreams == 0) {
@@ -341,6 +340,7 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           ml_state = sq.get(sh.ml_t.log);
         }
         uint64_t pos_r = sh.pos;       // per-lane replicated position
+        uint64_t wfrom = win_from;     // uniform: only changes between ops
         uint64_t lit_used = 0;
         const uint8_t* lit_base = (const uint8_t*)sh.lit_ptr;
         for (uint32_t i = 0; i < nseq; ++i) {
@@ -404,7 +404,7 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
             win[(p0 + k) & ZWMASK] = v;
           }
           uint64_t mp = p0 + ll;
-          if (dist <= ZWIN - 128) {
+          if (dist <= ZWIN - 128 && mp - dist >= wfrom) {
             // near match: source bytes live in the LDS window
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             if (dist >= len) {
@@ -485,7 +485,13 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
 extern "C" void launch_zstd_frames(const uint64_t* desc, int n_frames,
                                    int* /*unused*/, hipStream_t stream) {
   if (n_frames <= 0) return;
-  int blocks = n_frames < 2048 ? n_frames : 2048;
-  hipLaunchKernelGGL(zstd_kernel, dim3(blocks), dim3(64), 0, stream,
-                     (ZstdDesc*)desc, n_frames);
+  int blocks = n_frames < 4096 ? n_frames : 4096;
+  if (n_frames >= 768) {
+    // occupancy regime: small window, more resident workgroups
+    hipLaunchKernelGGL(zstd_kernel<16 * 1024>, dim3(blocks), dim3(64), 0,
+                       stream, (ZstdDesc*)desc, n_frames);
+  } else {
+    hipLaunchKernelGGL(zstd_kernel<64 * 1024>, dim3(blocks), dim3(64), 0,
+                       stream, (ZstdDesc*)desc, n_frames);
+  }
 }

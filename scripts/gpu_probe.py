@@ -151,6 +151,13 @@ def zstd_bench(n_frames=2048, frame_mb=2, payload="text"):
     if payload == "text":
         base = (b"some plainly compressible text payload flows here " * 200
                 + os.urandom(1 << 14))
+    elif payload == "words":
+        import numpy as np
+
+        rng = np.random.default_rng(5)
+        words = [f"w{i:04d}" for i in range(20000)]
+        idx = rng.integers(0, len(words), size=(2 << 20) // 6)
+        base = " ".join(words[i] for i in idx).encode()
     else:
         base = os.urandom(1 << 20)
     data = (base * (n // len(base) + 1))[:n]
