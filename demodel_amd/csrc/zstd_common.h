@@ -87,6 +87,16 @@ struct BBits {
     refill();
   }
   ZFN void refill() {
+    // batched reload: 4 bytes per step (backward stream = byteswapped
+    // little-endian load) instead of one global byte per iteration
+    while (nbits <= 32 && byte >= 3) {
+      uint32_t v;
+      __builtin_memcpy(&v, base + byte - 3, 4);
+      v = __builtin_bswap32(v);
+      cont = (cont << 32) | v;
+      nbits += 32;
+      byte -= 4;
+    }
     while (nbits <= 56 && byte >= 0) {
       cont = (cont << 8) | base[byte];
       nbits += 8;

@@ -31,7 +31,6 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
   __shared__ ZShared sh;
   __shared__ SeqRec cur;
   __shared__ int seq_err;
-  __shared__ BBits sq;
   __shared__ uint64_t win_from;   // output pos from which win[] is valid
   __shared__ uint8_t win[ZWIN];
   int lane = threadIdx.x;
@@ -324,6 +323,7 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         // Single-wave workgroup: LDS producer->consumer ordering inside
         // one wave needs only lgkmcnt, so the per-sequence loop runs
         // with NO barriers; matches read the LDS window.
+        BBits sq;  // lane-0 private (registers); others never touch it
         if (lane == 0) {
           seq_err = sh.herr;
           if (sh.n_seqs > 0 && !seq_err) {
