@@ -90,9 +90,12 @@ struct BBits {
     // batched reload: 4 bytes per step (backward stream = byteswapped
     // little-endian load) instead of one global byte per iteration
     while (nbits <= 32 && byte >= 3) {
-      uint32_t v;
-      __builtin_memcpy(&v, base + byte - 3, 4);
-      v = __builtin_bswap32(v);
+      // four independent byte loads (explicit: unaligned wide loads
+      // misbehave on device)
+      uint32_t v = ((uint32_t)base[byte] << 24) |
+                   ((uint32_t)base[byte - 1] << 16) |
+                   ((uint32_t)base[byte - 2] << 8) |
+                   (uint32_t)base[byte - 3];
       cont = (cont << 32) | v;
       nbits += 32;
       byte -= 4;
