@@ -319,22 +319,20 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         }
         __syncthreads();
 
-        // ---- 2) sequences: lane 0 decodes, the wave executes --------
-        // Single-wave workgroup: LDS producer->consumer ordering inside
-        // one wave needs only lgkmcnt, so the per-sequence loop runs
-        // with NO barriers; matches read the LDS window.
-        BBits sq;  // lane-0 private (registers); others never touch it
-        if (lane == 0) {
-          seq_err = sh.herr;
-          if (sh.n_seqs > 0 && !seq_err) {
-            sq.init((const uint8_t*)sh.a, (int64_t)sh.c);
-            if (sq.fail) seq_err = 1;
-          }
-        }
+        // ---- 2) sequences: every lane decodes redundantly -----------
+        // All 64 lanes run the identical FSE decode in lockstep on their
+        // own register state (LDS table reads broadcast, bitstream reads
+        // coalesce to one address) — zero cross-lane publication, zero
+        // waits in the per-sequence loop; matches read the LDS window.
         __syncthreads();
         uint32_t nseq = sh.n_seqs;
+        int err = sh.herr;
+        BBits sq;
+        sq.init((const uint8_t*)sh.a, (int64_t)sh.c);
         uint32_t ll_state = 0, of_state = 0, ml_state = 0;
-        if (nseq && lane == 0 && !seq_err) {
+        uint32_t rep0 = sh.rep[0], rep1 = sh.rep[1], rep2 = sh.rep[2];
+        if (nseq) {
+          if (sq.fail) err = 1;
           ll_state = sq.get(sh.ll_t.log);
           of_state = sq.get(sh.of_t.log);
           ml_state = sq.get(sh.ml_t.log);
@@ -343,59 +341,48 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         uint64_t wfrom = win_from;     // uniform: only changes between ops
         uint64_t lit_used = 0;
         const uint8_t* lit_base = (const uint8_t*)sh.lit_ptr;
-        for (uint32_t i = 0; i < nseq; ++i) {
-          if (lane == 0 && !seq_err) {
-            uint8_t ofc = sh.of_t.e[of_state].sym;
-            uint8_t mlc = sh.ml_t.e[ml_state].sym;
-            uint8_t llc = sh.ll_t.e[ll_state].sym;
-            uint32_t ofv = 0;
-            if (ofc > 31 || mlc > 52 || llc > 35) seq_err = 1;
-            if (!seq_err) {
-              ofv = (1u << ofc) + sq.get(ofc);
-              uint32_t ml = kMLBase[mlc] + sq.get(kMLExtra[mlc]);
-              uint32_t ll = kLLBase[llc] + sq.get(kLLExtra[llc]);
-              uint32_t offset;
-              if (ofv > 3) {
-                offset = ofv - 3;
-                sh.rep[2] = sh.rep[1]; sh.rep[1] = sh.rep[0];
-                sh.rep[0] = offset;
-              } else {
-                uint32_t idx = ofv + (ll == 0 ? 1 : 0);
-                if (idx == 1) {
-                  offset = sh.rep[0];
-                } else if (idx == 2) {
-                  offset = sh.rep[1];
-                  sh.rep[1] = sh.rep[0]; sh.rep[0] = offset;
-                } else if (idx == 3) {
-                  offset = sh.rep[2];
-                  sh.rep[2] = sh.rep[1]; sh.rep[1] = sh.rep[0];
-                  sh.rep[0] = offset;
-                } else {
-                  offset = sh.rep[0] - 1;
-                  if (offset == 0 || sh.rep[0] == 0) seq_err = 1;
-                  sh.rep[2] = sh.rep[1]; sh.rep[1] = sh.rep[0];
-                  sh.rep[0] = offset;
-                }
-              }
-              if (i + 1 < nseq) {
-                ll_state = sh.ll_t.e[ll_state].base
-                           + sq.get(sh.ll_t.e[ll_state].nbits);
-                ml_state = sh.ml_t.e[ml_state].base
-                           + sq.get(sh.ml_t.e[ml_state].nbits);
-                of_state = sh.of_t.e[of_state].base
-                           + sq.get(sh.of_t.e[of_state].nbits);
-              }
-              cur.ll = ll; cur.ml = ml; cur.off = offset;
-              if (lit_used + ll > sh.lit_len ||
-                  pos_r + ll + ml > d->dst_cap ||
-                  (uint64_t)offset > pos_r + ll)
-                seq_err = 1;
+        for (uint32_t i = 0; i < nseq && !err; ++i) {
+          uint8_t ofc = sh.of_t.e[of_state].sym;
+          uint8_t mlc = sh.ml_t.e[ml_state].sym;
+          uint8_t llc = sh.ll_t.e[ll_state].sym;
+          if (ofc > 31 || mlc > 52 || llc > 35) { err = 1; break; }
+          uint32_t ofv = (1u << ofc) + sq.get(ofc);
+          uint32_t len = kMLBase[mlc] + sq.get(kMLExtra[mlc]);
+          uint32_t ll = kLLBase[llc] + sq.get(kLLExtra[llc]);
+          uint32_t dist;
+          if (ofv > 3) {
+            dist = ofv - 3;
+            rep2 = rep1; rep1 = rep0; rep0 = dist;
+          } else {
+            uint32_t idx = ofv + (ll == 0 ? 1 : 0);
+            if (idx == 1) {
+              dist = rep0;
+            } else if (idx == 2) {
+              dist = rep1;
+              rep1 = rep0; rep0 = dist;
+            } else if (idx == 3) {
+              dist = rep2;
+              rep2 = rep1; rep1 = rep0; rep0 = dist;
+            } else {
+              dist = rep0 - 1;
+              if (dist == 0 || rep0 == 0) { err = 1; break; }
+              rep2 = rep1; rep1 = rep0; rep0 = dist;
             }
           }
-          // within-wave LDS publish of cur/seq_err
-          asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-          if (seq_err) break;
-          uint32_t ll = cur.ll, len = cur.ml, dist = cur.off;
+          if (i + 1 < nseq) {
+            ll_state = sh.ll_t.e[ll_state].base
+                       + sq.get(sh.ll_t.e[ll_state].nbits);
+            ml_state = sh.ml_t.e[ml_state].base
+                       + sq.get(sh.ml_t.e[ml_state].nbits);
+            of_state = sh.of_t.e[of_state].base
+                       + sq.get(sh.of_t.e[of_state].nbits);
+          }
+          if (lit_used + ll > sh.lit_len ||
+              pos_r + ll + (uint64_t)len > d->dst_cap ||
+              (uint64_t)dist > pos_r + ll) {
+            err = 1;
+            break;
+          }
           uint64_t p0 = pos_r;
           const uint8_t* lsrc = lit_base + lit_used;
           for (uint32_t k = lane; k < ll; k += 64) {
@@ -439,7 +426,11 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           pos_r += ll + len;
           lit_used += ll;
         }
-        if (lane == 0 && !seq_err) sh.pos = pos_r;
+        if (lane == 0) {
+          seq_err = err;
+          sh.rep[0] = rep0; sh.rep[1] = rep1; sh.rep[2] = rep2;
+        }
+        if (lane == 0 && !err) sh.pos = pos_r;
         __syncthreads();
         // trailing literals
         if (lane == 0 && !seq_err) {
