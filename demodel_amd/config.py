@@ -69,6 +69,7 @@ class Config:
 
     # --- cache ---
     cache_dir: str = ".cache"
+    cache_max_bytes: int | None = None   # LRU-evict over this (None = off)
 
     # --- upstream TLS ---
     upstream_cafile: str | None = None  # extra CA for origin verification (tests)
@@ -106,6 +107,9 @@ def load_config(**overrides) -> Config:
         host=os.environ.get("DEMODEL_HOST", "0.0.0.0"),
         port=_env_int("DEMODEL_PORT", 8080),
         cache_dir=os.environ.get("DEMODEL_CACHE_DIR", ".cache"),
+        cache_max_bytes=(
+            int(float(os.environ["DEMODEL_CACHE_MAX_GB"]) * 1e9)
+            if os.environ.get("DEMODEL_CACHE_MAX_GB") else None),
         upstream_cafile=os.environ.get("DEMODEL_UPSTREAM_CAFILE") or None,
         upstream_insecure=_env_bool("DEMODEL_UPSTREAM_INSECURE"),
         chunk_bytes=_env_int("DEMODEL_CHUNK_BYTES", 32 << 20),

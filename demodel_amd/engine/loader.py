@@ -28,6 +28,30 @@ def load_into(blob, header: SafetensorsHeader, targets: dict,
     """
     import torch
 
+    if blob.device == "cpu":
+        # CPU landing target: plain torch copies from zero-copy views
+        from .formats.safetensors import torch_views
+
+        views = torch_views(header, blob.torch_u8())
+        loaded = []
+        for name, dst in targets.items():
+            v = views.get(name)
+            if v is None:
+                if strict:
+                    raise KeyError(f"tensor {name!r} not in checkpoint")
+                continue
+            if tuple(dst.shape) != tuple(v.shape):
+                raise ValueError(f"{name}: shape mismatch")
+            with torch.no_grad():
+                dst.copy_(v.to(dst.dtype))
+            loaded.append(name)
+        if strict:
+            missing = {t.name for t in header.tensors} - set(targets)
+            if missing:
+                raise KeyError(f"model is missing tensors: "
+                               f"{sorted(missing)[:5]}")
+        return loaded
+
     h = hip()
     own = stream is None
     s = h.Stream(0) if own else stream
@@ -76,6 +100,53 @@ def load_into(blob, header: SafetensorsHeader, targets: dict,
     # the launches
     s.sync()
     return loaded
+
+
+def pull_pretrained(repo: str, endpoint: str | None = None,
+                    device: str | None = None, model_builder=None,
+                    **pull_kw):
+    """Pull an HF repo and materialize a transformers model with weights
+    scattered straight from the landed HBM blobs (no host round-trip on
+    GPU machines).
+
+    model_builder(config) -> nn.Module; defaults to
+    transformers.AutoModelForCausalLM.from_config.
+    Returns (model, PullResult)."""
+    import json
+
+    import torch
+
+    from ..gpu import have_gpu
+    from .pull import pull_hf
+
+    res = pull_hf(repo, endpoint=endpoint, **pull_kw)
+    cfg_file = next((f for f in res.files if f.name == "config.json"),
+                    None)
+    if cfg_file is None:
+        raise FileNotFoundError("repo has no config.json")
+    cfg_bytes = (bytes(cfg_file.blob.head[:cfg_file.nbytes])
+                 if cfg_file.blob.device != "cpu"
+                 else bytes(cfg_file.blob.buffer))
+    cfg_dict = json.loads(cfg_bytes)
+
+    import transformers
+
+    config = transformers.AutoConfig.for_model(
+        cfg_dict.get("model_type"), **{
+            k: v for k, v in cfg_dict.items() if k != "model_type"})
+    if device is None:
+        device = "cuda" if have_gpu() else "cpu"
+    if model_builder is None:
+        def model_builder(c):
+            return transformers.AutoModelForCausalLM.from_config(c)
+    with torch.device(device):
+        model = model_builder(config)
+    model = model.to(device)
+    model.eval()  # match from_pretrained semantics
+    n = load_model_from_pull(res, model)
+    if n == 0:
+        raise RuntimeError("no tensors loaded from pull")
+    return model, res
 
 
 def load_model_from_pull(result, model, strict: bool = False) -> int:

@@ -237,6 +237,22 @@ class ProxyServer:
                     u.hostname, port, is_tls, t
         raise ProtocolError(f"no reverse route for {t!r}")
 
+    _gc_countdown = 0
+
+    def _maybe_gc(self) -> None:
+        """LRU-evict over the configured budget (DEMODEL_CACHE_MAX_GB),
+        amortized: runs at most every 16 fills, off the event loop."""
+        if not self.cfg.cache_max_bytes:
+            return
+        self._gc_countdown -= 1
+        if self._gc_countdown > 0:
+            return
+        self._gc_countdown = 16
+        from ..utils.netio import _pool
+
+        max_bytes = self.cfg.cache_max_bytes
+        _pool().submit(lambda: self.cache.gc(max_bytes))
+
     async def _serve_digests(self, head: RequestHead, writer) -> bool:
         """GET /__demodel/digests/<path> — the cache's recorded per-chunk
         sha256 digests for the entry <path> resolves to (following the
@@ -554,6 +570,7 @@ class ProxyServer:
                 cache_writer.abort()
             else:
                 cache_writer.finalize()
+                self._maybe_gc()
         # ---- hook 3: response hook ------------------------------------
         self.transfers.record(event="miss", uri=uri, status=resp.status,
                               bytes=total)

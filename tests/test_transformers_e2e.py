@@ -61,3 +61,28 @@ def test_transformers_from_pretrained_via_proxy(tiny_gpt2_dir, tmp_path,
         assert torch.equal(out1, out2)
     finally:
         stack.close()
+
+
+def test_pull_pretrained_cpu(tiny_gpt2_dir, tmp_path):
+    """pull_pretrained: pull + build + weight-load in one call (CPU
+    landing; the GPU path is the same scatter covered by gpu tests)."""
+    import torch
+
+    from demodel_amd.engine.loader import pull_pretrained
+
+    files = {p.name: str(p) for p in tiny_gpt2_dir.iterdir()}
+    stack = Stack(tmp_path)
+    try:
+        stack.origin.add_hf_repo("tiny-random/gpt2", files)
+        model, res = pull_pretrained("tiny-random/gpt2",
+                                     endpoint=stack.origin_base,
+                                     workers=2)
+        assert res.total_bytes > 0
+        ref = transformers.GPT2LMHeadModel.from_pretrained(
+            str(tiny_gpt2_dir))
+        x = torch.randint(0, 512, (1, 8))
+        with torch.no_grad():
+            assert torch.allclose(model(x).logits, ref(x).logits,
+                                  atol=1e-5)
+    finally:
+        stack.close()
