@@ -24,8 +24,12 @@ namespace {
 // Window size trades per-wave speed (bigger window = fewer far matches)
 // against occupancy (LDS-limited resident workgroups); the launcher picks
 // 16 KiB when many frames provide parallelism, 64 KiB for few frames.
+// second launch_bounds arg = waves/SIMD: LDS (window + tables) already
+// caps residency at ~5 (16K) / 2 (64K) workgroups per CU, so granting the
+// full register budget is free — without it the compiler spills the
+// sequence loop into AGPRs/scratch (1626 v_accvgpr_read in the ISA).
 template <int ZWIN>
-__global__ void __launch_bounds__(64)
+__global__ void __launch_bounds__(64, ZWIN <= 16 * 1024 ? 2 : 1)
 zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
   constexpr int ZWMASK = ZWIN - 1;
   __shared__ ZShared sh;
