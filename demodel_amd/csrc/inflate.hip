@@ -1,21 +1,22 @@
-// CDNA4 DEFLATE (RFC 1951) inflate kernel.
+// CDNA4 DEFLATE (RFC 1951) inflate kernel, v2.
 //
-// Cached proxy bodies keep their original Content-Encoding (gzip —
-// reference CONTRIBUTING.md:116), and dataset streams carry
-// deflate/gzip-compressed members; this kernel decompresses them without
-// a host round-trip.
+// Design (lessons measured on the zstd kernel, profiles/zstd_words_pmc):
 //
-// DEFLATE is serially entropy-coded, so a single stream has no internal
-// parallelism (SURVEY.md §7 hard part (a)); the design is therefore
-// wave-per-stream: lane 0 owns the bit reader and Huffman decode, and the
-// whole wave64 executes long match copies / stored-block copies in
-// parallel.  Aggregate throughput comes from decompressing many streams
-// (gzip members, parquet pages, per-file bodies) concurrently — the
-// honest decomposition for this format.
+// * wave-per-stream, but ALL 64 lanes run the bit-serial decode
+//   REDUNDANTLY in lockstep on register state — no cross-lane
+//   publication, no barriers anywhere in the symbol loop.  LDS table
+//   reads broadcast; input reads coalesce to one address.
+// * a 32 KiB LDS output window — exactly DEFLATE's maximum match
+//   distance, so EVERY back-reference is an LDS read (wave-internal
+//   lgkm ordering); output bytes stream to HBM fire-and-forget and are
+//   never read back.
+// * 10-bit primary Huffman LUTs (litlen + dist) with a canonical
+//   bit-by-bit fallback for longer codes; 4-byte batched bit refills.
+// * launch_bounds grants the register budget (LDS caps occupancy).
 //
-// Back-references read the already-written output straight from HBM/L2;
-// no LDS window.  Huffman decode is canonical count/offset bit-by-bit
-// with tables in LDS.
+// Aggregate throughput comes from many streams (gzip members, pages,
+// bodies); within a stream DEFLATE remains serially entropy-coded
+// (SURVEY.md §7 hard part (a)).
 
 #include <hip/hip_runtime.h>
 
@@ -29,13 +30,13 @@ enum {
 };
 
 struct __align__(16) InflateDesc {
-  uint64_t src;       // device ptr to raw DEFLATE stream
+  uint64_t src;
   uint64_t src_len;
-  uint64_t dst;       // device ptr to output
+  uint64_t dst;
   uint64_t dst_cap;
   uint64_t written;   // out
   int64_t status;     // out
-  uint64_t consumed;  // out: input bytes consumed (multi-member scans)
+  uint64_t consumed;  // out
   uint64_t _pad1;
 };
 
@@ -50,12 +51,28 @@ struct BitReader {
     p = s; len = n; pos = 0; buf = 0; nbits = 0;
   }
   __device__ void fill() {
+    while (nbits <= 32 && pos + 4 <= len) {
+      uint64_t v = (uint64_t)p[pos] | ((uint64_t)p[pos + 1] << 8) |
+                   ((uint64_t)p[pos + 2] << 16) |
+                   ((uint64_t)p[pos + 3] << 24);
+      buf |= v << nbits;
+      nbits += 32;
+      pos += 4;
+    }
     while (nbits <= 56) {
       uint64_t byte = pos < len ? p[pos] : 0;  // zero-pad past EOF
       buf |= byte << nbits;
       nbits += 8;
       ++pos;
     }
+  }
+  __device__ uint32_t peek(int n) {
+    if (nbits < n) fill();
+    return (uint32_t)(buf & ((1ull << n) - 1));
+  }
+  __device__ void drop(int n) {
+    buf >>= n;
+    nbits -= n;
   }
   __device__ uint32_t bits(int n) {
     if (nbits < n) fill();
@@ -66,9 +83,9 @@ struct BitReader {
   }
   __device__ uint32_t bit1() { return bits(1); }
   __device__ void align_byte() {
-    int drop = nbits & 7;
-    buf >>= drop;
-    nbits -= drop;
+    int d = nbits & 7;
+    buf >>= d;
+    nbits -= d;
   }
   __device__ uint64_t byte_pos() const {
     return pos - (uint64_t)(nbits >> 3);
@@ -76,6 +93,7 @@ struct BitReader {
   __device__ bool overran() const { return byte_pos() > len; }
 };
 
+// canonical fallback tables (also the source for LUT construction)
 struct HuffTable {
   uint16_t count[16];
   uint16_t offset[16];
@@ -83,6 +101,8 @@ struct HuffTable {
   uint16_t sym[288];
 };
 
+// all-lane redundant build: every lane computes/writes the same values
+// (same-address LDS stores are benign)
 __device__ bool huff_build(HuffTable* t, const uint8_t* lens, int n) {
   for (int i = 0; i < 16; ++i) t->count[i] = 0;
   for (int i = 0; i < n; ++i) t->count[lens[i]]++;
@@ -105,7 +125,7 @@ __device__ bool huff_build(HuffTable* t, const uint8_t* lens, int n) {
   return total > 0;
 }
 
-__device__ int huff_decode(BitReader* br, const HuffTable* t) {
+__device__ int huff_decode_slow(BitReader* br, const HuffTable* t) {
   uint32_t code = 0;
   for (int l = 1; l < 16; ++l) {
     code |= br->bit1();
@@ -114,6 +134,46 @@ __device__ int huff_decode(BitReader* br, const HuffTable* t) {
     code <<= 1;
   }
   return -1;
+}
+
+#define LUT_BITS 10
+#define LUT_SIZE (1 << LUT_BITS)
+
+// Build a primary LUT over the next LUT_BITS raw (LSB-first) bits.
+// entry = (sym << 5) | code_len; 0 => fall back to bit-by-bit.
+// Lanes cooperate: symbol loop is redundant, replica fill is striped.
+__device__ void lut_build(uint16_t* lut, const HuffTable* t,
+                          const uint8_t* lens, int n, int lane) {
+  for (int i = lane; i < LUT_SIZE; i += 64) lut[i] = 0;
+  // codes per canonical order: walk lengths
+  uint32_t code = 0;
+  for (int l = 1; l <= LUT_BITS; ++l) {
+    code = (code + t->count[l - 1]) << 1;
+    for (int k = 0; k < t->count[l]; ++k) {
+      uint32_t c = code + k;
+      uint16_t sym = t->sym[t->offset[l] + k];
+      // bit-reverse the l-bit code (stream is LSB-first)
+      uint32_t rev = __brev(c) >> (32 - l);
+      uint16_t entry = (uint16_t)((sym << 5) | l);
+      int reps = 1 << (LUT_BITS - l);
+      for (int r = lane; r < reps; r += 64)
+        lut[rev | (r << l)] = entry;
+    }
+  }
+  (void)lens;
+  (void)n;
+}
+
+__device__ __forceinline__ int huff_decode_lut(BitReader* br,
+                                               const uint16_t* lut,
+                                               const HuffTable* t) {
+  uint32_t v = br->peek(LUT_BITS);
+  uint16_t e = lut[v];
+  if (e) {
+    br->drop(e & 31);
+    return e >> 5;
+  }
+  return huff_decode_slow(br, t);  // code longer than LUT_BITS
 }
 
 __constant__ uint16_t kLenBase[29] = {
@@ -132,65 +192,70 @@ __constant__ uint8_t kDistExtra[30] = {
 __constant__ uint8_t kClOrder[19] = {
     16, 17, 18, 0, 8, 7, 9, 6, 10, 5, 11, 4, 12, 3, 13, 2, 14, 1, 15};
 
-// op kinds handed from lane 0 to the wave
-enum { OP_NONE = 0, OP_MATCH = 1, OP_STORED = 2, OP_DONE = 3, OP_ERR = 4 };
+#define DWIN 32768          // DEFLATE max distance == window size
+#define DWMASK (DWIN - 1)
 
-struct Shared {
-  HuffTable litlen;
-  HuffTable dist;
-  BitReader br;
-  uint8_t lens_buf[320];
-  uint64_t pos;        // output position
-  int stage;           // 0 = at block boundary, 1 = inside huffman block
-  int bfinal;
-  int op;
-  uint64_t op_pos;     // op-specific: match dst / stored dst
-  uint64_t op_src;     // stored: src byte offset
-  uint32_t op_len;
-  uint32_t op_dist;
-  int64_t status;
-};
+#define LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
 
-// lane-0 decode step: runs until a wave op is required; returns op kind.
-__device__ int decode_until_op(Shared* sh, const uint8_t* src,
-                               uint8_t* out, const InflateDesc* d) {
-  BitReader br = sh->br;
-  uint64_t pos = sh->pos;
-  int op = OP_NONE;
+__global__ void __launch_bounds__(64, 2)
+inflate_kernel(InflateDesc* __restrict__ descs, int n_streams) {
+  __shared__ struct {
+    HuffTable litlen_t;
+    HuffTable dist_t;
+    uint16_t litlen_lut[LUT_SIZE];
+    uint16_t dist_lut[LUT_SIZE];
+    uint8_t lens_buf[320];
+    uint8_t win[DWIN];
+  } sh;
+  int lane = threadIdx.x;
 
-  while (op == OP_NONE) {
-    if (br.overran()) { sh->status = INF_ERR_UNDERRUN; op = OP_ERR; break; }
-    if (sh->stage == 0) {
-      sh->bfinal = (int)br.bit1();
+  for (int s = blockIdx.x; s < n_streams; s += gridDim.x) {
+    InflateDesc* d = &descs[s];
+    const uint8_t* src = (const uint8_t*)d->src;
+    uint8_t* out = (uint8_t*)d->dst;
+
+    // every lane holds identical decoder state in registers
+    BitReader br;
+    br.init(src, d->src_len);
+    uint64_t pos = 0;
+    int64_t status = INF_OK;
+    bool done = false;
+
+    while (!done) {
+      if (br.overran()) { status = INF_ERR_UNDERRUN; break; }
+      uint32_t bfinal = br.bit1();
       uint32_t btype = br.bits(2);
-      if (btype == 0) {
+      if (btype == 0) {                                   // stored
         br.align_byte();
-        uint32_t len = br.bits(16);
+        uint32_t blen = br.bits(16);
         uint32_t nlen = br.bits(16);
-        if ((len ^ 0xFFFFu) != nlen) {
-          sh->status = INF_ERR_FORMAT; op = OP_ERR; break;
-        }
+        if ((blen ^ 0xFFFFu) != nlen) { status = INF_ERR_FORMAT; break; }
         uint64_t so = br.byte_pos();
-        if (so + len > br.len) {
-          sh->status = INF_ERR_UNDERRUN; op = OP_ERR; break;
+        if (so + blen > br.len) { status = INF_ERR_UNDERRUN; break; }
+        if (pos + blen > d->dst_cap) { status = INF_ERR_OVERFLOW; break; }
+        for (uint32_t i = lane; i < blen; i += 64) {
+          uint8_t v = src[so + i];
+          out[pos + i] = v;
+          sh.win[(pos + i) & DWMASK] = v;
         }
-        if (pos + len > d->dst_cap) {
-          sh->status = INF_ERR_OVERFLOW; op = OP_ERR; break;
-        }
-        br.buf = 0; br.nbits = 0; br.pos = so + len;
-        sh->op_pos = pos; sh->op_src = so; sh->op_len = len;
-        pos += len;
-        op = OP_STORED;  // stage stays 0; bfinal checked after the copy
-        break;
+        LGKM0();
+        pos += blen;
+        br.buf = 0;
+        br.nbits = 0;
+        br.pos = so + blen;
+        if (bfinal) done = true;
+        continue;
       }
-      if (btype == 3) { sh->status = INF_ERR_FORMAT; op = OP_ERR; break; }
+      if (btype == 3) { status = INF_ERR_FORMAT; break; }
+
+      // ---- table setup (redundant build; striped LUT fill) ---------
       bool okt = true;
       if (btype == 1) {
         for (int i = 0; i < 288; ++i)
-          sh->lens_buf[i] = i < 144 ? 8 : i < 256 ? 9 : i < 280 ? 7 : 8;
-        okt = huff_build(&sh->litlen, sh->lens_buf, 288);
-        for (int i = 0; i < 30; ++i) sh->lens_buf[i] = 5;
-        okt = okt && huff_build(&sh->dist, sh->lens_buf, 30);
+          sh.lens_buf[i] = i < 144 ? 8 : i < 256 ? 9 : i < 280 ? 7 : 8;
+        okt = huff_build(&sh.litlen_t, sh.lens_buf, 288);
+        for (int i = 0; i < 30; ++i) sh.lens_buf[i] = 5;
+        okt = okt && huff_build(&sh.dist_t, sh.lens_buf, 30);
       } else {
         int hlit = (int)br.bits(5) + 257;
         int hdist = (int)br.bits(5) + 1;
@@ -199,145 +264,95 @@ __device__ int decode_until_op(Shared* sh, const uint8_t* src,
         for (int i = 0; i < 19; ++i) cl_lens[i] = 0;
         for (int i = 0; i < hclen; ++i)
           cl_lens[kClOrder[i]] = (uint8_t)br.bits(3);
-        if (!huff_build(&sh->dist, cl_lens, 19)) {  // dist as CL scratch
-          sh->status = INF_ERR_FORMAT; op = OP_ERR; break;
+        if (!huff_build(&sh.dist_t, cl_lens, 19)) {  // dist as CL scratch
+          status = INF_ERR_FORMAT;
+          break;
         }
         int total = hlit + hdist;
         int i = 0;
         while (i < total) {
-          int symc = huff_decode(&br, &sh->dist);
+          int symc = huff_decode_slow(&br, &sh.dist_t);
           if (symc < 0 || br.overran()) { okt = false; break; }
           if (symc < 16) {
-            sh->lens_buf[i++] = (uint8_t)symc;
+            sh.lens_buf[i++] = (uint8_t)symc;
           } else if (symc == 16) {
             if (i == 0) { okt = false; break; }
             int rep = 3 + (int)br.bits(2);
-            uint8_t v = sh->lens_buf[i - 1];
-            while (rep-- > 0 && i < total) sh->lens_buf[i++] = v;
+            uint8_t v = sh.lens_buf[i - 1];
+            while (rep-- > 0 && i < total) sh.lens_buf[i++] = v;
           } else if (symc == 17) {
             int rep = 3 + (int)br.bits(3);
-            while (rep-- > 0 && i < total) sh->lens_buf[i++] = 0;
+            while (rep-- > 0 && i < total) sh.lens_buf[i++] = 0;
           } else {
             int rep = 11 + (int)br.bits(7);
-            while (rep-- > 0 && i < total) sh->lens_buf[i++] = 0;
+            while (rep-- > 0 && i < total) sh.lens_buf[i++] = 0;
           }
         }
         okt = okt && i == total &&
-              huff_build(&sh->litlen, sh->lens_buf, hlit) &&
-              huff_build(&sh->dist, sh->lens_buf + hlit, hdist);
+              huff_build(&sh.litlen_t, sh.lens_buf, hlit) &&
+              huff_build(&sh.dist_t, sh.lens_buf + hlit, hdist);
       }
-      if (!okt) { sh->status = INF_ERR_FORMAT; op = OP_ERR; break; }
-      sh->stage = 1;
-      continue;
-    }
+      if (!okt) { status = INF_ERR_FORMAT; break; }
+      lut_build(sh.litlen_lut, &sh.litlen_t, sh.lens_buf, 288, lane);
+      lut_build(sh.dist_lut, &sh.dist_t, sh.lens_buf, 30, lane);
+      LGKM0();
 
-    // stage 1: symbol loop
-    while (true) {
-      int sym = huff_decode(&br, &sh->litlen);
-      if (sym < 0 || br.overran()) {
-        sh->status = br.overran() ? INF_ERR_UNDERRUN : INF_ERR_FORMAT;
-        op = OP_ERR;
-        break;
-      }
-      if (sym < 256) {
-        if (pos >= d->dst_cap) {
-          sh->status = INF_ERR_OVERFLOW; op = OP_ERR; break;
+      // ---- symbol loop (all lanes lockstep, zero barriers) ---------
+      while (true) {
+        int sym = huff_decode_lut(&br, sh.litlen_lut, &sh.litlen_t);
+        if (sym < 0 || br.overran()) {
+          status = br.overran() ? INF_ERR_UNDERRUN : INF_ERR_FORMAT;
+          break;
         }
-        out[pos++] = (uint8_t)sym;
-        continue;
-      }
-      if (sym == 256) {
-        sh->stage = 0;
-        if (sh->bfinal) { sh->status = INF_OK; op = OP_DONE; }
-        break;  // back to block-boundary handling (or done)
-      }
-      sym -= 257;
-      if (sym >= 29) { sh->status = INF_ERR_FORMAT; op = OP_ERR; break; }
-      uint32_t mlen = kLenBase[sym] + br.bits(kLenExtra[sym]);
-      int dsym = huff_decode(&br, &sh->dist);
-      if (dsym < 0 || dsym >= 30) {
-        sh->status = INF_ERR_FORMAT; op = OP_ERR; break;
-      }
-      uint32_t dist = kDistBase[dsym] + br.bits(kDistExtra[dsym]);
-      if (dist > pos) { sh->status = INF_ERR_FORMAT; op = OP_ERR; break; }
-      if (pos + mlen > d->dst_cap) {
-        sh->status = INF_ERR_OVERFLOW; op = OP_ERR; break;
-      }
-      if (mlen >= 64 && dist >= 16) {
-        sh->op_pos = pos; sh->op_len = mlen; sh->op_dist = dist;
-        pos += mlen;
-        op = OP_MATCH;
-        break;
-      }
-      for (uint32_t i = 0; i < mlen; ++i, ++pos)
-        out[pos] = out[pos - dist];
-    }
-  }
-
-  sh->br = br;
-  sh->pos = pos;
-  return op;
-}
-
-__global__ void __launch_bounds__(64)
-inflate_kernel(InflateDesc* __restrict__ descs, int n_streams) {
-  __shared__ Shared sh;
-  int lane = threadIdx.x;
-
-  for (int s = blockIdx.x; s < n_streams; s += gridDim.x) {
-    InflateDesc* d = &descs[s];
-    const uint8_t* src = (const uint8_t*)d->src;
-    uint8_t* out = (uint8_t*)d->dst;
-
-    if (lane == 0) {
-      sh.br.init(src, d->src_len);
-      sh.pos = 0;
-      sh.stage = 0;
-      sh.bfinal = 0;
-      sh.status = INF_OK;
-      sh.op = OP_NONE;
-    }
-    __syncthreads();
-
-    while (true) {
-      if (lane == 0) sh.op = decode_until_op(&sh, src, out, d);
-      __syncthreads();
-      int op = sh.op;
-      if (op == OP_DONE || op == OP_ERR) break;
-      if (op == OP_MATCH) {
-        uint64_t p0 = sh.op_pos;
-        uint32_t dist = sh.op_dist, len = sh.op_len;
-        if (dist >= len) {
-          for (uint32_t i = lane; i < len; i += 64)
-            out[p0 + i] = out[p0 + i - dist];
+        if (sym < 256) {
+          if (pos >= d->dst_cap) { status = INF_ERR_OVERFLOW; break; }
+          if (lane == 0) {
+            out[pos] = (uint8_t)sym;
+            sh.win[pos & DWMASK] = (uint8_t)sym;
+          }
+          ++pos;
+          continue;
+        }
+        if (sym == 256) break;  // end of block
+        sym -= 257;
+        if (sym >= 29) { status = INF_ERR_FORMAT; break; }
+        uint32_t mlen = kLenBase[sym] + br.bits(kLenExtra[sym]);
+        int dsym = huff_decode_lut(&br, sh.dist_lut, &sh.dist_t);
+        if (dsym < 0 || dsym >= 30) { status = INF_ERR_FORMAT; break; }
+        uint32_t dist = kDistBase[dsym] + br.bits(kDistExtra[dsym]);
+        if (dist > pos) { status = INF_ERR_FORMAT; break; }
+        if (pos + mlen > d->dst_cap) { status = INF_ERR_OVERFLOW; break; }
+        // every source byte is within DWIN -> always LDS
+        LGKM0();
+        if (dist >= mlen) {
+          for (uint32_t k = lane; k < mlen; k += 64) {
+            uint8_t v = sh.win[(pos + k - dist) & DWMASK];
+            out[pos + k] = v;
+            sh.win[(pos + k) & DWMASK] = v;
+          }
         } else {
-          uint64_t copied = 0;
-          while (copied < len) {
-            uint32_t n = (uint32_t)min((uint64_t)dist, len - copied);
-            for (uint32_t i = lane; i < n; i += 64)
-              out[p0 + copied + i] = out[p0 + copied + i - dist];
-            __syncthreads();
+          uint32_t copied = 0;
+          while (copied < mlen) {
+            uint32_t n = min(dist, mlen - copied);
+            for (uint32_t k = lane; k < n; k += 64) {
+              uint8_t v = sh.win[(pos + copied + k - dist) & DWMASK];
+              out[pos + copied + k] = v;
+              sh.win[(pos + copied + k) & DWMASK] = v;
+            }
+            LGKM0();
             copied += n;
           }
         }
-      } else if (op == OP_STORED) {
-        for (uint64_t i = lane; i < sh.op_len; i += 64)
-          out[sh.op_pos + i] = src[sh.op_src + i];
-        __syncthreads();
-        if (lane == 0 && sh.bfinal) {
-          sh.status = INF_OK;
-          sh.op = OP_DONE;
-        }
-        __syncthreads();
-        if (sh.op == OP_DONE) break;
+        pos += mlen;
       }
-      __syncthreads();
+      if (status != INF_OK) break;
+      if (bfinal) done = true;
     }
 
     if (lane == 0) {
-      d->written = sh.pos;
-      d->status = sh.status;
-      d->consumed = sh.br.byte_pos();
+      d->written = pos;
+      d->status = status;
+      d->consumed = br.byte_pos();
     }
     __syncthreads();
   }
@@ -349,7 +364,7 @@ extern "C" void launch_inflate_streams(const uint64_t* desc, int n_streams,
                                        int* /*status_unused*/,
                                        hipStream_t stream) {
   if (n_streams <= 0) return;
-  int blocks = n_streams < 2048 ? n_streams : 2048;
+  int blocks = n_streams < 4096 ? n_streams : 4096;
   hipLaunchKernelGGL(inflate_kernel, dim3(blocks), dim3(64), 0, stream,
                      (InflateDesc*)desc, n_streams);
 }
