@@ -104,3 +104,39 @@ def test_gpu_pull_ollama_dequant(stack, tmp_path):
     gf = out.float().cpu()
     mask = torch.isfinite(wf)
     assert torch.allclose(gf[mask], wf[mask], rtol=1 / 64, atol=1e-3)
+
+
+def test_gpu_peer_verified_pull(stack, tmp_path):
+    """Peer-verified pull on the GPU lander: digests recorded by the
+    proxy cache (1 MiB chunks) are verified by sha256_batch at landing."""
+    _require_gpu()
+    import json
+    import time
+    import urllib.request
+
+    from demodel_amd.engine import pull as pull_mod
+
+    data = os.urandom(5 << 20)
+    p = tmp_path / "pv.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/pv", {"pv.bin": str(p)})
+    url = f"{stack.endpoint}/org/pv/resolve/main/pv.bin"
+    with urllib.request.urlopen(url, timeout=20) as r:
+        assert r.read() == data
+    # wait for the peer's async digest record
+    for _ in range(100):
+        try:
+            with urllib.request.urlopen(
+                    f"{stack.endpoint}/__demodel/digests"
+                    f"/org/pv/resolve/main/pv.bin", timeout=10) as r:
+                if json.loads(r.read()).get("chunk_sha256"):
+                    break
+        except urllib.error.HTTPError:
+            pass
+        time.sleep(0.05)
+    res = pull_mod.pull_hf("org/pv", endpoint=stack.endpoint,
+                           verify="chunked", workers=1, peer_verify=True)
+    f = res.files[0]
+    assert f.blob.device.startswith("cuda")
+    assert f.blob.verify_chunk == 1 << 20
+    assert bytes(f.blob.torch_u8().cpu().numpy().tobytes()) == data
