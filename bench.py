@@ -90,7 +90,7 @@ def main():
     ap.add_argument("--workers", type=int, default=8)
     ap.add_argument("--verify", default="chunked",
                     choices=["chunked", "digest", "gpu-digest", "off"])
-    ap.add_argument("--slab-mib", type=int, default=128)
+    ap.add_argument("--slab-mib", type=int, default=32)
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--shards", type=int, default=None,
                     help="override safetensors shard count")
