@@ -345,32 +345,37 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         uint64_t wfrom = win_from;     // uniform: only changes between ops
         uint64_t lit_used = 0;
         const uint8_t* lit_base = (const uint8_t*)sh.lit_ptr;
+        // software-pipelined: decode sequence i+1 BEFORE executing the
+        // copies of sequence i, so the decode's LDS/bit latency chain
+        // overlaps the literal run's global load latency.
+        uint32_t c_ll = 0, c_len = 0, c_dist = 0;
+        bool have = false;
         for (uint32_t i = 0; i < nseq && !err; ++i) {
           uint8_t ofc = sh.of_t.e[of_state].sym;
           uint8_t mlc = sh.ml_t.e[ml_state].sym;
           uint8_t llc = sh.ll_t.e[ll_state].sym;
           if (ofc > 31 || mlc > 52 || llc > 35) { err = 1; break; }
           uint32_t ofv = (1u << ofc) + sq.get(ofc);
-          uint32_t len = kMLBase[mlc] + sq.get(kMLExtra[mlc]);
-          uint32_t ll = kLLBase[llc] + sq.get(kLLExtra[llc]);
-          uint32_t dist;
+          uint32_t n_len = kMLBase[mlc] + sq.get(kMLExtra[mlc]);
+          uint32_t n_ll = kLLBase[llc] + sq.get(kLLExtra[llc]);
+          uint32_t n_dist;
           if (ofv > 3) {
-            dist = ofv - 3;
-            rep2 = rep1; rep1 = rep0; rep0 = dist;
+            n_dist = ofv - 3;
+            rep2 = rep1; rep1 = rep0; rep0 = n_dist;
           } else {
-            uint32_t idx = ofv + (ll == 0 ? 1 : 0);
+            uint32_t idx = ofv + (n_ll == 0 ? 1 : 0);
             if (idx == 1) {
-              dist = rep0;
+              n_dist = rep0;
             } else if (idx == 2) {
-              dist = rep1;
-              rep1 = rep0; rep0 = dist;
+              n_dist = rep1;
+              rep1 = rep0; rep0 = n_dist;
             } else if (idx == 3) {
-              dist = rep2;
-              rep2 = rep1; rep1 = rep0; rep0 = dist;
+              n_dist = rep2;
+              rep2 = rep1; rep1 = rep0; rep0 = n_dist;
             } else {
-              dist = rep0 - 1;
-              if (dist == 0 || rep0 == 0) { err = 1; break; }
-              rep2 = rep1; rep1 = rep0; rep0 = dist;
+              n_dist = rep0 - 1;
+              if (n_dist == 0 || rep0 == 0) { err = 1; break; }
+              rep2 = rep1; rep1 = rep0; rep0 = n_dist;
             }
           }
           if (i + 1 < nseq) {
@@ -381,12 +386,63 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
             of_state = sh.of_t.e[of_state].base
                        + sq.get(sh.of_t.e[of_state].nbits);
           }
-          if (lit_used + ll > sh.lit_len ||
-              pos_r + ll + (uint64_t)len > d->dst_cap ||
-              (uint64_t)dist > pos_r + ll) {
+          // execute the PREVIOUS sequence while this decode's loads land
+          if (have) {
+            uint32_t ll = c_ll, len = c_len, dist = c_dist;
+            uint64_t p0 = pos_r;
+            const uint8_t* lsrc = lit_base + lit_used;
+            for (uint32_t k = lane; k < ll; k += 64) {
+              uint8_t v = lsrc[k];
+              out[p0 + k] = v;
+              win[(p0 + k) & ZWMASK] = v;
+            }
+            uint64_t mp = p0 + ll;
+            if (dist <= ZWIN - 128 && mp - dist >= wfrom) {
+              asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+              if (dist >= len) {
+                for (uint32_t k = lane; k < len; k += 64) {
+                  uint8_t v = win[(mp + k - dist) & ZWMASK];
+                  out[mp + k] = v;
+                  win[(mp + k) & ZWMASK] = v;
+                }
+              } else {
+                uint64_t copied = 0;
+                while (copied < len) {
+                  uint32_t n = (uint32_t)min((uint64_t)dist,
+                                             (uint64_t)len - copied);
+                  for (uint32_t k = lane; k < n; k += 64) {
+                    uint8_t v = win[(mp + copied + k - dist) & ZWMASK];
+                    out[mp + copied + k] = v;
+                    win[(mp + copied + k) & ZWMASK] = v;
+                  }
+                  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+                  copied += n;
+                }
+              }
+            } else {
+              asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+              for (uint32_t k = lane; k < len; k += 64) {
+                uint8_t v = out[mp + k - dist];
+                out[mp + k] = v;
+                win[(mp + k) & ZWMASK] = v;
+              }
+            }
+            pos_r += ll + len;
+            lit_used += ll;
+          }
+          // bounds for the NEW sequence (against post-exec position)
+          if (lit_used + n_ll > sh.lit_len ||
+              pos_r + n_ll + (uint64_t)n_len > d->dst_cap ||
+              (uint64_t)n_dist > pos_r + n_ll) {
             err = 1;
             break;
           }
+          c_ll = n_ll; c_len = n_len; c_dist = n_dist;
+          have = true;
+        }
+        // drain the last decoded sequence
+        if (!err && have) {
+          uint32_t ll = c_ll, len = c_len, dist = c_dist;
           uint64_t p0 = pos_r;
           const uint8_t* lsrc = lit_base + lit_used;
           for (uint32_t k = lane; k < ll; k += 64) {
@@ -396,7 +452,6 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           }
           uint64_t mp = p0 + ll;
           if (dist <= ZWIN - 128 && mp - dist >= wfrom) {
-            // near match: source bytes live in the LDS window
             asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
             if (dist >= len) {
               for (uint32_t k = lane; k < len; k += 64) {
@@ -419,7 +474,6 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
               }
             }
           } else {
-            // far match: read old output from HBM; drain our stores first
             asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
             for (uint32_t k = lane; k < len; k += 64) {
               uint8_t v = out[mp + k - dist];
