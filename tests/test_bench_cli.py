@@ -73,3 +73,28 @@ def test_bench_torchrun_shard_mode(tmp_path):
     line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
     out = _check_line(line[0], world=2)
     assert out["scaling"] == "strong"
+
+
+def test_demodel_pull_cli(tmp_path):
+    """`demodel pull hf://...` end-to-end via subprocess."""
+    import threading
+
+    sys.path.insert(0, os.path.join(REPO, "tests"))
+    from helpers import Stack
+
+    stack = Stack(tmp_path)
+    try:
+        blob = tmp_path / "w.bin"
+        blob.write_bytes(os.urandom(50_000))
+        stack.origin.add_hf_repo("org/cli", {"w.bin": str(blob)})
+        out_dir = tmp_path / "out"
+        r = subprocess.run(
+            [sys.executable, "-m", "demodel_amd", "pull", "hf://org/cli",
+             "--endpoint", stack.origin_base, "--out", str(out_dir)],
+            cwd=REPO, capture_output=True, text=True, timeout=120)
+        assert r.returncode == 0, r.stderr[-1500:]
+        result = json.loads(r.stdout)
+        assert result["total_bytes"] == 50_000
+        assert (out_dir / "w.bin").read_bytes() == blob.read_bytes()
+    finally:
+        stack.close()

@@ -79,3 +79,24 @@ def test_export_unknown_dest_raises(xdg_tmp):
     ca_mod.read_or_new_ca(use_ecdsa=True)
     with pytest.raises(ValueError):
         ca_mod.export_ca("netscape")
+
+
+def test_export_python_ssl_capath(xdg_tmp, tmp_path):
+    """--for python-ssl drops the CA into capath WITH a hash symlink
+    (the reference forgot the c_rehash link — SURVEY §2.1)."""
+    import json as _json
+
+    ca_mod.read_or_new_ca(use_ecdsa=True)
+    capath = tmp_path / "capath"
+    capath.mkdir()
+    fake_python = tmp_path / "fakepython"
+    fake_python.write_text(
+        "#!/bin/sh\necho '%s'\n" % _json.dumps(
+            {"cafile": None, "capath": str(capath),
+             "openssl_cafile": None, "openssl_capath": None}))
+    fake_python.chmod(0o755)
+    dest = ca_mod.export_ca("python-ssl", python_exe=str(fake_python))
+    assert dest == str(capath / "demodel-ca.crt")
+    assert (capath / "demodel-ca.crt").exists()
+    links = [p for p in capath.iterdir() if p.name.endswith(".0")]
+    assert len(links) == 1 and links[0].is_symlink()
