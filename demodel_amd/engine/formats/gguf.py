@@ -314,6 +314,42 @@ def dequant_cpu(type_id: int, raw: bytes, n_elems: int):
 # ------------------------------------------------------------------ #
 # GPU dequantization from a landed blob
 
+def dequant_all_gpu(gg: GGUFModel, stream=None) -> dict:
+    """Dequantize every tensor of a landed GGUF blob -> {name: bf16
+    torch tensor}.  One output arena, all launches async on one stream
+    (avoids per-tensor allocation serialization)."""
+    import torch
+
+    from ...gpu import hip
+
+    h = hip()
+    own = stream is None
+    stream = stream or h.Stream(0)
+    quants = [t for t in gg.tensors if t.type_id in (2, 8, 12, 14)]
+    out_bytes = sum(t.n_elems * 2 for t in quants)
+    arena = h.DeviceBuffer(max(out_bytes, 1))
+    offsets = {}
+    off = 0
+    for t in quants:
+        h.gguf_dequant(t.type_id,
+                       gg.blob.buffer.ptr + gg.data_offset + t.offset,
+                       arena.ptr + off, t.n_blocks, stream.handle)
+        offsets[t.name] = off
+        off += t.n_elems * 2
+    if own:
+        stream.sync()
+    u8 = torch.from_dlpack(arena.to_dlpack())
+    out = {}
+    for t in gg.tensors:
+        if t.type_id in (2, 8, 12, 14):
+            o = offsets[t.name]
+            out[t.name] = (u8[o:o + t.n_elems * 2]
+                           .view(torch.bfloat16).view(t.dims[::-1]))
+        else:
+            out[t.name] = dequant_tensor_gpu(gg, t, stream=stream)
+    return out
+
+
 def dequant_tensor_gpu(gg: GGUFModel, t: GGUFTensor, stream=None):
     """Dequantize one tensor from the landed blob -> torch bf16 tensor
     (shape reversed from ggml dims: torch shape = dims[::-1])."""

@@ -140,3 +140,43 @@ def test_gpu_peer_verified_pull(stack, tmp_path):
     assert f.blob.device.startswith("cuda")
     assert f.blob.verify_chunk == 1 << 20
     assert bytes(f.blob.torch_u8().cpu().numpy().tobytes()) == data
+
+
+def test_gpu_dequant_all(stack, tmp_path):
+    """dequant_all_gpu: whole GGUF -> bf16 state dict in one arena."""
+    _require_gpu()
+    import numpy as np
+    import torch
+
+    from demodel_amd.engine import pull as pull_mod
+    from demodel_amd.engine.formats import gguf
+
+    gg_path = tmp_path / "all.gguf"
+    gguf.build_file(str(gg_path), [
+        ("a.weight", (64, 4), 2),      # q4_0
+        ("b.weight", (256, 2), 12),    # q4_K
+        ("c.weight", (256, 2), 14),    # q6_K
+        ("n.weight", (32,), 0),        # f32
+    ])
+    stack.origin.add_ollama_model("library/da", "latest", [
+        ("application/vnd.ollama.image.model", str(gg_path)),
+    ])
+    res = pull_mod.pull_ollama("da", "latest", endpoint=stack.origin_base,
+                               verify="chunked", workers=1)
+    gg = res.meta["gguf_model"]
+    sd = gguf.dequant_all_gpu(gg)
+    assert set(sd) == {"a.weight", "b.weight", "c.weight", "n.weight"}
+    raw = gg_path.read_bytes()
+    for t in gg.tensors:
+        got = sd[t.name].float().cpu().reshape(-1)
+        want = torch.from_numpy(gguf.dequant_cpu(
+            t.type_id, raw[gg.data_offset + t.offset:
+                           gg.data_offset + t.offset + t.nbytes],
+            t.n_elems))
+        if t.type_id in (2, 8, 12, 14):
+            want = want.to(torch.bfloat16).float()
+        else:
+            want = want.to(torch.bfloat16).float()
+        mask = torch.isfinite(want)
+        assert torch.allclose(got[mask], want[mask], rtol=1 / 64,
+                              atol=1e-3), t.name
