@@ -15,7 +15,7 @@ from __future__ import annotations
 
 import os
 import struct
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 MAGIC = 0x46554747  # 'GGUF' little-endian
 

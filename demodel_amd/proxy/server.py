@@ -641,7 +641,6 @@ class ProxyServer:
 
 
 async def run_proxy(cfg: Config) -> None:
-    from . import server as _self  # noqa: F401
     from ..ca import read_or_new_ca
 
     ca = read_or_new_ca(cfg.ca_use_ecdsa)
