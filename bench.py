@@ -91,6 +91,7 @@ def main():
     ap.add_argument("--verify", default="chunked",
                     choices=["chunked", "digest", "gpu-digest", "off"])
     ap.add_argument("--slab-mib", type=int, default=32)
+    ap.add_argument("--n-slabs", type=int, default=4)
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--shards", type=int, default=None,
                     help="override safetensors shard count")
@@ -161,7 +162,8 @@ def main():
     endpoint = f"http://127.0.0.1:{port}"
 
     landers = LanderPool(local_rank if have_gpu else 0,
-                         slab_bytes=args.slab_mib << 20)
+                         slab_bytes=args.slab_mib << 20,
+                         n_slabs=args.n_slabs)
 
     def sync():
         if have_gpu:

@@ -21,6 +21,54 @@ namespace {
 // slot p & (ZWIN-1) holds output byte p; every producer (literals,
 // matches, raw/RLE blocks) maintains it.  Matches farther back than the
 // window take a rare global-read path behind an explicit vmcnt drain.
+
+// one sequence's copies: literal run from the (global) literal buffer,
+// then the match via the LDS window (or the rare far-global path)
+template <int ZWIN>
+__device__ __forceinline__ void exec_seq(
+    uint8_t* __restrict__ out, uint8_t* win, uint64_t wfrom,
+    const uint8_t* lit_src, uint64_t p0, uint32_t ll, uint32_t len,
+    uint32_t dist, int lane) {
+  constexpr int ZWMASK = ZWIN - 1;
+  for (uint32_t k = lane; k < ll; k += 64) {
+    uint8_t v = lit_src[k];
+    out[p0 + k] = v;
+    win[(p0 + k) & ZWMASK] = v;
+  }
+  uint64_t mp = p0 + ll;
+  if (dist <= ZWIN - 128 && mp - dist >= wfrom) {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    if (dist >= len) {
+      for (uint32_t k = lane; k < len; k += 64) {
+        uint8_t v = win[(mp + k - dist) & ZWMASK];
+        out[mp + k] = v;
+        win[(mp + k) & ZWMASK] = v;
+      }
+    } else {
+      uint64_t copied = 0;
+      while (copied < len) {
+        uint32_t n = (uint32_t)min((uint64_t)dist,
+                                   (uint64_t)len - copied);
+        for (uint32_t k = lane; k < n; k += 64) {
+          uint8_t v = win[(mp + copied + k - dist) & ZWMASK];
+          out[mp + copied + k] = v;
+          win[(mp + copied + k) & ZWMASK] = v;
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        copied += n;
+      }
+    }
+  } else {
+    // far match: read old output from HBM; drain our stores first
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    for (uint32_t k = lane; k < len; k += 64) {
+      uint8_t v = out[mp + k - dist];
+      out[mp + k] = v;
+      win[(mp + k) & ZWMASK] = v;
+    }
+  }
+}
+
 // Window size trades per-wave speed (bigger window = fewer far matches)
 // against occupancy (LDS-limited resident workgroups); the launcher picks
 // 16 KiB when many frames provide parallelism, 64 KiB for few frames.
@@ -388,47 +436,10 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           }
           // execute the PREVIOUS sequence while this decode's loads land
           if (have) {
-            uint32_t ll = c_ll, len = c_len, dist = c_dist;
-            uint64_t p0 = pos_r;
-            const uint8_t* lsrc = lit_base + lit_used;
-            for (uint32_t k = lane; k < ll; k += 64) {
-              uint8_t v = lsrc[k];
-              out[p0 + k] = v;
-              win[(p0 + k) & ZWMASK] = v;
-            }
-            uint64_t mp = p0 + ll;
-            if (dist <= ZWIN - 128 && mp - dist >= wfrom) {
-              asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-              if (dist >= len) {
-                for (uint32_t k = lane; k < len; k += 64) {
-                  uint8_t v = win[(mp + k - dist) & ZWMASK];
-                  out[mp + k] = v;
-                  win[(mp + k) & ZWMASK] = v;
-                }
-              } else {
-                uint64_t copied = 0;
-                while (copied < len) {
-                  uint32_t n = (uint32_t)min((uint64_t)dist,
-                                             (uint64_t)len - copied);
-                  for (uint32_t k = lane; k < n; k += 64) {
-                    uint8_t v = win[(mp + copied + k - dist) & ZWMASK];
-                    out[mp + copied + k] = v;
-                    win[(mp + copied + k) & ZWMASK] = v;
-                  }
-                  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-                  copied += n;
-                }
-              }
-            } else {
-              asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-              for (uint32_t k = lane; k < len; k += 64) {
-                uint8_t v = out[mp + k - dist];
-                out[mp + k] = v;
-                win[(mp + k) & ZWMASK] = v;
-              }
-            }
-            pos_r += ll + len;
-            lit_used += ll;
+            exec_seq<ZWIN>(out, win, wfrom, lit_base + lit_used, pos_r,
+                           c_ll, c_len, c_dist, lane);
+            pos_r += c_ll + c_len;
+            lit_used += c_ll;
           }
           // bounds for the NEW sequence (against post-exec position)
           if (lit_used + n_ll > sh.lit_len ||
@@ -442,47 +453,10 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         }
         // drain the last decoded sequence
         if (!err && have) {
-          uint32_t ll = c_ll, len = c_len, dist = c_dist;
-          uint64_t p0 = pos_r;
-          const uint8_t* lsrc = lit_base + lit_used;
-          for (uint32_t k = lane; k < ll; k += 64) {
-            uint8_t v = lsrc[k];
-            out[p0 + k] = v;
-            win[(p0 + k) & ZWMASK] = v;
-          }
-          uint64_t mp = p0 + ll;
-          if (dist <= ZWIN - 128 && mp - dist >= wfrom) {
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-            if (dist >= len) {
-              for (uint32_t k = lane; k < len; k += 64) {
-                uint8_t v = win[(mp + k - dist) & ZWMASK];
-                out[mp + k] = v;
-                win[(mp + k) & ZWMASK] = v;
-              }
-            } else {
-              uint64_t copied = 0;
-              while (copied < len) {
-                uint32_t n = (uint32_t)min((uint64_t)dist,
-                                           (uint64_t)len - copied);
-                for (uint32_t k = lane; k < n; k += 64) {
-                  uint8_t v = win[(mp + copied + k - dist) & ZWMASK];
-                  out[mp + copied + k] = v;
-                  win[(mp + copied + k) & ZWMASK] = v;
-                }
-                asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-                copied += n;
-              }
-            }
-          } else {
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            for (uint32_t k = lane; k < len; k += 64) {
-              uint8_t v = out[mp + k - dist];
-              out[mp + k] = v;
-              win[(mp + k) & ZWMASK] = v;
-            }
-          }
-          pos_r += ll + len;
-          lit_used += ll;
+          exec_seq<ZWIN>(out, win, wfrom, lit_base + lit_used, pos_r,
+                         c_ll, c_len, c_dist, lane);
+          pos_r += c_ll + c_len;
+          lit_used += c_ll;
         }
         if (lane == 0) {
           seq_err = err;

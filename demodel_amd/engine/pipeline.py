@@ -422,4 +422,4 @@ def make_lander(device_index: int = 0, **kw):
         return Lander(device_index=device_index, **kw)
     return HostLander(**{k: v for k, v in kw.items()
                          if k in ("slab_bytes", "verify_chunk",
-                                  "head_bytes")})
+                                  "head_bytes")})  # n_slabs: GPU-only

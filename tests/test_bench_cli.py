@@ -77,8 +77,6 @@ def test_bench_torchrun_shard_mode(tmp_path):
 
 def test_demodel_pull_cli(tmp_path):
     """`demodel pull hf://...` end-to-end via subprocess."""
-    import threading
-
     sys.path.insert(0, os.path.join(REPO, "tests"))
     from helpers import Stack
 
