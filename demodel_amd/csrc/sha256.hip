@@ -216,6 +216,10 @@ sha256_chain_update_kernel(uint32_t* __restrict__ state,
       for (int t = 0; t < 64; ++t) kwreg[t] = kw[b][t];
       uint32_t a = s[0], bb = s[1], c = s[2], d = s[3];
       uint32_t e = s[4], f = s[5], g = s[6], h = s[7];
+      // all lanes compute identical values, so the compiler scalarizes
+      // the rounds onto the SALU — lengthening the dependence chain with
+      // cross-unit hops.  Pin the two chain registers to VGPRs.
+      asm("" : "+v"(a), "+v"(e));
 #pragma unroll
       for (int t = 0; t < 64; ++t) {
         uint32_t S1 = rotr(e, 6) ^ rotr(e, 11) ^ rotr(e, 25);
