@@ -219,7 +219,8 @@ sha256_chain_update_kernel(uint32_t* __restrict__ state,
       // all lanes compute identical values, so the compiler scalarizes
       // the rounds onto the SALU — lengthening the dependence chain with
       // cross-unit hops.  Pin the two chain registers to VGPRs.
-      asm("" : "+v"(a), "+v"(e));
+      asm("" : "+v"(a), "+v"(bb), "+v"(c), "+v"(d),
+              "+v"(e), "+v"(f), "+v"(g), "+v"(h));
 #pragma unroll
       for (int t = 0; t < 64; ++t) {
         uint32_t S1 = rotr(e, 6) ^ rotr(e, 11) ^ rotr(e, 25);
