@@ -21,3 +21,20 @@ see cmd/demodel/main.go:59) rebuilt from scratch MI355X-first:
 __version__ = "0.1.0"
 
 from .config import Config, load_config  # noqa: F401
+
+
+def __getattr__(name):
+    # convenience re-exports without forcing heavy imports at package load
+    if name in ("pull_hf", "pull_ollama", "pull_spec"):
+        from .engine import pull as _p
+
+        return getattr(_p, name)
+    if name == "stream_dataset":
+        from .engine.datasets import stream_dataset
+
+        return stream_dataset
+    if name in ("pull_pretrained", "load_into", "load_model_from_pull"):
+        from .engine import loader as _l
+
+        return getattr(_l, name)
+    raise AttributeError(name)
