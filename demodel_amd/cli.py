@@ -11,8 +11,8 @@ Parity with the reference CLI (cmd/demodel/main.go:56-81):
 New subcommands beyond the reference:
 
 * ``demodel pull <spec>``  — pull a model/dataset directly (no client
-  needed): hf://org/repo[@rev] or ollama://name[:tag], with optional GPU
-  landing (--gpu / --gpus).
+  needed): hf://org/repo[@rev] or ollama://name[:tag], with GPU landing
+  auto-selected (force with --gpu / --cpu).
 * ``demodel verify <spec>`` — re-verify cached blobs against digests.
 """
 
@@ -72,7 +72,9 @@ def _cmd_pull(args) -> int:
         print("error: --gpu requested but no AMD GPU is available",
               file=sys.stderr)
         return 1
-    result = pull_spec(args.spec, endpoint=args.endpoint, out_dir=args.out)
+    gpu = True if args.gpu else (False if args.cpu else None)
+    result = pull_spec(args.spec, endpoint=args.endpoint, out_dir=args.out,
+                       gpu=gpu)
     print(json.dumps(result, indent=2, default=str))
     return 0
 
@@ -122,7 +124,10 @@ def main(argv: list[str] | None = None) -> int:
     pp.add_argument("--endpoint", default=None,
                     help="upstream endpoint override (e.g. a fake origin)")
     pp.add_argument("--gpu", action="store_true",
-                    help="land blobs in GPU HBM via the HIP pipeline")
+                    help="land blobs in GPU HBM via the HIP pipeline "
+                         "(default: auto)")
+    pp.add_argument("--cpu", action="store_true",
+                    help="force host-RAM landing even with a GPU present")
     pp.add_argument("--out", default=None, help="materialize files here")
     pp.set_defaults(fn=_cmd_pull)
 

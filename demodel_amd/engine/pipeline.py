@@ -417,8 +417,14 @@ class HostLander:
         return blob
 
 
-def make_lander(device_index: int = 0, **kw):
-    if have_gpu():
+def make_lander(device_index: int = 0, gpu: bool | None = None, **kw):
+    """gpu=True forces HBM landing (raises without a device), gpu=False
+    forces host RAM, None auto-selects."""
+    if gpu is None:
+        gpu = have_gpu()
+    elif gpu and not have_gpu():
+        raise RuntimeError("GPU landing requested but no AMD GPU present")
+    if gpu:
         return Lander(device_index=device_index, **kw)
     return HostLander(**{k: v for k, v in kw.items()
                          if k in ("slab_bytes", "verify_chunk",

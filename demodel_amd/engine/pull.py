@@ -502,7 +502,11 @@ def pull_ollama(name: str, tag: str = "latest",
 
 def pull_spec(spec: str, cfg: Config | None = None, endpoint=None,
               gpu: bool | None = None, out_dir=None, **kw) -> dict:
-    """CLI entry: parse hf://org/repo[@rev] or ollama://name[:tag]."""
+    """CLI entry: parse hf://org/repo[@rev] or ollama://name[:tag].
+
+    gpu=True forces HBM landing, gpu=False host RAM, None auto."""
+    if gpu is not None:
+        kw.setdefault("landers", LanderPool(0, gpu=gpu))
     if spec.startswith("hf://"):
         body = spec[len("hf://"):]
         repo, _, rev = body.partition("@")

@@ -88,7 +88,8 @@ def test_demodel_pull_cli(tmp_path):
         out_dir = tmp_path / "out"
         r = subprocess.run(
             [sys.executable, "-m", "demodel_amd", "pull", "hf://org/cli",
-             "--endpoint", stack.origin_base, "--out", str(out_dir)],
+             "--cpu", "--endpoint", stack.origin_base,
+             "--out", str(out_dir)],
             cwd=REPO, capture_output=True, text=True, timeout=120)
         assert r.returncode == 0, r.stderr[-1500:]
         result = json.loads(r.stdout)
