@@ -317,31 +317,73 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     ] if n_segs > 1 else [0, total]
     assert bounds[-1] == total, bounds
 
+    from .pipeline import LandingError
+
     def land_range(i):
         lo, hi = bounds[i], bounds[i + 1]
         if hi <= lo:
             return 0.0
-        rng = {"Range": f"bytes={lo}-{hi - 1}"}
-        if headers:
-            rng.update(headers)
-        s = fetch.http_get(url, cafile=cafile, insecure=insecure,
-                           headers=rng)
-        try:
-            if s.status != 206:
-                raise fetch.FetchError(
-                    f"range GET {url} [{lo},{hi}) -> {s.status}")
-            lander = landers.get()
-            lander.land_into(buf, lo, s.fill, hi - lo, file_size=total)
-            lander.sync()
-        finally:
-            s.close()
+        lander = landers.get()
+        at = lo
+        attempt = 0
+        while at < hi:
+            rng = {"Range": f"bytes={at}-{hi - 1}"}
+            if headers:
+                rng.update(headers)
+            s = fetch.http_get(url, cafile=cafile, insecure=insecure,
+                               headers=rng)
+            try:
+                if s.status != 206:
+                    raise fetch.FetchError(
+                        f"range GET {url} [{at},{hi}) -> {s.status}")
+                lander.land_into(buf, at, s.fill, hi - at, file_size=total)
+                at = hi
+            except LandingError as e:
+                at += e.landed
+                attempt += 1
+                if attempt > RESUME_RETRIES:
+                    raise IOError(
+                        f"segment [{lo},{hi}) failed after {attempt} "
+                        f"attempts at byte {at}") from e
+                log.info("resuming segment [%d,%d) at byte %d (attempt "
+                         "%d)", lo, hi, at, attempt)
+            finally:
+                s.close()
+        lander.sync()
         return 0.0
 
     futs = [seg_executor.submit(land_range, i) for i in range(1, n_segs)]
     # this thread lands segment 0 from the open stream (it was asked for
     # [0, SEGMENT_MIN) which equals bounds[1] when total >= SEGMENT_MIN)
-    head, _ = lander0.land_into(buf, 0, src0.fill, bounds[1],
-                                file_size=total, keep_head=True)
+    try:
+        head, _ = lander0.land_into(buf, 0, src0.fill, bounds[1],
+                                    file_size=total, keep_head=True)
+    except LandingError as e:
+        # the probe stream died mid-segment-0: finish it with range GETs
+        head = bytearray()
+        at = e.landed
+        attempt = 0
+        while at < bounds[1]:
+            rng = {"Range": f"bytes={at}-{bounds[1] - 1}"}
+            if headers:
+                rng.update(headers)
+            s = fetch.http_get(url, cafile=cafile, insecure=insecure,
+                               headers=rng)
+            try:
+                if s.status != 206:
+                    raise fetch.FetchError(
+                        f"range GET {url} [{at},{bounds[1]}) -> "
+                        f"{s.status}")
+                lander0.land_into(buf, at, s.fill, bounds[1] - at,
+                                  file_size=total)
+                at = bounds[1]
+            except LandingError as e2:
+                at += e2.landed
+                attempt += 1
+                if attempt > RESUME_RETRIES:
+                    raise
+            finally:
+                s.close()
     lander0.sync()
     for f in futs:
         f.result()

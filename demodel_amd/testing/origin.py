@@ -350,11 +350,6 @@ class FakeOrigin:
 
     async def _serve_file(self, writer, req: RequestHead, path: str,
                           etag: str, extra: list[tuple[str, str]] | None = None):
-        drop_at = None
-        for key in list(self.drop_once):
-            if key in req.target:
-                drop_at = self.drop_once.pop(key)
-                break
         size = os.path.getsize(path)
         start, end = 0, size - 1
         status, reason = 200, "OK"
@@ -368,6 +363,16 @@ class FakeOrigin:
             else:  # suffix range
                 start = max(0, size - int(e))
             status, reason = 206, "Partial Content"
+        drop_at = None
+        for key in list(self.drop_once):
+            # key "name" fires on the first request for name; key
+            # "name@OFF" only on a request whose Range starts at OFF
+            # (targets one segment of a range-parallel pull)
+            name, _, want_start = key.partition("@")
+            if name in req.target and \
+                    (not want_start or int(want_start) == start):
+                drop_at = self.drop_once.pop(key)
+                break
         length = end - start + 1
         headers = [("Content-Type", "application/octet-stream"),
                    ("Content-Length", str(length)),

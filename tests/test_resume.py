@@ -81,3 +81,35 @@ def test_proxy_serves_range_from_cache(stack, tmp_path):
     with urllib.request.urlopen(req, timeout=20) as r:
         assert r.status == 206
         assert r.read() == data[-100:]
+
+
+@pytest.mark.gpu
+class TestSegmentedResume:
+    """Range-parallel pulls survive drops in any segment."""
+
+    def _pull(self, stack, tmp_path, monkeypatch, drop_key, drop_at):
+        import demodel_amd.engine.pull as pm
+
+        monkeypatch.setattr(pm, "SEGMENT_MIN", 1 << 20)
+        data = os.urandom(5 << 20)
+        p = tmp_path / "seg.bin"
+        p.write_bytes(data)
+        stack.origin.add_hf_repo("org/seg", {"seg.bin": str(p)})
+        stack.origin.drop_once[drop_key] = drop_at
+        res = pull_mod.pull_hf("org/seg", endpoint=stack.origin_base,
+                               verify="chunked", workers=2)
+        f = [x for x in res.files if x.name == "seg.bin"][0]
+        assert f.nbytes == len(data)
+        assert not stack.origin.drop_once  # fault fired
+        got = bytes(f.blob.torch_u8().cpu().numpy().tobytes())
+        assert got == data
+        del res, f
+
+    def test_drop_in_segment0(self, stack, tmp_path, monkeypatch):
+        # the probe stream (segment 0) dies after 300 KiB
+        self._pull(stack, tmp_path, monkeypatch, "seg.bin", 300 << 10)
+
+    def test_drop_in_later_segment(self, stack, tmp_path, monkeypatch):
+        # the range GET for the segment starting at 2 MiB dies early
+        self._pull(stack, tmp_path, monkeypatch,
+                   f"seg.bin@{2 << 20}", 100 << 10)
