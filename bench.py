@@ -41,10 +41,10 @@ def log(msg):
 class _FanoutResult:
     """Minimal result shim for fan-out steps (log fields only)."""
 
-    def __init__(self, nbytes):
+    def __init__(self, nbytes, seconds=0.0):
         self.total_bytes = nbytes
-        self.seconds_to_ready = 0.0
-        self.gbps = 0.0
+        self.seconds_to_ready = seconds
+        self.gbps = nbytes / max(seconds, 1e-9) / 1e9 if seconds else 0.0
 
 
 TINY_GEOM = {"hidden": 256, "inter": 688, "layers": 4, "heads": 8,
@@ -228,103 +228,36 @@ def main():
         res.meta["dequant_bf16_bytes"] = out_bytes
         return res, len(gg.tensors)
 
-    def parquet_step(record_digests=False):
-        """Real parquet streaming (config 5): land shards, read footers +
-        page headers back from HBM, GPU-decompress every ZSTD page into
-        the ring."""
-        from demodel_amd.engine.formats import parquet as pqf
-        from demodel_amd.engine.formats.compress import zstd_gpu
-        from demodel_amd.gpu import have_gpu as _hg, hip
+    def _stream_step(record_digests, patterns):
+        """config 5 via the user-facing stream_dataset API: each landed
+        shard's GPU decompression overlaps with the remaining pulls;
+        counts DECOMPRESSED bytes landed in HBM rings."""
+        from demodel_amd.engine.datasets import stream_dataset
 
-        res = pull_mod.pull_hf(
-            "bench/model", endpoint=endpoint, workers=args.workers,
-            verify=args.verify, landers=landers,
-            digest_map=digest_map or None)
-        if record_digests:
-            for f in res.files:
-                if f.blob.digest_blob:
-                    digest_map[f.name] = f.blob.digest_blob
-        decompressed = 0
-        if _hg():
-            h = hip()
-            frames = []
-            plan = []
-            total_out = 0
-            for f in res.files:
-                pages = pqf.blob_pages(f.blob)
-                for codec, p in pages:
-                    assert codec == pqf.CODEC_ZSTD, codec
-                    plan.append((f.blob, p))
-                    total_out += p.uncomp_size
-            ring = h.DeviceBuffer(total_out)
-            off = 0
-            for blob, p in plan:
-                frames.append((blob.buffer.ptr + p.comp_offset,
-                               p.comp_size, ring.ptr + off, p.uncomp_size))
-                off += p.uncomp_size
-            results = zstd_gpu(frames)
-            for i, r in enumerate(results):
-                assert r.ok and r.written == plan[i][1].uncomp_size, (i, r)
-            decompressed = total_out
-            del ring
-        else:
-            decompressed = res.total_bytes
+        def on_file(f):
+            if record_digests and f.blob.digest_blob:
+                digest_map[f.name] = f.blob.digest_blob
+
+        t0 = time.perf_counter()
+        out_bytes = 0
+        n = 0
+        for b in stream_dataset("bench/model", endpoint=endpoint,
+                                workers=args.workers, verify=args.verify,
+                                landers=landers, patterns=patterns,
+                                digest_map=digest_map or None,
+                                on_file=on_file, inflight=64):
+            out_bytes += int(b.data.numel())
+            n += 1
+            del b
         if have_gpu:
             torch.cuda.synchronize()
-        res.meta["decompressed_bytes"] = decompressed
-        res.total_bytes = decompressed
-        return res, len(res.files)
+        return _FanoutResult(out_bytes, time.perf_counter() - t0), n
+
+    def parquet_step(record_digests=False):
+        return _stream_step(record_digests, ("*.parquet",))
 
     def dataset_step(record_digests=False):
-        """Dataset streaming (config 5): pull zstd shards, decompress all
-        frames wave-parallel into an HBM ring; counts DECOMPRESSED bytes
-        landed."""
-        import json as _json
-
-        from demodel_amd.engine.formats.compress import zstd_gpu
-        from demodel_amd.gpu import have_gpu as _hg, hip
-
-        res = pull_mod.pull_hf(
-            "bench/model", endpoint=endpoint, workers=args.workers,
-            verify=args.verify, landers=landers,
-            digest_map=digest_map or None)
-        if record_digests:
-            for f in res.files:
-                digest_map[f.name] = f.blob.digest_blob
-        byname = {f.name: f for f in res.files}
-        decompressed = 0
-        if _hg():
-            h = hip()
-            frames = []
-            total_out = 0
-            plan = []
-            for name, f in byname.items():
-                if not name.endswith(".zst"):
-                    continue
-                idx = _json.loads(
-                    bytes(byname[name + ".idx.json"].blob.head[
-                        :byname[name + ".idx.json"].blob.nbytes]))
-                for fr in idx["frames"]:
-                    plan.append((f.blob.buffer.ptr + fr["offset"],
-                                 fr["compressed"], fr["decompressed"]))
-                    total_out += fr["decompressed"]
-            ring = h.DeviceBuffer(total_out)
-            off = 0
-            for src, clen, dlen in plan:
-                frames.append((src, clen, ring.ptr + off, dlen))
-                off += dlen
-            results = zstd_gpu(frames)
-            for i, r in enumerate(results):
-                assert r.ok and r.written == plan[i][2], (i, r)
-            decompressed = total_out
-            del ring
-        else:
-            decompressed = res.total_bytes  # CPU plumbing mode
-        if have_gpu:
-            torch.cuda.synchronize()
-        res.meta["decompressed_bytes"] = decompressed
-        res.total_bytes = decompressed  # the bytes that LAND in the ring
-        return res, len(byname)
+        return _stream_step(record_digests, ("*.zst", "*.idx.json"))
 
     file_sizes = [(n, os.path.getsize(p) if p else sizes[n])
                   for n, p in sorted(files.items())]

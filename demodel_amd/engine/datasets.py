@@ -5,6 +5,14 @@ or .zst frame shards stream through the landing pipeline and decompress
 wave-parallel into HBM rings; the iterator yields one shard at a time so
 a training/ingest loop can consume while later shards still pull.
 
+Overlap is two-level: the pull runs on worker threads (pull_hf_stream)
+while this thread LAUNCHES each completed shard's decompression async on
+its own HIP stream (ZstdJob), so downloads, several shards' decodes, and
+the consumer all run concurrently.  Per-shard frame counts are far below
+the chip's wave capacity (256 frames vs 2048 wave slots), so concurrent
+per-shard launches are what keep the zstd kernel occupancy at the level
+a single whole-dataset launch would get.
+
 On CPU-only machines the same API decompresses with pyarrow, so the data
 path is testable anywhere.
 """
@@ -12,10 +20,11 @@ path is testable anywhere.
 from __future__ import annotations
 
 import json
+from collections import deque
 from dataclasses import dataclass
 
 from ..utils.log import get_logger
-from .pull import LanderPool, pull_hf
+from .pull import LanderPool
 
 log = get_logger("datasets")
 
@@ -30,21 +39,21 @@ class ShardBatch:
         return [self.data[o:o + n] for o, n in self.spans]
 
 
-def _decompress_parquet_gpu(blob):
+def _launch_parquet_gpu(blob):
     from .formats import parquet as pqf
 
     pages = pqf.blob_pages(blob)
-    ring, spans = pqf.decompress_pages_gpu(blob, pages)
+    job, ring, spans = pqf.launch_pages_gpu(blob, pages)
     import torch
 
-    return torch.from_dlpack(ring.to_dlpack()), spans
+    return job, torch.from_dlpack(ring.to_dlpack()), spans
 
 
-def _decompress_zst_frames_gpu(blob, idx):
+def _launch_zst_frames_gpu(blob, idx):
     import torch
 
     from ..gpu import hip
-    from .formats.compress import zstd_gpu
+    from .formats.compress import ZstdJob
 
     h = hip()
     total = sum(fr["decompressed"] for fr in idx["frames"])
@@ -57,11 +66,8 @@ def _decompress_zst_frames_gpu(blob, idx):
                        ring.ptr + off, fr["decompressed"]))
         spans.append((off, fr["decompressed"]))
         off += fr["decompressed"]
-    results = zstd_gpu(frames)
-    bad = [(i, r) for i, r in enumerate(results) if not r.ok]
-    if bad:
-        raise IOError(f"GPU frame decompress failed: {bad[:3]}")
-    return torch.from_dlpack(ring.to_dlpack()), spans
+    job = ZstdJob(frames)
+    return job, torch.from_dlpack(ring.to_dlpack()), spans
 
 
 def _decompress_cpu(blob, idx):
@@ -81,47 +87,93 @@ def _decompress_cpu(blob, idx):
     return torch.frombuffer(bytearray(out), dtype=torch.uint8), spans
 
 
+def _sidecar_idx(sidecar):
+    raw = (bytes(sidecar.blob.buffer) if sidecar.blob.device == "cpu"
+           else bytes(sidecar.blob.head[:sidecar.blob.nbytes]))
+    return json.loads(raw)
+
+
 def stream_dataset(repo: str, endpoint: str | None = None,
                    patterns: tuple = ("*.parquet", "*.zst", "*.idx.json"),
                    device_index: int = 0, workers: int = 4,
                    verify: str = "chunked",
-                   landers: LanderPool | None = None):
+                   landers: LanderPool | None = None,
+                   digest_map: dict | None = None,
+                   on_file=None, inflight: int = 4):
     """Yield ShardBatch per data shard of an HF dataset repo.
 
-    Pull happens up-front (concurrent, verified); decompression runs
-    shard-by-shard as the iterator advances, so HBM holds one
-    decompressed ring at a time plus the compressed blobs.
-    """
+    inflight bounds how many decompressed shard rings can be in flight
+    (launched but not yet yielded) at once — the HBM high-water mark is
+    roughly (inflight + 1) decompressed shards plus the compressed
+    blobs.  on_file(pulled_file) fires per landed file (digest capture,
+    progress)."""
     from ..gpu import have_gpu
 
-    res = pull_hf(repo, endpoint=endpoint, workers=workers,
-                  verify=verify, patterns=list(patterns),
-                  device_index=device_index, landers=landers)
-    byname = {f.name: f for f in res.files}
-    for name in sorted(byname):
-        f = byname[name]
-        if name.endswith(".idx.json"):
-            continue
-        if name.endswith(".parquet"):
-            if not have_gpu():
+    from .pull import pull_hf_stream
+
+    _, names, gen = pull_hf_stream(
+        repo, endpoint=endpoint, workers=workers, verify=verify,
+        patterns=list(patterns), device_index=device_index,
+        landers=landers, digest_map=digest_map)
+    expected = set(names)
+    gpu = have_gpu()
+
+    def launch(f):
+        """-> (job, data, spans); job None on the CPU path."""
+        if f.name.endswith(".parquet"):
+            if not gpu:
                 raise RuntimeError(
                     "parquet streaming needs a GPU (CPU fallback covers "
                     ".zst shards)")
-            data, spans = _decompress_parquet_gpu(f.blob)
-        elif name.endswith(".zst"):
-            sidecar = byname.get(name + ".idx.json")
-            if sidecar is None:
-                raise FileNotFoundError(f"{name}: missing .idx.json")
-            raw = (bytes(sidecar.blob.buffer)
-                   if sidecar.blob.device == "cpu"
-                   else bytes(sidecar.blob.head[:sidecar.blob.nbytes]))
-            idx = json.loads(raw)
-            if have_gpu():
-                data, spans = _decompress_zst_frames_gpu(f.blob, idx)
-            else:
-                data, spans = _decompress_cpu(f.blob, idx)
-        else:
-            continue
+            return _launch_parquet_gpu(f.blob)
+        idx = _sidecar_idx(f.sidecar)
+        if gpu:
+            return _launch_zst_frames_gpu(f.blob, idx)
+        data, spans = _decompress_cpu(f.blob, idx)
+        return None, data, spans
+
+    def finish(item):
+        shard, job, data, spans = item
+        if job is not None:
+            results = job.wait()
+            bad = [(i, r) for i, r in enumerate(results) if not r.ok]
+            if bad:
+                raise IOError(
+                    f"GPU decompress of {shard.name} failed: {bad[:3]}")
         log.info("dataset shard %s: %d spans, %d bytes decompressed",
-                 name, len(spans), int(data.numel()))
-        yield ShardBatch(name=name, data=data, spans=spans)
+                 shard.name, len(spans), int(data.numel()))
+        return ShardBatch(name=shard.name, data=data, spans=spans)
+
+    # pending holds the PulledFile too: its HBM blob must stay alive
+    # until the decode kernel reading it has finished
+    pending: deque = deque()
+    byname: dict = {}
+    for f in gen:
+        if on_file is not None:
+            on_file(f)
+        byname[f.name] = f
+        ready = []
+        if f.name.endswith(".parquet"):
+            ready.append(f)
+        elif f.name.endswith(".zst"):
+            if f.name + ".idx.json" not in expected:
+                raise FileNotFoundError(f"{f.name}: missing .idx.json")
+            if f.name + ".idx.json" in byname:
+                f.sidecar = byname[f.name + ".idx.json"]
+                ready.append(f)
+        elif f.name.endswith(".zst.idx.json"):
+            shard = byname.get(f.name[:-len(".idx.json")])
+            if shard is not None:
+                shard.sidecar = f
+                ready.append(shard)
+        for shard in ready:
+            job, data, spans = launch(shard)
+            pending.append((shard, job, data, spans))
+        # yield whatever has finished decoding; block only over the
+        # inflight cap (bounds HBM rings held by un-yielded shards)
+        while pending and (len(pending) > inflight
+                           or pending[0][1] is None
+                           or pending[0][1].done()):
+            yield finish(pending.popleft())
+    while pending:
+        yield finish(pending.popleft())

@@ -96,6 +96,58 @@ def inflate_gpu(streams: list[tuple[int, int, int, int]],
 ZSTD_WS_BYTES = 144 << 10
 
 
+def _parse_results(desc: bytearray, n: int) -> list[InflateResult]:
+    out = []
+    for i in range(n):
+        vals = struct.unpack_from("<8Q", desc, i * DESC_WORDS * 8)
+        written, status_u, consumed = vals[4], vals[5], vals[6]
+        status = status_u - (1 << 64) if status_u >= (1 << 63) else status_u
+        out.append(InflateResult(written=written, status=int(status),
+                                 consumed=consumed))
+    return out
+
+
+class ZstdJob:
+    """An in-flight GPU zstd batch on its own HIP stream: the launch
+    returns immediately so several batches (e.g. dataset shards landing
+    at different times) decode CONCURRENTLY, which keeps wave occupancy
+    high even when each batch alone has fewer frames than the chip has
+    wave slots.  pre_launch(stream_handle), when given, queues extra
+    async work (device copies) on the same stream before the kernel."""
+
+    def __init__(self, frames: list[tuple[int, int, int, int]],
+                 pre_launch=None):
+        from ...gpu import hip
+
+        h = hip()
+        n = self._n = len(frames)
+        self._s = h.Stream(0)
+        if pre_launch is not None:
+            pre_launch(self._s.handle)
+        if n:
+            self._ws = h.DeviceBuffer(n * ZSTD_WS_BYTES)
+            desc = self._desc = bytearray(n * DESC_WORDS * 8)
+            for i, (src, slen, dst, cap) in enumerate(frames):
+                struct.pack_into("<8Q", desc, i * DESC_WORDS * 8,
+                                 src, slen, dst, cap, 0, 0, 0,
+                                 self._ws.ptr + i * ZSTD_WS_BYTES)
+            self._dbuf = h.DeviceBuffer(len(desc))
+            self._carr = (ctypes.c_char * len(desc)).from_buffer(desc)
+            addr = ctypes.addressof(self._carr)
+            h.h2d_async(self._dbuf.ptr, addr, len(desc), self._s.handle)
+            h.zstd_frames(self._dbuf.ptr, n, self._s.handle)
+            h.d2h_async(addr, self._dbuf.ptr, len(desc), self._s.handle)
+        self._ev = h.Event()
+        self._ev.record(self._s.handle)
+
+    def done(self) -> bool:
+        return self._ev.query()
+
+    def wait(self) -> list[InflateResult]:
+        self._s.sync()
+        return _parse_results(self._desc, self._n) if self._n else []
+
+
 def zstd_gpu(frames: list[tuple[int, int, int, int]],
              stream_handle=None) -> list[InflateResult]:
     """Decompress zstd frames on the GPU (csrc/zstd_kernel.hip).
@@ -105,13 +157,12 @@ def zstd_gpu(frames: list[tuple[int, int, int, int]],
     """
     from ...gpu import hip
 
-    h = hip()
     n = len(frames)
     if n == 0:
         return []
-    own = stream_handle is None
-    s = h.Stream(0) if own else None
-    handle = s.handle if own else stream_handle
+    if stream_handle is None:
+        return ZstdJob(frames).wait()
+    h = hip()
     ws = h.DeviceBuffer(n * ZSTD_WS_BYTES)
     desc = bytearray(n * DESC_WORDS * 8)
     for i, (src, slen, dst, cap) in enumerate(frames):
@@ -120,21 +171,11 @@ def zstd_gpu(frames: list[tuple[int, int, int, int]],
                          ws.ptr + i * ZSTD_WS_BYTES)
     dbuf = h.DeviceBuffer(len(desc))
     carr = (ctypes.c_char * len(desc)).from_buffer(desc)
-    h.h2d_async(dbuf.ptr, ctypes.addressof(carr), len(desc), handle)
-    h.zstd_frames(dbuf.ptr, n, handle)
-    h.d2h_async(ctypes.addressof(carr), dbuf.ptr, len(desc), handle)
-    if own:
-        s.sync()
-    else:
-        h.device_sync()
-    out = []
-    for i in range(n):
-        vals = struct.unpack_from("<8Q", desc, i * DESC_WORDS * 8)
-        written, status_u, consumed = vals[4], vals[5], vals[6]
-        status = status_u - (1 << 64) if status_u >= (1 << 63) else status_u
-        out.append(InflateResult(written=written, status=int(status),
-                                 consumed=consumed))
-    return out
+    h.h2d_async(dbuf.ptr, ctypes.addressof(carr), len(desc), stream_handle)
+    h.zstd_frames(dbuf.ptr, n, stream_handle)
+    h.d2h_async(ctypes.addressof(carr), dbuf.ptr, len(desc), stream_handle)
+    h.device_sync()
+    return _parse_results(desc, n)
 
 
 def gunzip_blob_gpu(blob, out_size: int | None = None):

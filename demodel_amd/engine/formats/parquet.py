@@ -321,19 +321,20 @@ def file_pages(path: str):
     return raw, out
 
 
-def decompress_pages_gpu(blob, pages, ring=None):
-    """GPU-decompress a landed parquet blob's ZSTD pages into an HBM
-    ring.  pages: [(codec, PageInfo)] with offsets relative to the blob.
-    Returns (ring_buffer, [(out_offset, size)]) covering every page
-    (UNCOMPRESSED pages are device-copied)."""
+def launch_pages_gpu(blob, pages, ring=None):
+    """Queue GPU decompression of a landed parquet blob's ZSTD pages
+    into an HBM ring, ASYNC on the job's own stream.  pages: [(codec,
+    PageInfo)] with offsets relative to the blob.  Returns (job, ring,
+    spans); call job.wait() and check the results before reading the
+    ring (UNCOMPRESSED pages are device-copied on the same stream)."""
     from ...gpu import hip
-    from .compress import zstd_gpu
+    from .compress import ZstdJob
 
     h = hip()
     total = sum(p.uncomp_size for _, p in pages)
     ring = ring or h.DeviceBuffer(max(total, 1))
-    s = h.Stream(0)
     frames = []
+    copies = []
     spans = []
     off = 0
     for codec, p in pages:
@@ -341,17 +342,27 @@ def decompress_pages_gpu(blob, pages, ring=None):
             frames.append((blob.buffer.ptr + p.comp_offset, p.comp_size,
                            ring.ptr + off, p.uncomp_size))
         elif codec == CODEC_UNCOMPRESSED:
-            h.d2d_async(ring.ptr + off, blob.buffer.ptr + p.comp_offset,
-                        p.comp_size, s.handle)
+            copies.append((ring.ptr + off,
+                           blob.buffer.ptr + p.comp_offset, p.comp_size))
         else:
             raise ValueError(f"GPU path supports ZSTD/UNCOMPRESSED, "
                              f"got codec {codec}")
         spans.append((off, p.uncomp_size))
         off += p.uncomp_size
-    if frames:
-        results = zstd_gpu(frames, stream_handle=s.handle)
-        bad = [(i, r) for i, r in enumerate(results) if not r.ok]
-        if bad:
-            raise IOError(f"GPU page decompress failed: {bad[:3]}")
-    s.sync()
+
+    def pre(handle):
+        for dst, src, n in copies:
+            h.d2d_async(dst, src, n, handle)
+
+    return ZstdJob(frames, pre_launch=pre), ring, spans
+
+
+def decompress_pages_gpu(blob, pages, ring=None):
+    """Synchronous wrapper around launch_pages_gpu: returns
+    (ring_buffer, [(out_offset, size)]) covering every page."""
+    job, ring, spans = launch_pages_gpu(blob, pages, ring)
+    results = job.wait()
+    bad = [(i, r) for i, r in enumerate(results) if not r.ok]
+    if bad:
+        raise IOError(f"GPU page decompress failed: {bad[:3]}")
     return ring, spans

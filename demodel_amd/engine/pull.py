@@ -415,6 +415,69 @@ def fetch_peer_digests(endpoint: str, path: str, cafile=None,
     return bytes.fromhex("".join(digs)), int(obj["chunk_bytes"])
 
 
+def pull_hf_stream(repo: str, rev: str = "main",
+                   endpoint: str | None = None,
+                   device_index: int = 0, workers: int = 4,
+                   verify: str = "chunked", cafile=None,
+                   insecure: bool = False,
+                   patterns: list[str] | None = None,
+                   landers: LanderPool | None = None,
+                   slab_bytes: int = 32 << 20,
+                   digest_map: dict[str, bytes] | None = None,
+                   peer_verify: bool = False):
+    """Streaming pull: returns (info, names, generator) where the
+    generator yields each PulledFile AS IT FINISHES landing, so a
+    consumer (e.g. stream_dataset's GPU decompression) overlaps with the
+    remaining downloads.  The generator owns the worker pools; closing
+    it early cancels unstarted pulls."""
+    endpoint = (endpoint or os.environ.get("HF_ENDPOINT")
+                or HF_DEFAULT_ENDPOINT).rstrip("/")
+    info = fetch.get_json(f"{endpoint}/api/models/{repo}/revision/{rev}",
+                          cafile=cafile, insecure=insecure)
+    names = [s["rfilename"] for s in info.get("siblings", [])]
+    if patterns:
+        names = [n for n in names
+                 if any(fnmatch.fnmatch(n, p) for p in patterns)]
+    landers = landers or LanderPool(device_index, slab_bytes=slab_bytes)
+    from ..gpu import have_gpu
+
+    def peer_expected(n):
+        if not peer_verify:
+            return None
+        return fetch_peer_digests(endpoint, f"/{repo}/resolve/{rev}/{n}",
+                                  cafile=cafile, insecure=insecure)
+
+    def gen():
+        seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
+                  if have_gpu() else None)
+        try:
+            with cf.ThreadPoolExecutor(max_workers=workers) as ex:
+                futs = {}
+                for n in names:
+                    pd = peer_expected(n)
+                    exp = (digest_map or {}).get(n)
+                    vc = None
+                    if pd is not None and exp is None:
+                        exp, vc = pd
+                    futs[ex.submit(
+                        _pull_blob, landers, n,
+                        f"{endpoint}/{repo}/resolve/{rev}/{n}",
+                        None, verify, cafile, insecure, None,
+                        exp, seg_ex, vc)] = n
+                try:
+                    for fut in cf.as_completed(futs):
+                        yield fut.result()
+                except GeneratorExit:
+                    for f in futs:
+                        f.cancel()
+                    raise
+        finally:
+            if seg_ex:
+                seg_ex.shutdown()
+
+    return info, names, gen()
+
+
 def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             device_index: int = 0, workers: int = 4,
             verify: str = "chunked", out_dir: str | None = None,
@@ -427,45 +490,14 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
     """peer_verify: when `endpoint` is another demodel node, fetch its
     recorded chunk digests per blob and GPU-verify the pull against them
     (verified distribution)."""
-    endpoint = (endpoint or os.environ.get("HF_ENDPOINT")
-                or HF_DEFAULT_ENDPOINT).rstrip("/")
     t0 = time.perf_counter()
-    info = fetch.get_json(f"{endpoint}/api/models/{repo}/revision/{rev}",
-                          cafile=cafile, insecure=insecure)
-    names = [s["rfilename"] for s in info.get("siblings", [])]
-    if patterns:
-        names = [n for n in names
-                 if any(fnmatch.fnmatch(n, p) for p in patterns)]
-    landers = landers or LanderPool(device_index, slab_bytes=slab_bytes)
+    info, _, gen = pull_hf_stream(
+        repo, rev, endpoint=endpoint, device_index=device_index,
+        workers=workers, verify=verify, cafile=cafile, insecure=insecure,
+        patterns=patterns, landers=landers, slab_bytes=slab_bytes,
+        digest_map=digest_map, peer_verify=peer_verify)
     result = PullResult(spec=f"hf://{repo}@{rev}")
-    from ..gpu import have_gpu
-
-    seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
-              if have_gpu() else None)
-
-    def peer_expected(n):
-        if not peer_verify:
-            return None
-        return fetch_peer_digests(endpoint, f"/{repo}/resolve/{rev}/{n}",
-                                  cafile=cafile, insecure=insecure)
-
-    with cf.ThreadPoolExecutor(max_workers=workers) as ex:
-        futs = {}
-        for n in names:
-            pd = peer_expected(n)
-            exp = (digest_map or {}).get(n)
-            vc = None
-            if pd is not None and exp is None:
-                exp, vc = pd
-            futs[ex.submit(
-                _pull_blob, landers, n,
-                f"{endpoint}/{repo}/resolve/{rev}/{n}",
-                None, verify, cafile, insecure, None,
-                exp, seg_ex, vc)] = n
-        for fut in cf.as_completed(futs):
-            result.files.append(fut.result())
-    if seg_ex:
-        seg_ex.shutdown()
+    result.files = list(gen)
     result.files.sort(key=lambda f: f.name)
     result.total_bytes = sum(f.nbytes for f in result.files)
     result.seconds_to_ready = time.perf_counter() - t0
