@@ -54,7 +54,7 @@ void launch_gguf_dequant(int qtype, const void* src, uint16_t* dst_bf16,
 void launch_inflate_streams(const uint64_t* desc, int n_streams,
                             int* status, hipStream_t stream);
 void launch_zstd_frames(const uint64_t* desc, int n_frames, int* status,
-                        hipStream_t stream);
+                        hipStream_t stream, int window);
 }
 
 // ------------------------------------------------------------------------
@@ -380,12 +380,14 @@ PYBIND11_MODULE(_hip, m) {
         "DEFLATE-inflate n_streams descriptors (8 u64 each: src, src_len, "
         "dst, dst_cap, written, status, consumed, pad), wave per stream");
   m.def("zstd_frames",
-        [](uintptr_t desc, int n_frames, uintptr_t stream) {
+        [](uintptr_t desc, int n_frames, uintptr_t stream, int window) {
           launch_zstd_frames((const uint64_t*)desc, n_frames, nullptr,
-                             (hipStream_t)stream);
+                             (hipStream_t)stream, window);
         },
-        py::call_guard<py::gil_scoped_release>(),
+        py::call_guard<py::gil_scoped_release>(), py::arg("desc"),
+        py::arg("n_frames"), py::arg("stream"), py::arg("window") = 0,
         "zstd-decompress n_frames descriptors (8 u64 each: src, src_len, "
         "dst, dst_cap, written, status, consumed, ws), wave per frame; ws "
-        "needs >= 144 KiB per frame");
+        "needs >= 144 KiB per frame; window 0=auto/16384/65536 selects "
+        "the LDS window template");
 }

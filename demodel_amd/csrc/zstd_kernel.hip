@@ -506,11 +506,18 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
 }  // namespace
 
 extern "C" void launch_zstd_frames(const uint64_t* desc, int n_frames,
-                                   int* /*unused*/, hipStream_t stream) {
+                                   int* /*unused*/, hipStream_t stream,
+                                   int window /*0=auto,16384,65536*/) {
   if (n_frames <= 0) return;
   int blocks = n_frames < 4096 ? n_frames : 4096;
-  if (n_frames >= 768) {
-    // occupancy regime: small window, more resident workgroups
+  // both windows are CORRECT for any match distance (exec_seq falls
+  // back to a global read past the LDS window); the choice is pure
+  // occupancy vs far-match speed.  Callers that run MANY CONCURRENT
+  // small launches (per-shard dataset decode) pass 16384 explicitly:
+  // LDS is shared chip-wide, so 64 KiB windows would cap residency at
+  // ~2 workgroups/CU across all their launches combined.
+  bool small = window == 16 * 1024 || (window == 0 && n_frames >= 768);
+  if (small) {
     hipLaunchKernelGGL(zstd_kernel<16 * 1024>, dim3(blocks), dim3(64), 0,
                        stream, (ZstdDesc*)desc, n_frames);
   } else {

@@ -66,7 +66,9 @@ def _launch_zst_frames_gpu(blob, idx):
                        ring.ptr + off, fr["decompressed"]))
         spans.append((off, fr["decompressed"]))
         off += fr["decompressed"]
-    job = ZstdJob(frames)
+    # 16 KiB window: concurrent shard jobs share chip LDS, and word-
+    # level text matches are short (far matches take the global path)
+    job = ZstdJob(frames, window=16 << 10)
     return job, torch.from_dlpack(ring.to_dlpack()), spans
 
 

@@ -116,7 +116,7 @@ class ZstdJob:
     async work (device copies) on the same stream before the kernel."""
 
     def __init__(self, frames: list[tuple[int, int, int, int]],
-                 pre_launch=None):
+                 pre_launch=None, window: int = 0):
         from ...gpu import hip
 
         h = hip()
@@ -135,7 +135,8 @@ class ZstdJob:
             self._carr = (ctypes.c_char * len(desc)).from_buffer(desc)
             addr = ctypes.addressof(self._carr)
             h.h2d_async(self._dbuf.ptr, addr, len(desc), self._s.handle)
-            h.zstd_frames(self._dbuf.ptr, n, self._s.handle)
+            h.zstd_frames(self._dbuf.ptr, n, self._s.handle,
+                          window=window)
             h.d2h_async(addr, self._dbuf.ptr, len(desc), self._s.handle)
         self._ev = h.Event()
         self._ev.record(self._s.handle)
@@ -149,7 +150,7 @@ class ZstdJob:
 
 
 def zstd_gpu(frames: list[tuple[int, int, int, int]],
-             stream_handle=None) -> list[InflateResult]:
+             stream_handle=None, window: int = 0) -> list[InflateResult]:
     """Decompress zstd frames on the GPU (csrc/zstd_kernel.hip).
 
     frames: (src_ptr, src_len, dst_ptr, dst_cap) device addresses.
@@ -161,7 +162,7 @@ def zstd_gpu(frames: list[tuple[int, int, int, int]],
     if n == 0:
         return []
     if stream_handle is None:
-        return ZstdJob(frames).wait()
+        return ZstdJob(frames, window=window).wait()
     h = hip()
     ws = h.DeviceBuffer(n * ZSTD_WS_BYTES)
     desc = bytearray(n * DESC_WORDS * 8)
@@ -172,7 +173,7 @@ def zstd_gpu(frames: list[tuple[int, int, int, int]],
     dbuf = h.DeviceBuffer(len(desc))
     carr = (ctypes.c_char * len(desc)).from_buffer(desc)
     h.h2d_async(dbuf.ptr, ctypes.addressof(carr), len(desc), stream_handle)
-    h.zstd_frames(dbuf.ptr, n, stream_handle)
+    h.zstd_frames(dbuf.ptr, n, stream_handle, window=window)
     h.d2h_async(ctypes.addressof(carr), dbuf.ptr, len(desc), stream_handle)
     h.device_sync()
     return _parse_results(desc, n)

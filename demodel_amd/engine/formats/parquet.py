@@ -354,7 +354,9 @@ def launch_pages_gpu(blob, pages, ring=None):
         for dst, src, n in copies:
             h.d2d_async(dst, src, n, handle)
 
-    return ZstdJob(frames, pre_launch=pre), ring, spans
+    # 16 KiB LDS window: page batches run as concurrent jobs in
+    # stream_dataset, so occupancy beats far-match locality
+    return ZstdJob(frames, pre_launch=pre, window=16 << 10), ring, spans
 
 
 def decompress_pages_gpu(blob, pages, ring=None):

@@ -34,7 +34,7 @@ def test_stream_zst_shards_cpu(tmp_path):
             want = bytes(codec.decompress(
                 raw[fr["offset"]:fr["offset"] + fr["compressed"]],
                 fr["decompressed"]))
-            got = bytes(b.tensors()[0].numpy().tobytes())
+            got = bytes(b.tensors()[0].cpu().numpy().tobytes())
             assert got == want
     finally:
         stack.close()
