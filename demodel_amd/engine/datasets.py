@@ -170,6 +170,8 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                 ready.append(shard)
         for shard in ready:
             job, data, spans = launch(shard)
+            log.info("launched decode of %s (%d pending)", shard.name,
+                     len(pending) + 1)
             pending.append((shard, job, data, spans))
         # yield whatever has finished decoding; block only over the
         # inflight cap (bounds HBM rings held by un-yielded shards)
