@@ -50,14 +50,18 @@ def _launch_parquet_gpu(blob):
 
 
 def _launch_zst_frames_gpu(blob, idx):
+    import time as _t
+
     import torch
 
     from ..gpu import hip
     from .formats.compress import ZstdJob
 
     h = hip()
+    _t0 = _t.perf_counter()
     total = sum(fr["decompressed"] for fr in idx["frames"])
     ring = h.DeviceBuffer(max(total, 1))
+    _t1 = _t.perf_counter()
     frames = []
     spans = []
     off = 0
@@ -69,6 +73,8 @@ def _launch_zst_frames_gpu(blob, idx):
     # 16 KiB window: concurrent shard jobs share chip LDS, and word-
     # level text matches are short (far matches take the global path)
     job = ZstdJob(frames, window=16 << 10)
+    log.info("zst launch: alloc %.3fs job %.3fs", _t1 - _t0,
+             _t.perf_counter() - _t1)
     return job, torch.from_dlpack(ring.to_dlpack()), spans
 
 
@@ -169,9 +175,13 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                 shard.sidecar = f
                 ready.append(shard)
         for shard in ready:
+            import time as _t
+            _t0 = _t.perf_counter()
             job, data, spans = launch(shard)
-            log.info("launched decode of %s (%d pending)", shard.name,
-                     len(pending) + 1)
+            log.info("launched decode of %s (%d pending, launch %.3fs, "
+                     "done=%s)", shard.name, len(pending) + 1,
+                     _t.perf_counter() - _t0,
+                     job.done() if job is not None else "cpu")
             pending.append((shard, job, data, spans))
         # yield whatever has finished decoding; block only over the
         # inflight cap (bounds HBM rings held by un-yielded shards)
