@@ -126,18 +126,23 @@ class ZstdJob:
             pre_launch(self._s.handle)
         if n:
             self._ws = h.DeviceBuffer(n * ZSTD_WS_BYTES)
-            desc = self._desc = bytearray(n * DESC_WORDS * 8)
+            nd = n * DESC_WORDS * 8
+            # PINNED staging for the descriptor block: hipMemcpyAsync
+            # to/from pageable memory BLOCKS the host thread until every
+            # prior op on the stream (the kernel!) completes, which would
+            # make this "async" launch synchronous
+            self._pin = h.PinnedPool(nd, 1)
+            desc = self._pin.slab_view(0)
             for i, (src, slen, dst, cap) in enumerate(frames):
                 struct.pack_into("<8Q", desc, i * DESC_WORDS * 8,
                                  src, slen, dst, cap, 0, 0, 0,
                                  self._ws.ptr + i * ZSTD_WS_BYTES)
-            self._dbuf = h.DeviceBuffer(len(desc))
-            self._carr = (ctypes.c_char * len(desc)).from_buffer(desc)
-            addr = ctypes.addressof(self._carr)
-            h.h2d_async(self._dbuf.ptr, addr, len(desc), self._s.handle)
+            self._dbuf = h.DeviceBuffer(nd)
+            addr = self._pin.slab_ptr(0)
+            h.h2d_async(self._dbuf.ptr, addr, nd, self._s.handle)
             h.zstd_frames(self._dbuf.ptr, n, self._s.handle,
                           window=window)
-            h.d2h_async(addr, self._dbuf.ptr, len(desc), self._s.handle)
+            h.d2h_async(addr, self._dbuf.ptr, nd, self._s.handle)
         self._ev = h.Event()
         self._ev.record(self._s.handle)
 
@@ -146,7 +151,9 @@ class ZstdJob:
 
     def wait(self) -> list[InflateResult]:
         self._s.sync()
-        return _parse_results(self._desc, self._n) if self._n else []
+        if not self._n:
+            return []
+        return _parse_results(bytearray(self._pin.slab_view(0)), self._n)
 
 
 def zstd_gpu(frames: list[tuple[int, int, int, int]],
