@@ -44,24 +44,16 @@ def _launch_parquet_gpu(blob):
 
     pages = pqf.blob_pages(blob)
     job, ring, spans = pqf.launch_pages_gpu(blob, pages)
-    import torch
-
-    return job, torch.from_dlpack(ring.to_dlpack()), spans
+    return job, ring, spans
 
 
 def _launch_zst_frames_gpu(blob, idx):
-    import time as _t
-
-    import torch
-
     from ..gpu import hip
     from .formats.compress import ZstdJob
 
     h = hip()
-    _t0 = _t.perf_counter()
     total = sum(fr["decompressed"] for fr in idx["frames"])
     ring = h.DeviceBuffer(max(total, 1))
-    _t1 = _t.perf_counter()
     frames = []
     spans = []
     off = 0
@@ -73,9 +65,7 @@ def _launch_zst_frames_gpu(blob, idx):
     # 16 KiB window: concurrent shard jobs share chip LDS, and word-
     # level text matches are short (far matches take the global path)
     job = ZstdJob(frames, window=16 << 10)
-    log.info("zst launch: alloc %.3fs job %.3fs", _t1 - _t0,
-             _t.perf_counter() - _t1)
-    return job, torch.from_dlpack(ring.to_dlpack()), spans
+    return job, ring, spans
 
 
 def _decompress_cpu(blob, idx):
@@ -127,7 +117,10 @@ def stream_dataset(repo: str, endpoint: str | None = None,
     gpu = have_gpu()
 
     def launch(f):
-        """-> (job, data, spans); job None on the CPU path."""
+        """-> (job, ring_or_data, spans); job None on the CPU path.
+        The torch wrap of a GPU ring happens at FINISH time:
+        torch.from_dlpack synchronizes with the device, which would
+        serialize the launches."""
         if f.name.endswith(".parquet"):
             if not gpu:
                 raise RuntimeError(
@@ -148,6 +141,9 @@ def stream_dataset(repo: str, endpoint: str | None = None,
             if bad:
                 raise IOError(
                     f"GPU decompress of {shard.name} failed: {bad[:3]}")
+            import torch
+
+            data = torch.from_dlpack(data.to_dlpack())
         log.info("dataset shard %s: %d spans, %d bytes decompressed",
                  shard.name, len(spans), int(data.numel()))
         return ShardBatch(name=shard.name, data=data, spans=spans)
@@ -175,13 +171,9 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                 shard.sidecar = f
                 ready.append(shard)
         for shard in ready:
-            import time as _t
-            _t0 = _t.perf_counter()
             job, data, spans = launch(shard)
-            log.info("launched decode of %s (%d pending, launch %.3fs, "
-                     "done=%s)", shard.name, len(pending) + 1,
-                     _t.perf_counter() - _t0,
-                     job.done() if job is not None else "cpu")
+            log.debug("launched decode of %s (%d pending)", shard.name,
+                      len(pending) + 1)
             pending.append((shard, job, data, spans))
         # yield whatever has finished decoding; block only over the
         # inflight cap (bounds HBM rings held by un-yielded shards)
