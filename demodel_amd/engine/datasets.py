@@ -39,17 +39,15 @@ class ShardBatch:
         return [self.data[o:o + n] for o, n in self.spans]
 
 
-def _launch_parquet_gpu(blob):
+def _prep_parquet_gpu(blob):
     from .formats import parquet as pqf
 
     pages = pqf.blob_pages(blob)
-    job, ring, spans = pqf.launch_pages_gpu(blob, pages)
-    return job, ring, spans
+    return pqf.prep_pages_gpu(blob, pages)
 
 
-def _launch_zst_frames_gpu(blob, idx):
+def _prep_zst_frames_gpu(blob, idx):
     from ..gpu import hip
-    from .formats.compress import ZstdJob
 
     h = hip()
     total = sum(fr["decompressed"] for fr in idx["frames"])
@@ -62,10 +60,7 @@ def _launch_zst_frames_gpu(blob, idx):
                        ring.ptr + off, fr["decompressed"]))
         spans.append((off, fr["decompressed"]))
         off += fr["decompressed"]
-    # 16 KiB window: concurrent shard jobs share chip LDS, and word-
-    # level text matches are short (far matches take the global path)
-    job = ZstdJob(frames, window=16 << 10)
-    return job, ring, spans
+    return frames, [], ring, spans
 
 
 def _decompress_cpu(blob, idx):
@@ -112,26 +107,56 @@ def stream_dataset(repo: str, endpoint: str | None = None,
     _, names, gen = pull_hf_stream(
         repo, endpoint=endpoint, workers=workers, verify=verify,
         patterns=list(patterns), device_index=device_index,
-        landers=landers, digest_map=digest_map)
+        landers=landers, digest_map=digest_map, batched=True)
     expected = set(names)
     gpu = have_gpu()
 
-    def launch(f):
-        """-> (job, ring_or_data, spans); job None on the CPU path.
-        The torch wrap of a GPU ring happens at FINISH time:
-        torch.from_dlpack synchronizes with the device, which would
-        serialize the launches."""
-        if f.name.endswith(".parquet"):
-            if not gpu:
-                raise RuntimeError(
-                    "parquet streaming needs a GPU (CPU fallback covers "
-                    ".zst shards)")
-            return _launch_parquet_gpu(f.blob)
-        idx = _sidecar_idx(f.sidecar)
-        if gpu:
-            return _launch_zst_frames_gpu(f.blob, idx)
-        data, spans = _decompress_cpu(f.blob, idx)
-        return None, data, spans
+    def launch_batch(shards):
+        """Coalesce every shard that became ready at the same time into
+        ONE kernel launch (a shared ZstdJob; each shard holds a frame-
+        range view).  Per-shard frame counts (~hundreds) are far below
+        the chip's wave slots, and kernels on the same HW queue
+        serialize — batching restores big-launch occupancy while the
+        launch itself stays async.  Returns [(shard, jobview_or_None,
+        ring_or_data, spans)].  The torch wrap of a GPU ring happens at
+        FINISH time: torch.from_dlpack synchronizes with the device,
+        which would serialize the launches."""
+        from ..gpu import hip
+        from .formats.compress import ZstdJob
+
+        out = []
+        all_frames = []
+        all_copies = []
+        entries = []
+        for f in shards:
+            if f.name.endswith(".parquet"):
+                if not gpu:
+                    raise RuntimeError(
+                        "parquet streaming needs a GPU (CPU fallback "
+                        "covers .zst shards)")
+                frames, copies, ring, spans = _prep_parquet_gpu(f.blob)
+            else:
+                idx = _sidecar_idx(f.sidecar)
+                if not gpu:
+                    data, spans = _decompress_cpu(f.blob, idx)
+                    out.append((f, None, data, spans))
+                    continue
+                frames, copies, ring, spans = _prep_zst_frames_gpu(
+                    f.blob, idx)
+            entries.append((f, len(all_frames), len(frames), ring, spans))
+            all_frames += frames
+            all_copies += copies
+        if entries:
+            h = hip()
+
+            def pre(handle):
+                for dst, src, n in all_copies:
+                    h.d2d_async(dst, src, n, handle)
+
+            job = ZstdJob(all_frames, pre_launch=pre, window=16 << 10)
+            for f, lo, n, ring, spans in entries:
+                out.append((f, job.view(lo, n), ring, spans))
+        return out
 
     def finish(item):
         shard, job, data, spans = item
@@ -152,29 +177,31 @@ def stream_dataset(repo: str, endpoint: str | None = None,
     # until the decode kernel reading it has finished
     pending: deque = deque()
     byname: dict = {}
-    for f in gen:
-        if on_file is not None:
-            on_file(f)
-        byname[f.name] = f
+    for burst in gen:
         ready = []
-        if f.name.endswith(".parquet"):
-            ready.append(f)
-        elif f.name.endswith(".zst"):
-            if f.name + ".idx.json" not in expected:
-                raise FileNotFoundError(f"{f.name}: missing .idx.json")
-            if f.name + ".idx.json" in byname:
-                f.sidecar = byname[f.name + ".idx.json"]
+        for f in burst:
+            if on_file is not None:
+                on_file(f)
+            byname[f.name] = f
+            if f.name.endswith(".parquet"):
                 ready.append(f)
-        elif f.name.endswith(".zst.idx.json"):
-            shard = byname.get(f.name[:-len(".idx.json")])
-            if shard is not None:
-                shard.sidecar = f
-                ready.append(shard)
-        for shard in ready:
-            job, data, spans = launch(shard)
-            log.debug("launched decode of %s (%d pending)", shard.name,
-                      len(pending) + 1)
-            pending.append((shard, job, data, spans))
+            elif f.name.endswith(".zst"):
+                if f.name + ".idx.json" not in expected:
+                    raise FileNotFoundError(
+                        f"{f.name}: missing .idx.json")
+                if f.name + ".idx.json" in byname:
+                    f.sidecar = byname[f.name + ".idx.json"]
+                    ready.append(f)
+            elif f.name.endswith(".zst.idx.json"):
+                shard = byname.get(f.name[:-len(".idx.json")])
+                if shard is not None:
+                    shard.sidecar = f
+                    ready.append(shard)
+        if ready:
+            batch = launch_batch(ready)
+            log.debug("launched decode of %d shard(s) (%d pending)",
+                      len(batch), len(pending) + len(batch))
+            pending.extend(batch)
         # yield whatever has finished decoding; block only over the
         # inflight cap (bounds HBM rings held by un-yielded shards)
         while pending and (len(pending) > inflight

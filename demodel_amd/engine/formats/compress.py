@@ -150,10 +150,29 @@ class ZstdJob:
         return self._ev.query()
 
     def wait(self) -> list[InflateResult]:
-        self._s.sync()
-        if not self._n:
-            return []
-        return _parse_results(bytearray(self._pin.slab_view(0)), self._n)
+        if not hasattr(self, "_results"):
+            self._s.sync()
+            self._results = (_parse_results(
+                bytearray(self._pin.slab_view(0)), self._n)
+                if self._n else [])
+        return self._results
+
+    def view(self, lo: int, n: int) -> "ZstdJobView":
+        return ZstdJobView(self, lo, n)
+
+
+class ZstdJobView:
+    """A contiguous frame-range of a (possibly shared) ZstdJob — several
+    shards coalesced into ONE launch each hold a view of it."""
+
+    def __init__(self, job: ZstdJob, lo: int, n: int):
+        self._job, self._lo, self._vn = job, lo, n
+
+    def done(self) -> bool:
+        return self._job.done()
+
+    def wait(self) -> list[InflateResult]:
+        return self._job.wait()[self._lo:self._lo + self._vn]
 
 
 def zstd_gpu(frames: list[tuple[int, int, int, int]],

@@ -321,14 +321,11 @@ def file_pages(path: str):
     return raw, out
 
 
-def launch_pages_gpu(blob, pages, ring=None):
-    """Queue GPU decompression of a landed parquet blob's ZSTD pages
-    into an HBM ring, ASYNC on the job's own stream.  pages: [(codec,
-    PageInfo)] with offsets relative to the blob.  Returns (job, ring,
-    spans); call job.wait() and check the results before reading the
-    ring (UNCOMPRESSED pages are device-copied on the same stream)."""
+def prep_pages_gpu(blob, pages, ring=None):
+    """Build the decode plan for a landed parquet blob's pages: returns
+    (frames, copies, ring, spans) where frames feed a ZstdJob and copies
+    are (dst, src, n) device copies for UNCOMPRESSED pages."""
     from ...gpu import hip
-    from .compress import ZstdJob
 
     h = hip()
     total = sum(p.uncomp_size for _, p in pages)
@@ -349,6 +346,19 @@ def launch_pages_gpu(blob, pages, ring=None):
                              f"got codec {codec}")
         spans.append((off, p.uncomp_size))
         off += p.uncomp_size
+    return frames, copies, ring, spans
+
+
+def launch_pages_gpu(blob, pages, ring=None):
+    """Queue GPU decompression of a landed parquet blob's ZSTD pages
+    into an HBM ring, ASYNC on the job's own stream.  Returns (job,
+    ring, spans); call job.wait() and check the results before reading
+    the ring."""
+    from ...gpu import hip
+    from .compress import ZstdJob
+
+    h = hip()
+    frames, copies, ring, spans = prep_pages_gpu(blob, pages, ring)
 
     def pre(handle):
         for dst, src, n in copies:

@@ -424,12 +424,15 @@ def pull_hf_stream(repo: str, rev: str = "main",
                    landers: LanderPool | None = None,
                    slab_bytes: int = 32 << 20,
                    digest_map: dict[str, bytes] | None = None,
-                   peer_verify: bool = False):
+                   peer_verify: bool = False, batched: bool = False):
     """Streaming pull: returns (info, names, generator) where the
     generator yields each PulledFile AS IT FINISHES landing, so a
     consumer (e.g. stream_dataset's GPU decompression) overlaps with the
-    remaining downloads.  The generator owns the worker pools; closing
-    it early cancels unstarted pulls."""
+    remaining downloads.  With batched=True it yields LISTS — every file
+    that has finished since the last wake-up arrives together, so the
+    consumer can coalesce per-burst work (one decode launch per burst).
+    The generator owns the worker pools; closing it early cancels
+    unstarted pulls."""
     endpoint = (endpoint or os.environ.get("HF_ENDPOINT")
                 or HF_DEFAULT_ENDPOINT).rstrip("/")
     info = fetch.get_json(f"{endpoint}/api/models/{repo}/revision/{rev}",
@@ -465,8 +468,15 @@ def pull_hf_stream(repo: str, rev: str = "main",
                         None, verify, cafile, insecure, None,
                         exp, seg_ex, vc)] = n
                 try:
-                    for fut in cf.as_completed(futs):
-                        yield fut.result()
+                    if batched:
+                        left = set(futs)
+                        while left:
+                            finished, left = cf.wait(
+                                left, return_when=cf.FIRST_COMPLETED)
+                            yield [f.result() for f in finished]
+                    else:
+                        for fut in cf.as_completed(futs):
+                            yield fut.result()
                 except GeneratorExit:
                     for f in futs:
                         f.cancel()

@@ -14,6 +14,14 @@ import os
 _hip = None
 _import_error: Exception | None = None
 
+# The engine runs concurrent long kernels on many streams (per-shard
+# decodes, per-worker verify/copy streams); ROCm multiplexes streams
+# onto GPU_MAX_HW_QUEUES hardware queues (default 4) and kernels that
+# share a queue SERIALIZE — measured: per-shard zstd decodes ran
+# back-to-back instead of concurrently.  Must be set before the HIP
+# runtime initializes (i.e. before torch loads).
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")
+
 # torch MUST load before our extension: torch-rocm wheels bundle their own
 # libamdhip64, and whichever HIP runtime initializes second in a process
 # sees no devices (observed on MI355X: _hip-first broke torch.cuda,
