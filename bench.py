@@ -245,7 +245,8 @@ def main():
                                 workers=args.workers, verify=args.verify,
                                 landers=landers, patterns=patterns,
                                 digest_map=digest_map or None,
-                                on_file=on_file, inflight=64):
+                                on_file=on_file, inflight=64,
+                                eager=False):
             out_bytes += int(b.data.numel())
             n += 1
             del b

@@ -92,14 +92,21 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                    verify: str = "chunked",
                    landers: LanderPool | None = None,
                    digest_map: dict | None = None,
-                   on_file=None, inflight: int = 4):
+                   on_file=None, inflight: int = 4, eager: bool = True):
     """Yield ShardBatch per data shard of an HF dataset repo.
 
     inflight bounds how many decompressed shard rings can be in flight
     (launched but not yet yielded) at once — the HBM high-water mark is
     roughly (inflight + 1) decompressed shards plus the compressed
     blobs.  on_file(pulled_file) fires per landed file (digest capture,
-    progress)."""
+    progress).
+
+    eager=True (default) launches each burst's decode while later
+    shards still download — best latency to the first batch.  On
+    payloads where decode saturates the chip for long stretches the
+    decode waves starve the in-flight pulls' verify kernels, so
+    eager=False defers everything into ONE maximal launch after the
+    last shard lands — best whole-dataset seconds-to-ready."""
     from ..gpu import have_gpu
 
     from .pull import pull_hf_stream
@@ -176,6 +183,7 @@ def stream_dataset(repo: str, endpoint: str | None = None,
     # pending holds the PulledFile too: its HBM blob must stay alive
     # until the decode kernel reading it has finished
     pending: deque = deque()
+    deferred: list = []
     byname: dict = {}
     for burst in gen:
         ready = []
@@ -197,6 +205,9 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                 if shard is not None:
                     shard.sidecar = f
                     ready.append(shard)
+        if not eager:
+            deferred.extend(ready)
+            continue
         if ready:
             batch = launch_batch(ready)
             log.debug("launched decode of %d shard(s) (%d pending)",
@@ -208,5 +219,7 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                            or pending[0][1] is None
                            or pending[0][1].done()):
             yield finish(pending.popleft())
+    if deferred:
+        pending.extend(launch_batch(deferred))
     while pending:
         yield finish(pending.popleft())

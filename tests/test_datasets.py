@@ -41,7 +41,8 @@ def test_stream_zst_shards_cpu(tmp_path):
 
 
 @pytest.mark.gpu
-def test_stream_parquet_gpu(tmp_path):
+@pytest.mark.parametrize("eager", [True, False])
+def test_stream_parquet_gpu(tmp_path, eager):
     import pyarrow.parquet as pq
 
     from demodel_amd.engine.datasets import stream_dataset
@@ -55,7 +56,7 @@ def test_stream_parquet_gpu(tmp_path):
         batches = list(stream_dataset("ds/pq",
                                       endpoint=stack.origin_base,
                                       patterns=("*.parquet",),
-                                      workers=2))
+                                      workers=2, eager=eager))
         assert len(batches) == 2
         for b in batches:
             assert b.data.is_cuda
