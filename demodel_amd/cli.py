@@ -88,6 +88,14 @@ def _cmd_verify(args) -> int:
     return 0 if result.get("ok", False) else 1
 
 
+def _cmd_stats(args) -> int:
+    from .engine import fetch
+
+    base = (args.endpoint or "http://127.0.0.1:8080").rstrip("/")
+    print(json.dumps(fetch.get_json(base + "/__demodel/stats"), indent=1))
+    return 0
+
+
 def _cmd_gc(args) -> int:
     from .cache import CacheStore
 
@@ -134,6 +142,12 @@ def main(argv: list[str] | None = None) -> int:
     vp = sub.add_parser("verify", help="re-verify cached blobs")
     vp.add_argument("--uri", default=None)
     vp.set_defaults(fn=_cmd_verify)
+
+    st = sub.add_parser("stats", help="transfer metrics of a running "
+                                      "proxy (GET /__demodel/stats)")
+    st.add_argument("--endpoint", default=None,
+                    help="proxy base URL (default http://127.0.0.1:8080)")
+    st.set_defaults(fn=_cmd_stats)
 
     gp = sub.add_parser("gc", help="evict LRU cache entries over a size "
                                    "budget")

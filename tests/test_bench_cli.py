@@ -97,3 +97,28 @@ def test_demodel_pull_cli(tmp_path):
         assert (out_dir / "w.bin").read_bytes() == blob.read_bytes()
     finally:
         stack.close()
+
+
+def test_demodel_stats_cli(tmp_path, capsys):
+    """`demodel stats` prints a running proxy's transfer metrics."""
+    sys.path.insert(0, os.path.join(REPO, "tests"))
+    from helpers import Stack
+
+    from demodel_amd.cli import main as cli_main
+
+    stack = Stack(tmp_path)
+    try:
+        blob = tmp_path / "s.bin"
+        blob.write_bytes(b"x" * 1000)
+        stack.origin.add_hf_repo("org/st", {"s.bin": str(blob)})
+        import urllib.request
+
+        urllib.request.urlopen(
+            f"{stack.endpoint}/org/st/resolve/main/s.bin").read()
+        rc = cli_main(["stats", "--endpoint", stack.endpoint])
+        assert rc == 0
+        out = json.loads(capsys.readouterr().out)
+        assert out["requests"] >= 1
+        assert out["cache_misses"] >= 1
+    finally:
+        stack.close()

@@ -7,7 +7,8 @@ from helpers import Stack
 pa = pytest.importorskip("pyarrow")
 
 
-def test_stream_zst_shards_cpu(tmp_path):
+@pytest.mark.parametrize("eager", [True, False])
+def test_stream_zst_shards_cpu(tmp_path, eager):
     from demodel_amd.engine.datasets import stream_dataset
     from demodel_amd.testing import synth
 
@@ -19,7 +20,7 @@ def test_stream_zst_shards_cpu(tmp_path):
         stack.origin.add_hf_repo("ds/c4", files)
         batches = list(stream_dataset("ds/c4",
                                       endpoint=stack.origin_base,
-                                      workers=2))
+                                      workers=2, eager=eager))
         assert len(batches) == 2
         codec = pa.Codec("zstd")
         import json
