@@ -53,8 +53,8 @@ def test_peer_digests_endpoint_and_verified_pull(stack, tmp_path):
     data, url = _prime(stack, tmp_path)
     obj = _wait_digests(stack, "/org/p/resolve/main/w.bin")
     assert obj["body_size"] == len(data)
-    assert obj["chunk_bytes"] == 1 << 20
-    assert len(obj["chunk_sha256"]) == 3
+    assert obj["chunk_bytes"] == 64 << 10
+    assert len(obj["chunk_sha256"]) == 48  # 3 MiB / 64 KiB
     import hashlib
 
     assert obj["sha256"] == hashlib.sha256(data).hexdigest()
