@@ -59,13 +59,13 @@ def test_peer_digests_endpoint_and_verified_pull(stack, tmp_path):
 
     assert obj["sha256"] == hashlib.sha256(data).hexdigest()
     assert obj["chunk_sha256"][0] == \
-        hashlib.sha256(data[:1 << 20]).hexdigest()
+        hashlib.sha256(data[:64 << 10]).hexdigest()
 
     # peer-verified pull: engine compares every chunk to the record
     res = pull_mod.pull_hf("org/p", endpoint=stack.endpoint,
                            verify="chunked", workers=1, peer_verify=True)
     f = res.files[0]
-    assert f.blob.verify_chunk == 1 << 20
+    assert f.blob.verify_chunk == 64 << 10
     assert bytes(f.blob.buffer) == data
 
 
@@ -84,4 +84,4 @@ def test_peer_verify_detects_tamper(stack, tmp_path):
     with pytest.raises(DigestMismatch) as ei:
         pull_mod.pull_hf("org/p", endpoint=stack.endpoint,
                          verify="chunked", workers=1, peer_verify=True)
-    assert ei.value.chunk_index == 1  # byte 2,000,000 sits in chunk 1
+    assert ei.value.chunk_index == 2_000_000 // (64 << 10)
