@@ -197,36 +197,35 @@ def main():
         from demodel_amd.engine.formats import gguf
         from demodel_amd.gpu import have_gpu as _hg, hip
 
+        pd = gguf.ProgressiveDequant() if _hg() else None
         res = pull_mod.pull_hf(
             "bench/model", endpoint=endpoint, workers=args.workers,
             verify=args.verify, landers=landers,
-            digest_map=digest_map or None)
+            digest_map=digest_map or None,
+            on_range=pd.on_range if pd else None)
         assert res.total_bytes == total_bytes
         if record_digests:
             for f in res.files:
                 digest_map[f.name] = f.blob.digest_blob
         blob = res.files[0].blob
-        gg = gguf.parse(blob)
-        quants = [t for t in gg.tensors if t.type_id in (2, 8, 12, 14)]
-        out_bytes = sum(t.n_elems * 2 for t in quants)
-        if _hg() and quants:
-            h = hip()
-            stream = h.Stream(0)
-            # one arena for all bf16 outputs: per-tensor hipMalloc would
-            # serialize ~300 allocations between the async launches
-            arena = h.DeviceBuffer(out_bytes)
-            off = 0
-            for t in quants:
-                h.gguf_dequant(t.type_id,
-                               blob.buffer.ptr + gg.data_offset + t.offset,
-                               arena.ptr + off, t.n_blocks, stream.handle)
-                off += t.n_elems * 2
-            stream.sync()
-            del arena
+        if pd is not None:
+            # per-tensor dequants launched WHILE segments landed; this
+            # finishes the stragglers and syncs (one output arena)
+            tensors = pd.finish(blob)
+            out_bytes = sum(t.numel() * 2 for t in tensors.values())
+            n_t = len(tensors)
+            log(f"dequant overlap: {pd.launched_early}/{n_t} tensors "
+                f"launched before pull completion")
+            del tensors, pd
+        else:
+            gg = gguf.parse(blob)
+            out_bytes = sum(t.n_elems * 2 for t in gg.tensors
+                            if t.type_id in (2, 8, 12, 14))
+            n_t = len(gg.tensors)
         if have_gpu:
             torch.cuda.synchronize()
         res.meta["dequant_bf16_bytes"] = out_bytes
-        return res, len(gg.tensors)
+        return res, n_t
 
     def _stream_step(record_digests, patterns):
         """config 5 via the user-facing stream_dataset API: each landed

@@ -324,6 +324,132 @@ def _dequant_cpu_impl(type_id: int, raw: bytes, n_elems: int):
 # ------------------------------------------------------------------ #
 # GPU dequantization from a landed blob
 
+class ProgressiveDequant:
+    """Dequantize a GGUF blob WHILE it lands.
+
+    Feed `on_range` events from a (segmented) pull
+    (pull_hf(..., on_range=pd.on_range)): the header parses from the
+    first range's head bytes, and each quant tensor's dequant kernel
+    launches the moment the landed contiguous prefix covers its byte
+    range — so the dequant (tens of ms of GPU work) hides entirely
+    under the download instead of running after it.  `finish(blob)`
+    launches whatever remains, syncs, and returns {name: bf16 tensor}.
+
+    Thread-safe: on_range arrives from segment worker threads.  If the
+    head bytes don't hold the whole header (huge embedded tokenizer
+    metadata), everything simply launches at finish() — correctness
+    never depends on the overlap.
+    """
+
+    QUANT_IDS = (2, 8, 12, 14)
+
+    def __init__(self, device_index: int = 0):
+        import threading
+
+        self._lock = threading.Lock()
+        self._ranges: list = []     # merged completed [lo, hi)
+        self._prefix = 0            # contiguous bytes landed from 0
+        self._gg: GGUFModel | None = None
+        self._buf = None            # DeviceBuffer being filled
+        self._h = None
+        self._stream = None
+        self._arena = None
+        self._offsets: dict = {}
+        self._quants: list = []
+        self._next = 0
+
+    # -- range bookkeeping -------------------------------------------- #
+    def _add_range(self, lo: int, hi: int) -> None:
+        rs = self._ranges
+        rs.append((lo, hi))
+        rs.sort()
+        merged = [rs[0]]
+        for a, b in rs[1:]:
+            if a <= merged[-1][1]:
+                merged[-1] = (merged[-1][0], max(merged[-1][1], b))
+            else:
+                merged.append((a, b))
+        self._ranges = merged
+        self._prefix = merged[0][1] if merged[0][0] == 0 else 0
+
+    def _init_plan(self, head: bytes) -> None:
+        from ...gpu import hip
+
+        try:
+            self._gg = parse_bytes(bytes(head))
+        except Exception:
+            return  # header bigger than head capture: finish() covers it
+        self._h = hip()
+        self._stream = self._h.Stream(0)
+        self._quants = [t for t in self._gg.tensors
+                        if t.type_id in self.QUANT_IDS]
+        out_bytes = sum(t.n_elems * 2 for t in self._quants)
+        self._arena = self._h.DeviceBuffer(max(out_bytes, 1))
+        off = 0
+        for t in self._quants:
+            self._offsets[t.name] = off
+            off += t.n_elems * 2
+
+    def _launch_covered(self) -> None:
+        if self._gg is None or self._buf is None:
+            return
+        while self._next < len(self._quants):
+            t = self._quants[self._next]
+            end = self._gg.data_offset + t.offset + t.nbytes
+            if end > self._prefix:
+                break
+            self._h.gguf_dequant(
+                t.type_id, self._buf.ptr + self._gg.data_offset + t.offset,
+                self._arena.ptr + self._offsets[t.name], t.n_blocks,
+                self._stream.handle)
+            self._next += 1
+
+    # -- pull hooks ---------------------------------------------------- #
+    def on_range(self, name, lo, hi, buf, head) -> None:
+        """Pull progress hook: [lo, hi) of `name` is resident in `buf`;
+        head is the first bytes (non-None when the range covers 0)."""
+        with self._lock:
+            if head is not None and self._gg is None:
+                self._init_plan(head)
+            if getattr(buf, "ptr", None) is not None:
+                self._buf = buf
+            self._add_range(lo, hi)
+            self._launch_covered()
+
+    def finish(self, blob) -> dict:
+        """Launch any remaining tensors, sync, return {name: bf16}."""
+        import torch
+
+        with self._lock:
+            if self._gg is None:
+                self._init_plan(blob.head)
+                if self._gg is None:
+                    raise ValueError("GGUF header did not parse")
+            self._early = self._next
+            self._buf = blob.buffer
+            self._prefix = blob.nbytes
+            self._launch_covered()
+            assert self._next == len(self._quants)
+        self._stream.sync()
+        self._gg.blob = blob
+        u8 = torch.from_dlpack(self._arena.to_dlpack())
+        out = {}
+        for t in self._gg.tensors:
+            if t.type_id in self.QUANT_IDS:
+                o = self._offsets[t.name]
+                out[t.name] = (u8[o:o + t.n_elems * 2]
+                               .view(torch.bfloat16).view(t.dims[::-1]))
+            else:
+                out[t.name] = dequant_tensor_gpu(self._gg, t,
+                                                 stream=self._stream)
+        return out
+
+    @property
+    def launched_early(self) -> int:
+        """How many tensors launched before finish() (observability)."""
+        return getattr(self, "_early", self._next if self._gg else 0)
+
+
 def dequant_all_gpu(gg: GGUFModel, stream=None) -> dict:
     """Dequantize every tensor of a landed GGUF blob -> {name: bf16
     torch tensor}.  One output arena, all launches async on one stream

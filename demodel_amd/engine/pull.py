@@ -200,7 +200,13 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
                cafile, insecure, headers=None,
                expected_digests: bytes | None = None,
                seg_executor: cf.ThreadPoolExecutor | None = None,
-               verify_chunk: int | None = None) -> PulledFile:
+               verify_chunk: int | None = None,
+               on_range=None) -> PulledFile:
+    """on_range(name, lo, hi, buf, head_or_None): progress hook fired
+    as byte ranges become resident (per segment for range-parallel
+    pulls, once at completion otherwise) — lets consumers start GPU
+    work (e.g. GGUF dequant) on the landed prefix while the rest still
+    downloads."""
     t0 = time.perf_counter()
     do_verify = verify in ("chunked", "digest", "gpu-digest")
     want_segments = (seg_executor is not None
@@ -219,7 +225,8 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
             if total > SEGMENT_MIN:
                 blob = _pull_segmented(landers, url, total, src,
                                        expected_digests, cafile, insecure,
-                                       headers, seg_executor, verify_chunk)
+                                       headers, seg_executor, verify_chunk,
+                                       name=name, on_range=on_range)
             else:
                 blob = landers.get().land(
                     src.fill, min(total, SEGMENT_MIN),
@@ -272,6 +279,8 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
                     for s2 in extra_sources:
                         s2.close()
         nbytes = blob.nbytes
+        if on_range is not None and not getattr(blob, "_ranged", False):
+            on_range(name, 0, nbytes, blob.buffer, blob.head)
     finally:
         src.close()
     etag = src.resp.get("x-linked-etag") or src.resp.get("etag")
@@ -297,7 +306,8 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
 
 def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
                     expected_digests, cafile, insecure, headers,
-                    seg_executor, verify_chunk=None) -> "object":
+                    seg_executor, verify_chunk=None, name=None,
+                    on_range=None) -> "object":
     """Range-parallel landing of one blob: segment 0 comes from the
     already-open 206 stream, the rest are parallel range GETs, all landing
     into disjoint ranges of one HBM buffer through per-thread pinned
@@ -350,6 +360,8 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
             finally:
                 s.close()
         lander.sync()
+        if on_range is not None:
+            on_range(name, lo, hi, buf, None)
         return 0.0
 
     futs = [seg_executor.submit(land_range, i) for i in range(1, n_segs)]
@@ -385,11 +397,14 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
             finally:
                 s.close()
     lander0.sync()
+    if on_range is not None:
+        on_range(name, 0, bounds[1], buf, bytes(head))
     for f in futs:
         f.result()
     vc = verify_chunk or lander0.verify_chunk
     blob = LandedBlob(nbytes=total, device=f"cuda:{lander0.device_index}",
                       buffer=buf, verify_chunk=vc, head=bytes(head))
+    blob._ranged = True  # on_range already fired per segment
     blob.digest_blob = lander0._gpu_chunk_digests(buf, total, vc)
     if expected_digests is not None:
         from .pipeline import check_digests
@@ -424,7 +439,8 @@ def pull_hf_stream(repo: str, rev: str = "main",
                    landers: LanderPool | None = None,
                    slab_bytes: int = 32 << 20,
                    digest_map: dict[str, bytes] | None = None,
-                   peer_verify: bool = False, batched: bool = False):
+                   peer_verify: bool = False, batched: bool = False,
+                   on_range=None):
     """Streaming pull: returns (info, names, generator) where the
     generator yields each PulledFile AS IT FINISHES landing, so a
     consumer (e.g. stream_dataset's GPU decompression) overlaps with the
@@ -466,7 +482,7 @@ def pull_hf_stream(repo: str, rev: str = "main",
                         _pull_blob, landers, n,
                         f"{endpoint}/{repo}/resolve/{rev}/{n}",
                         None, verify, cafile, insecure, None,
-                        exp, seg_ex, vc)] = n
+                        exp, seg_ex, vc, on_range)] = n
                 try:
                     if batched:
                         left = set(futs)
@@ -496,16 +512,17 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             landers: LanderPool | None = None,
             slab_bytes: int = 32 << 20,
             digest_map: dict[str, bytes] | None = None,
-            peer_verify: bool = False) -> PullResult:
+            peer_verify: bool = False, on_range=None) -> PullResult:
     """peer_verify: when `endpoint` is another demodel node, fetch its
     recorded chunk digests per blob and GPU-verify the pull against them
-    (verified distribution)."""
+    (verified distribution).  on_range: progress hook, see _pull_blob."""
     t0 = time.perf_counter()
     info, _, gen = pull_hf_stream(
         repo, rev, endpoint=endpoint, device_index=device_index,
         workers=workers, verify=verify, cafile=cafile, insecure=insecure,
         patterns=patterns, landers=landers, slab_bytes=slab_bytes,
-        digest_map=digest_map, peer_verify=peer_verify)
+        digest_map=digest_map, peer_verify=peer_verify,
+        on_range=on_range)
     result = PullResult(spec=f"hf://{repo}@{rev}")
     result.files = list(gen)
     result.files.sort(key=lambda f: f.name)

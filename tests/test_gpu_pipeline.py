@@ -138,7 +138,7 @@ def test_gpu_peer_verified_pull(stack, tmp_path):
                            verify="chunked", workers=1, peer_verify=True)
     f = res.files[0]
     assert f.blob.device.startswith("cuda")
-    assert f.blob.verify_chunk == 1 << 20
+    assert f.blob.verify_chunk == 64 << 10
     assert bytes(f.blob.torch_u8().cpu().numpy().tobytes()) == data
 
 
@@ -177,6 +177,43 @@ def test_gpu_dequant_all(stack, tmp_path):
             want = want.to(torch.bfloat16).float()
         else:
             want = want.to(torch.bfloat16).float()
+        mask = torch.isfinite(want)
+        assert torch.allclose(got[mask], want[mask], rtol=1 / 64,
+                              atol=1e-3), t.name
+
+
+def test_gpu_progressive_dequant(stack, tmp_path, monkeypatch):
+    """ProgressiveDequant: dequant launches fed by on_range events from
+    a segmented pull match the CPU reference."""
+    _require_gpu()
+    import torch
+
+    import demodel_amd.engine.pull as pm
+    from demodel_amd.engine.formats import gguf
+
+    monkeypatch.setattr(pm, "SEGMENT_MIN", 1 << 20)  # force segments
+    gg_path = tmp_path / "prog.gguf"
+    gguf.build_file(str(gg_path), [
+        ("t0.weight", (256, 512), 12),   # q4_K ~1.1 MB
+        ("t1.weight", (256, 512), 14),   # q6_K ~1.6 MB
+        ("t2.weight", (64, 1024), 2),    # q4_0
+        ("n.weight", (128,), 0),         # f32
+    ])
+    stack.origin.add_hf_repo("org/prog", {"m.gguf": str(gg_path)})
+    pd = gguf.ProgressiveDequant()
+    res = pm.pull_hf("org/prog", endpoint=stack.endpoint,
+                     verify="chunked", workers=2,
+                     patterns=["*.gguf"], on_range=pd.on_range)
+    sd = pd.finish(res.files[0].blob)
+    assert set(sd) == {"t0.weight", "t1.weight", "t2.weight", "n.weight"}
+    raw = gg_path.read_bytes()
+    gg = gguf.parse_bytes(raw[:1 << 20])
+    for t in gg.tensors:
+        got = sd[t.name].float().cpu().reshape(-1)
+        want = torch.from_numpy(gguf.dequant_cpu(
+            t.type_id, raw[gg.data_offset + t.offset:
+                           gg.data_offset + t.offset + t.nbytes],
+            t.n_elems)).to(torch.bfloat16).float()
         mask = torch.isfinite(want)
         assert torch.allclose(got[mask], want[mask], rtol=1 / 64,
                               atol=1e-3), t.name
