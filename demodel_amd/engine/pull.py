@@ -542,9 +542,15 @@ def pull_ollama(name: str, tag: str = "latest",
                 out_dir: str | None = None, cafile=None,
                 insecure: bool = False,
                 landers: LanderPool | None = None,
-                dequant: bool = True) -> PullResult:
+                dequant: bool = True,
+                dequant_tensors: bool = False) -> PullResult:
     """Pull an Ollama model: manifest + layer blobs by sha256 digest
-    (protocol shape per reference CONTRIBUTING.md:127-153)."""
+    (protocol shape per reference CONTRIBUTING.md:127-153).
+
+    dequant=True parses the GGUF header into meta["gguf_model"].
+    dequant_tensors=True (GPU) additionally dequantizes every tensor to
+    bf16 — overlapped with the download via ProgressiveDequant — into
+    meta["tensors"] ({name: torch bf16 tensor})."""
     endpoint = (endpoint or os.environ.get("OLLAMA_REGISTRY")
                 or OLLAMA_DEFAULT_ENDPOINT).rstrip("/")
     if "/" not in name:
@@ -561,6 +567,11 @@ def pull_ollama(name: str, tag: str = "latest",
 
     seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
               if have_gpu() and verify in ("chunked", "off") else None)
+    pd = None
+    if dequant and dequant_tensors and have_gpu():
+        from .formats.gguf import ProgressiveDequant
+
+        pd = ProgressiveDequant(device_index)
     with cf.ThreadPoolExecutor(max_workers=workers) as ex:
         futs = {}
         for layer in layers:
@@ -568,9 +579,13 @@ def pull_ollama(name: str, tag: str = "latest",
             url = f"{endpoint}/v2/{name}/blobs/{digest}"
             expected = digest.split(":", 1)[1] \
                 if digest.startswith("sha256:") else None
+            is_model = (layer.get("mediaType")
+                        == "application/vnd.ollama.image.model")
             futs[ex.submit(_pull_blob, landers, digest, url, expected,
                            verify, cafile, insecure, None, None,
-                           seg_ex)] = layer
+                           seg_ex, None,
+                           pd.on_range if (pd and is_model) else None)
+                 ] = layer
         for fut in cf.as_completed(futs):
             pf = fut.result()
             pf.name = futs[fut].get("mediaType", pf.name)
@@ -594,6 +609,8 @@ def pull_ollama(name: str, tag: str = "latest",
                 "types": sorted({t.type_name for t in gg.tensors}),
             }
             result.meta["gguf_model"] = gg
+            if pd is not None:
+                result.meta["tensors"] = pd.finish(model_layers[0].blob)
     if out_dir:
         for f in result.files:
             safe = f.name.replace("/", "_").replace(":", "_")
@@ -622,6 +639,7 @@ def pull_spec(spec: str, cfg: Config | None = None, endpoint=None,
         s = res.summary()
         s.pop("manifest", None)
         s.pop("gguf_model", None)
+        s.pop("tensors", None)
         return s
     raise ValueError(f"unknown spec {spec!r} (want hf:// or ollama://)")
 

@@ -217,3 +217,38 @@ def test_gpu_progressive_dequant(stack, tmp_path, monkeypatch):
         mask = torch.isfinite(want)
         assert torch.allclose(got[mask], want[mask], rtol=1 / 64,
                               atol=1e-3), t.name
+
+
+def test_gpu_pull_ollama_dequant_tensors(stack, tmp_path):
+    """pull_ollama(dequant_tensors=True): bf16 weights arrive with the
+    pull (ProgressiveDequant under the hood)."""
+    _require_gpu()
+    import torch
+
+    from demodel_amd.engine import pull as pull_mod
+    from demodel_amd.engine.formats import gguf
+
+    gg_path = tmp_path / "dt.gguf"
+    gguf.build_file(str(gg_path), [
+        ("w.weight", (256, 8), 12),
+        ("n.weight", (64,), 0),
+    ])
+    stack.origin.add_ollama_model("library/dt", "latest", [
+        ("application/vnd.ollama.image.model", str(gg_path)),
+    ])
+    res = pull_mod.pull_ollama("dt", "latest",
+                               endpoint=stack.origin_base,
+                               verify="chunked", workers=1,
+                               dequant_tensors=True)
+    sd = res.meta["tensors"]
+    assert set(sd) == {"w.weight", "n.weight"}
+    raw = gg_path.read_bytes()
+    gg = res.meta["gguf_model"]
+    t = [x for x in gg.tensors if x.name == "w.weight"][0]
+    got = sd["w.weight"].float().cpu().reshape(-1)
+    want = torch.from_numpy(gguf.dequant_cpu(
+        t.type_id, raw[gg.data_offset + t.offset:
+                       gg.data_offset + t.offset + t.nbytes],
+        t.n_elems)).to(torch.bfloat16).float()
+    mask = torch.isfinite(want)
+    assert torch.allclose(got[mask], want[mask], rtol=1 / 64, atol=1e-3)
