@@ -204,3 +204,20 @@ def test_gguf_parse_and_cpu_dequant(tmp_path):
         vals = gguf.dequant_cpu(t.type_id, data, t.n_elems)
         assert vals.shape == (t.n_elems,)
         assert np.isfinite(vals).all() or t.type_id in (1,)  # f16 may inf
+
+
+def test_on_range_fires_on_cpu_path(stack, tmp_path):
+    """The on_range progress hook fires once (whole blob) for plain
+    non-segmented pulls, with head bytes attached."""
+    data = os.urandom(200_000)
+    p = tmp_path / "r.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/rng", {"r.bin": str(p)})
+    events = []
+
+    def on_range(name, lo, hi, buf, head):
+        events.append((name, lo, hi, head is not None))
+
+    pull_mod.pull_hf("org/rng", endpoint=stack.origin_base,
+                     verify="chunked", workers=1, on_range=on_range)
+    assert events == [("r.bin", 0, len(data), True)]
