@@ -384,7 +384,11 @@ def main():
         res, _ = one_step(record_digests=(i == 0))
         log(f"warmup {i}: {res.seconds_to_ready:.2f}s "
             f"({res.gbps:.2f} GB/s)")
-        del res  # free HBM before the next pull (70B barely fits twice)
+        # recycle HBM buffers instead of free+realloc: near device
+        # capacity (70B = 141 GB twice on 288 GB) a fresh hipMalloc of
+        # just-freed pages costs SECONDS of driver page reclaim
+        landers.recycle(res)
+        del res
 
     sync()
     t0 = time.perf_counter()
@@ -392,6 +396,7 @@ def main():
     for i in range(args.steps):
         res, _ = one_step()
         per_step.append(res.seconds_to_ready)
+        landers.recycle(res)
         del res
     sync()
     elapsed = time.perf_counter() - t0

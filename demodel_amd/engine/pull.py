@@ -83,7 +83,13 @@ class PullResult:
 
 
 class LanderPool:
-    """One landing pipeline per worker thread (own pinned ring+streams)."""
+    """One landing pipeline per worker thread (own pinned ring+streams).
+
+    All landers share one BufferPool: re-allocating a just-freed large
+    HBM buffer costs seconds near device capacity (driver page
+    reclaim), so steady-state re-pulls recycle via
+    `pool.recycle(result)` — caller guarantees no outstanding tensor
+    views of the recycled blobs."""
 
     def __init__(self, device_index: int = 0, **kw):
         self._device_index = device_index
@@ -91,15 +97,38 @@ class LanderPool:
         self._local = threading.local()
         self._all: list = []
         self._lock = threading.Lock()
+        from .pipeline import BufferPool
+
+        self.buffer_pool = BufferPool()
 
     def get(self):
         lander = getattr(self._local, "lander", None)
         if lander is None:
-            lander = make_lander(self._device_index, **self._kw)
+            lander = make_lander(self._device_index,
+                                 buffer_pool=self.buffer_pool, **self._kw)
             self._local.lander = lander
             with self._lock:
                 self._all.append(lander)
         return lander
+
+    def recycle(self, result) -> int:
+        """Hand a PullResult's HBM blob buffers back for reuse; returns
+        the number of buffers recycled.  The caller must not touch the
+        result's blobs (or views of them) afterwards."""
+        n = 0
+        files = getattr(result, "files", None)
+        if files is None:
+            files = [result]
+        for f in files:
+            blob = getattr(f, "blob", f)
+            if getattr(blob, "device", "cpu") == "cpu":
+                continue
+            buf = blob.buffer
+            if buf is not None:
+                self.buffer_pool.put(buf, max(blob.nbytes, 1))
+                blob.buffer = None
+                n += 1
+        return n
 
 
 def blob_to_file(blob, path: str, chunk: int = 32 << 20) -> None:
@@ -315,8 +344,7 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     from .pipeline import LandedBlob
 
     lander0 = landers.get()
-    h = lander0._h
-    buf = h.DeviceBuffer(total)
+    buf = lander0.alloc(total)
     n_segs = min(MAX_SEGMENTS, (total + SEGMENT_MIN - 1) // SEGMENT_MIN)
     # segment 0 covers exactly what the already-open 206 stream serves
     # ([0, SEGMENT_MIN)); the rest of the file splits evenly

@@ -221,3 +221,14 @@ def test_on_range_fires_on_cpu_path(stack, tmp_path):
     pull_mod.pull_hf("org/rng", endpoint=stack.origin_base,
                      verify="chunked", workers=1, on_range=on_range)
     assert events == [("r.bin", 0, len(data), True)]
+
+
+def test_recycle_noop_on_cpu(stack, tmp_path):
+    data = os.urandom(10_000)
+    p = tmp_path / "c.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/cyc", {"c.bin": str(p)})
+    landers = pull_mod.LanderPool(0)
+    res = pull_mod.pull_hf("org/cyc", endpoint=stack.origin_base,
+                           verify="chunked", workers=1, landers=landers)
+    assert landers.recycle(res) == 0  # host blobs are not pooled

@@ -252,3 +252,32 @@ def test_gpu_pull_ollama_dequant_tensors(stack, tmp_path):
         t.n_elems)).to(torch.bfloat16).float()
     mask = torch.isfinite(want)
     assert torch.allclose(got[mask], want[mask], rtol=1 / 64, atol=1e-3)
+
+
+def test_gpu_buffer_recycle(stack, tmp_path):
+    """LanderPool.recycle: a re-pull of the same sizes reuses the HBM
+    buffers instead of free+realloc (near-capacity reallocs cost
+    seconds of driver page reclaim)."""
+    _require_gpu()
+    import os as _os
+
+    from demodel_amd.engine import pull as pull_mod
+
+    data = _os.urandom(2 << 20)
+    p = tmp_path / "cyc.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/cyc", {"cyc.bin": str(p)})
+    landers = pull_mod.LanderPool(0)
+    res = pull_mod.pull_hf("org/cyc", endpoint=stack.origin_base,
+                           verify="chunked", workers=1, landers=landers)
+    f = res.files[0]
+    ptr0 = f.blob.buffer.ptr
+    body0 = bytes(f.blob.torch_u8().cpu().numpy().tobytes())
+    assert body0 == data
+    assert landers.recycle(res) == 1
+    assert f.blob.buffer is None
+    res2 = pull_mod.pull_hf("org/cyc", endpoint=stack.origin_base,
+                            verify="chunked", workers=1, landers=landers)
+    f2 = res2.files[0]
+    assert f2.blob.buffer.ptr == ptr0  # same buffer came back
+    assert bytes(f2.blob.torch_u8().cpu().numpy().tobytes()) == data
