@@ -26,12 +26,37 @@ def _pool() -> cf.ThreadPoolExecutor:
     return _POOL
 
 
+_MEMFDS: dict[int, int] = {}
+
+
+def _pattern_memfd(pattern: bytes) -> int | None:
+    """A memfd holding `pattern` (created once per pattern object):
+    sendfile from it serves tiled virtual blobs with ZERO user-space
+    copies — a plain sock.send loop pays a user->kernel copy per byte
+    and capped the 141 GB bench ~27 GB/s vs sendfile's 39."""
+    key = id(pattern)
+    fd = _MEMFDS.get(key)
+    if fd is not None:
+        return fd
+    try:
+        fd = os.memfd_create("demodel-virtual-pattern")
+        off = 0
+        while off < len(pattern):
+            off += os.pwrite(fd, pattern[off:off + (8 << 20)], off)
+    except (AttributeError, OSError):
+        return None
+    _MEMFDS[key] = fd
+    return fd
+
+
 async def send_pattern_threaded(writer: asyncio.StreamWriter,
                                 pattern: bytes, start: int,
                                 length: int) -> None:
     """Send `length` bytes of an infinitely-tiled `pattern` beginning at
-    absolute offset `start`, straight from memory on a worker thread
-    (virtual blobs: benchmarks larger than the disk)."""
+    absolute offset `start`, on a worker thread (virtual blobs:
+    benchmarks larger than the disk).  Uses sendfile from a pattern
+    memfd (page-cache -> socket, no user-space copy); falls back to a
+    sock.send loop."""
     sock = writer.transport.get_extra_info("socket")
     tls = writer.transport.get_extra_info("sslcontext")
     if sock is None or tls is not None:
@@ -40,8 +65,24 @@ async def send_pattern_threaded(writer: asyncio.StreamWriter,
     writer.transport.pause_reading()
     mv = memoryview(pattern)
     plen = len(pattern)
+    memfd = _pattern_memfd(pattern)
+    out_fd = sock.fileno()
 
-    def run():
+    def run_sendfile():
+        sent_total = 0
+        while sent_total < length:
+            phase = (start + sent_total) % plen
+            want = min(plen - phase, length - sent_total)
+            try:
+                n = os.sendfile(out_fd, memfd, phase, want)
+            except BlockingIOError:
+                sel.select([], [out_fd], [], 10)
+                continue
+            if n == 0:
+                raise ConnectionResetError("peer went away")
+            sent_total += n
+
+    def run_send():
         sent_total = 0
         while sent_total < length:
             phase = (start + sent_total) % plen
@@ -60,7 +101,8 @@ async def send_pattern_threaded(writer: asyncio.StreamWriter,
 
     loop = asyncio.get_running_loop()
     try:
-        await loop.run_in_executor(_pool(), run)
+        await loop.run_in_executor(
+            _pool(), run_sendfile if memfd is not None else run_send)
     finally:
         try:
             writer.transport.resume_reading()
