@@ -399,13 +399,22 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
         uint32_t c_ll = 0, c_len = 0, c_dist = 0;
         bool have = false;
         for (uint32_t i = 0; i < nseq && !err; ++i) {
-          uint8_t ofc = sh.of_t.e[of_state].sym;
-          uint8_t mlc = sh.ml_t.e[ml_state].sym;
-          uint8_t llc = sh.ll_t.e[ll_state].sym;
+          // one LDS dword per table entry: exec_seq's asm memory
+          // clobbers would otherwise force re-loads of sym/base/nbits
+          FseEntry oe = sh.of_t.e[of_state];
+          FseEntry me = sh.ml_t.e[ml_state];
+          FseEntry le = sh.ll_t.e[ll_state];
+          uint8_t ofc = oe.sym, mlc = me.sym, llc = le.sym;
           if (ofc > 31 || mlc > 52 || llc > 35) { err = 1; break; }
           uint32_t ofv = (1u << ofc) + sq.get(ofc);
-          uint32_t n_len = kMLBase[mlc] + sq.get(kMLExtra[mlc]);
-          uint32_t n_ll = kLLBase[llc] + sq.get(kLLExtra[llc]);
+          // ml+ll extras in ONE bit read (<=32 bits): get(a)+get(b)
+          // composes as get(a+b) with the first read in the high bits
+          int mlx = kMLExtra[mlc], llx = kLLExtra[llc];
+          uint32_t ex = sq.get(mlx + llx);
+          uint32_t n_len = kMLBase[mlc] + (ex >> llx);
+          uint32_t n_ll = kLLBase[llc]
+                          + (ex & ((llx < 32) ? ((1u << llx) - 1)
+                                              : 0xFFFFFFFFu));
           uint32_t n_dist;
           if (ofv > 3) {
             n_dist = ofv - 3;
@@ -427,12 +436,13 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
             }
           }
           if (i + 1 < nseq) {
-            ll_state = sh.ll_t.e[ll_state].base
-                       + sq.get(sh.ll_t.e[ll_state].nbits);
-            ml_state = sh.ml_t.e[ml_state].base
-                       + sq.get(sh.ml_t.e[ml_state].nbits);
-            of_state = sh.of_t.e[of_state].base
-                       + sq.get(sh.of_t.e[of_state].nbits);
+            // the three state refreshes (<=9 bits each, ll|ml|of order)
+            // in ONE bit read: one refill check instead of three
+            int b_ll = le.nbits, b_ml = me.nbits, b_of = oe.nbits;
+            uint32_t bits = sq.get(b_ll + b_ml + b_of);
+            of_state = oe.base + (bits & ((1u << b_of) - 1));
+            ml_state = me.base + ((bits >> b_of) & ((1u << b_ml) - 1));
+            ll_state = le.base + (bits >> (b_of + b_ml));
           }
           // execute the PREVIOUS sequence while this decode's loads land
           if (have) {
