@@ -70,6 +70,7 @@ HIP_SRCS = [
     "gguf_dequant.hip",
     "inflate.hip",
     "zstd_kernel.hip",
+    "snappy.hip",
 ]
 
 

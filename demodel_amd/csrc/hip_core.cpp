@@ -55,6 +55,8 @@ void launch_inflate_streams(const uint64_t* desc, int n_streams,
                             int* status, hipStream_t stream);
 void launch_zstd_frames(const uint64_t* desc, int n_frames, int* status,
                         hipStream_t stream, int window);
+void launch_snappy_streams(const uint64_t* desc, int n_streams,
+                           hipStream_t stream);
 }
 
 // ------------------------------------------------------------------------
@@ -390,4 +392,13 @@ PYBIND11_MODULE(_hip, m) {
         "dst, dst_cap, written, status, consumed, ws), wave per frame; ws "
         "needs >= 144 KiB per frame; window 0=auto/16384/65536 selects "
         "the LDS window template");
+  m.def("snappy_streams",
+        [](uintptr_t desc, int n_streams, uintptr_t stream) {
+          launch_snappy_streams((const uint64_t*)desc, n_streams,
+                                (hipStream_t)stream);
+        },
+        py::call_guard<py::gil_scoped_release>(),
+        "snappy-decompress n_streams descriptors (8 u64 each: src, "
+        "src_len, dst, dst_cap, written, status, consumed, pad), wave "
+        "per stream (parquet's default page codec)");
 }

@@ -108,22 +108,40 @@ def _parse_results(desc: bytearray, n: int) -> list[InflateResult]:
 
 
 class ZstdJob:
-    """An in-flight GPU zstd batch on its own HIP stream: the launch
-    returns immediately so several batches (e.g. dataset shards landing
-    at different times) decode CONCURRENTLY, which keeps wave occupancy
-    high even when each batch alone has fewer frames than the chip has
-    wave slots.  pre_launch(stream_handle), when given, queues extra
-    async work (device copies) on the same stream before the kernel."""
+    """An in-flight GPU decompression batch on its own HIP stream: the
+    launch returns immediately so several batches (e.g. dataset shards
+    landing at different times) decode CONCURRENTLY, which keeps wave
+    occupancy high even when each batch alone has fewer frames than the
+    chip has wave slots.  pre_launch(stream_handle), when given, queues
+    extra async work (device copies) on the same stream before the
+    kernels.  snappy_frames run through the snappy kernel on the same
+    stream (parquet's default page codec); their results surface as
+    .snappy_results after wait()."""
 
     def __init__(self, frames: list[tuple[int, int, int, int]],
-                 pre_launch=None, window: int = 0):
+                 pre_launch=None, window: int = 0,
+                 snappy_frames: list[tuple[int, int, int, int]] | None
+                 = None):
         from ...gpu import hip
 
         h = hip()
         n = self._n = len(frames)
+        sn = self._sn = len(snappy_frames or ())
         self._s = h.Stream(0)
         if pre_launch is not None:
             pre_launch(self._s.handle)
+        if sn:
+            nd = sn * DESC_WORDS * 8
+            self._spin = h.PinnedPool(nd, 1)
+            sdesc = self._spin.slab_view(0)
+            for i, (src, slen, dst, cap) in enumerate(snappy_frames):
+                struct.pack_into("<8Q", sdesc, i * DESC_WORDS * 8,
+                                 src, slen, dst, cap, 0, 0, 0, 0)
+            self._sdbuf = h.DeviceBuffer(nd)
+            addr = self._spin.slab_ptr(0)
+            h.h2d_async(self._sdbuf.ptr, addr, nd, self._s.handle)
+            h.snappy_streams(self._sdbuf.ptr, sn, self._s.handle)
+            h.d2h_async(addr, self._sdbuf.ptr, nd, self._s.handle)
         if n:
             self._ws = h.DeviceBuffer(n * ZSTD_WS_BYTES)
             nd = n * DESC_WORDS * 8
@@ -155,24 +173,35 @@ class ZstdJob:
             self._results = (_parse_results(
                 bytearray(self._pin.slab_view(0)), self._n)
                 if self._n else [])
+            self.snappy_results = (_parse_results(
+                bytearray(self._spin.slab_view(0)), self._sn)
+                if self._sn else [])
         return self._results
 
-    def view(self, lo: int, n: int) -> "ZstdJobView":
-        return ZstdJobView(self, lo, n)
+    def view(self, lo: int, n: int, slo: int = 0,
+             sn: int = 0) -> "ZstdJobView":
+        return ZstdJobView(self, lo, n, slo, sn)
 
 
 class ZstdJobView:
-    """A contiguous frame-range of a (possibly shared) ZstdJob — several
-    shards coalesced into ONE launch each hold a view of it."""
+    """A frame-range of a (possibly shared) ZstdJob — several shards
+    coalesced into ONE launch each hold a view of it.  wait() returns
+    this view's zstd results followed by its snappy results."""
 
-    def __init__(self, job: ZstdJob, lo: int, n: int):
+    def __init__(self, job: ZstdJob, lo: int, n: int, slo: int = 0,
+                 sn: int = 0):
         self._job, self._lo, self._vn = job, lo, n
+        self._slo, self._svn = slo, sn
 
     def done(self) -> bool:
         return self._job.done()
 
     def wait(self) -> list[InflateResult]:
-        return self._job.wait()[self._lo:self._lo + self._vn]
+        res = self._job.wait()[self._lo:self._lo + self._vn]
+        if self._svn:
+            res = res + self._job.snappy_results[
+                self._slo:self._slo + self._svn]
+        return res
 
 
 def zstd_gpu(frames: list[tuple[int, int, int, int]],
@@ -236,3 +265,15 @@ def gunzip_blob_gpu(blob, out_size: int | None = None):
             continue
         raise IOError(f"GPU inflate failed: {res.error}")
     raise IOError("GPU inflate: capacity growth exhausted")
+
+
+def snappy_gpu(streams: list[tuple[int, int, int, int]]
+               ) -> list[InflateResult]:
+    """Decompress raw snappy streams on the GPU (csrc/snappy.hip).
+
+    streams: (src_ptr, src_len, dst_ptr, dst_cap) device addresses."""
+    if not streams:
+        return []
+    job = ZstdJob([], snappy_frames=streams)
+    job.wait()
+    return job.snappy_results

@@ -9,10 +9,10 @@ parquet library in the data path), and the GPU decompresses every page
 payload in one wave-parallel zstd/deflate launch into the HBM ring.
 
 Scope: v1 data pages + dictionary pages (what pyarrow writes by
-default); compressed codecs ZSTD and GZIP-less SNAPPY are NOT all
-supported — ZSTD pages go to the GPU, UNCOMPRESSED pass through, anything
-else fails loudly.  Data page v2 is rejected loudly (levels would need
-splitting).
+default); ZSTD and SNAPPY pages decompress on the GPU (csrc/
+zstd_kernel.hip, csrc/snappy.hip — snappy is parquet's default codec),
+UNCOMPRESSED pages device-copy, anything else fails loudly.  Data page
+v2 is rejected loudly (levels would need splitting).
 """
 
 from __future__ import annotations
@@ -323,14 +323,16 @@ def file_pages(path: str):
 
 def prep_pages_gpu(blob, pages, ring=None):
     """Build the decode plan for a landed parquet blob's pages: returns
-    (frames, copies, ring, spans) where frames feed a ZstdJob and copies
-    are (dst, src, n) device copies for UNCOMPRESSED pages."""
+    (frames, snappy, copies, ring, spans) where frames feed the zstd
+    kernel, snappy the snappy kernel, and copies are (dst, src, n)
+    device copies for UNCOMPRESSED pages."""
     from ...gpu import hip
 
     h = hip()
     total = sum(p.uncomp_size for _, p in pages)
     ring = ring or h.DeviceBuffer(max(total, 1))
     frames = []
+    snappy = []
     copies = []
     spans = []
     off = 0
@@ -338,15 +340,18 @@ def prep_pages_gpu(blob, pages, ring=None):
         if codec == CODEC_ZSTD:
             frames.append((blob.buffer.ptr + p.comp_offset, p.comp_size,
                            ring.ptr + off, p.uncomp_size))
+        elif codec == CODEC_SNAPPY:
+            snappy.append((blob.buffer.ptr + p.comp_offset, p.comp_size,
+                           ring.ptr + off, p.uncomp_size))
         elif codec == CODEC_UNCOMPRESSED:
             copies.append((ring.ptr + off,
                            blob.buffer.ptr + p.comp_offset, p.comp_size))
         else:
-            raise ValueError(f"GPU path supports ZSTD/UNCOMPRESSED, "
-                             f"got codec {codec}")
+            raise ValueError(f"GPU path supports ZSTD/SNAPPY/"
+                             f"UNCOMPRESSED, got codec {codec}")
         spans.append((off, p.uncomp_size))
         off += p.uncomp_size
-    return frames, copies, ring, spans
+    return frames, snappy, copies, ring, spans
 
 
 def launch_pages_gpu(blob, pages, ring=None):
@@ -358,7 +363,8 @@ def launch_pages_gpu(blob, pages, ring=None):
     from .compress import ZstdJob
 
     h = hip()
-    frames, copies, ring, spans = prep_pages_gpu(blob, pages, ring)
+    frames, snappy, copies, ring, spans = prep_pages_gpu(blob, pages,
+                                                         ring)
 
     def pre(handle):
         for dst, src, n in copies:
@@ -366,14 +372,16 @@ def launch_pages_gpu(blob, pages, ring=None):
 
     # 16 KiB LDS window: page batches run as concurrent jobs in
     # stream_dataset, so occupancy beats far-match locality
-    return ZstdJob(frames, pre_launch=pre, window=16 << 10), ring, spans
+    job = ZstdJob(frames, pre_launch=pre, window=16 << 10,
+                  snappy_frames=snappy)
+    return job, ring, spans
 
 
 def decompress_pages_gpu(blob, pages, ring=None):
     """Synchronous wrapper around launch_pages_gpu: returns
     (ring_buffer, [(out_offset, size)]) covering every page."""
     job, ring, spans = launch_pages_gpu(blob, pages, ring)
-    results = job.wait()
+    results = job.wait() + job.snappy_results
     bad = [(i, r) for i, r in enumerate(results) if not r.ok]
     if bad:
         raise IOError(f"GPU page decompress failed: {bad[:3]}")

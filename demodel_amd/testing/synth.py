@@ -165,9 +165,11 @@ def write_dataset_shards(out_dir: str, n_shards: int = 8,
 
 def write_parquet_shards(out_dir: str, n_shards: int = 8,
                          rows_per_shard: int = 300_000,
-                         reuse: bool = True) -> dict[str, str]:
-    """c4-en-like parquet shards (text/url/timestamp columns, ZSTD
-    pages) — BASELINE.json config 5's dataset."""
+                         reuse: bool = True,
+                         compression: str = "zstd") -> dict[str, str]:
+    """c4-en-like parquet shards (text/url/timestamp columns; ZSTD or
+    SNAPPY pages — snappy is parquet's default codec) — BASELINE.json
+    config 5's dataset."""
     import pyarrow as pa
     import pyarrow.parquet as pq
 
@@ -194,7 +196,7 @@ def write_parquet_shards(out_dir: str, n_shards: int = 8,
             "url": [f"https://example.com/{s}/{i}" for i in range(n)],
             "timestamp": np.arange(n, dtype=np.int64) + s,
         })
-        pq.write_table(table, path, compression="zstd",
+        pq.write_table(table, path, compression=compression,
                        data_page_version="1.0")
         files[fname] = path
     return files

@@ -96,7 +96,8 @@ def test_uncompressed_parquet_pages(tmp_path):
 
 
 @pytest.mark.gpu
-def test_gpu_page_decompress(tmp_path):
+@pytest.mark.parametrize("codec", ["zstd", "snappy"])
+def test_gpu_page_decompress(tmp_path, codec):
     import ctypes
 
     from demodel_amd.engine.pipeline import Lander
@@ -104,7 +105,7 @@ def test_gpu_page_decompress(tmp_path):
 
     assert have_gpu()
     p = tmp_path / "g.parquet"
-    _write_parquet(str(p), n_rows=50_000)
+    _write_parquet(str(p), n_rows=50_000, codec=codec)
     raw, pages = pqf.file_pages(str(p))
 
     pos = [0]
@@ -121,7 +122,7 @@ def test_gpu_page_decompress(tmp_path):
 
     h = hip()
     s = h.Stream(0)
-    c = pa.Codec("zstd")
+    c = pa.Codec(codec)
     total = sum(sz for _, sz in spans)
     out = bytearray(total)
     addr = ctypes.addressof((ctypes.c_char * total).from_buffer(out))
