@@ -8,11 +8,6 @@ import sys
 sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 from gpu_probe import zstd_bench  # noqa: E402
 
-if __name__ == "__main__":
-    for payload in ("words", "text", "random"):
-        zstd_bench(payload=payload)
-    for payload in ("words", "text", "random"):
-        snappy_bench(payload=payload)
 
 
 def snappy_bench(n_streams=2048, payload="words"):
@@ -60,3 +55,10 @@ def snappy_bench(n_streams=2048, payload="words"):
                       "MBps_per_wave": round(len(base) / dt / 1e6, 2),
                       "ratio": round(len(base) / len(comp), 2)}),
           flush=True)
+
+
+if __name__ == "__main__":
+    for payload in ("words", "text", "random"):
+        zstd_bench(payload=payload)
+    for payload in ("words", "text", "random"):
+        snappy_bench(payload=payload)
