@@ -11,8 +11,10 @@ payload in one wave-parallel zstd/deflate launch into the HBM ring.
 Scope: v1 data pages + dictionary pages (what pyarrow writes by
 default); ZSTD and SNAPPY pages decompress on the GPU (csrc/
 zstd_kernel.hip, csrc/snappy.hip — snappy is parquet's default codec),
-UNCOMPRESSED pages device-copy, anything else fails loudly.  Data page
-v2 is rejected loudly (levels would need splitting).
+UNCOMPRESSED pages device-copy, anything else fails loudly.  v2 data
+pages are handled: their rep+def levels (stored uncompressed at the
+front of the payload) device-copy and only the values region feeds the
+codec kernel.
 """
 
 from __future__ import annotations
@@ -119,9 +121,14 @@ class _Compact:
 @dataclass
 class PageInfo:
     page_type: int          # 0 data, 2 dictionary, 3 data_v2
-    comp_offset: int        # absolute offset of the compressed payload
+    comp_offset: int        # absolute offset of the page payload
     comp_size: int
     uncomp_size: int
+    # v2 data pages: rep+def levels are stored UNCOMPRESSED as the first
+    # lvl_bytes of the payload; only the rest is codec-compressed (and
+    # only when is_compressed)
+    lvl_bytes: int = 0
+    is_compressed: bool = True
 
 
 def parse_page_header(data, pos: int) -> tuple[PageInfo, int]:
@@ -129,6 +136,21 @@ def parse_page_header(data, pos: int) -> tuple[PageInfo, int]:
     c = _Compact(data, pos)
     page_type = None
     comp = uncomp = None
+    lvl = 0
+    is_comp = True
+
+    def parse_v2(cc):
+        nonlocal lvl, is_comp
+        for fid2, ft2 in cc.read_struct_fields():
+            if fid2 == 5:           # definition_levels_byte_length
+                lvl += cc.zigzag()
+            elif fid2 == 6:         # repetition_levels_byte_length
+                lvl += cc.zigzag()
+            elif fid2 == 7 and ft2 in (_CT_TRUE, _CT_FALSE):
+                is_comp = ft2 == _CT_TRUE
+            else:
+                cc.skip(ft2)
+
     for fid, ft in c.read_struct_fields():
         if fid == 1 and ft in (_CT_I32, _CT_BYTE, _CT_I16):
             page_type = c.zigzag() if ft != _CT_BYTE else c.byte()
@@ -136,13 +158,15 @@ def parse_page_header(data, pos: int) -> tuple[PageInfo, int]:
             uncomp = c.zigzag()
         elif fid == 3:
             comp = c.zigzag()
+        elif fid == 8 and ft == _CT_STRUCT:
+            parse_v2(c)             # DataPageHeaderV2
         else:
             c.skip(ft)
     if page_type is None or comp is None or uncomp is None:
         raise ValueError("malformed parquet PageHeader")
-    if page_type == 3:
-        raise ValueError("data page v2 not supported (levels split)")
-    return PageInfo(page_type, c.p, comp, uncomp), c.p
+    if page_type != 3:
+        lvl, is_comp = 0, True
+    return PageInfo(page_type, c.p, comp, uncomp, lvl, is_comp), c.p
 
 
 CODEC_UNCOMPRESSED = 0
@@ -337,15 +361,21 @@ def prep_pages_gpu(blob, pages, ring=None):
     spans = []
     off = 0
     for codec, p in pages:
-        if codec == CODEC_ZSTD:
-            frames.append((blob.buffer.ptr + p.comp_offset, p.comp_size,
-                           ring.ptr + off, p.uncomp_size))
-        elif codec == CODEC_SNAPPY:
-            snappy.append((blob.buffer.ptr + p.comp_offset, p.comp_size,
-                           ring.ptr + off, p.uncomp_size))
-        elif codec == CODEC_UNCOMPRESSED:
+        lvl = p.lvl_bytes
+        if lvl:
+            # v2: rep+def levels pass through uncompressed
             copies.append((ring.ptr + off,
-                           blob.buffer.ptr + p.comp_offset, p.comp_size))
+                           blob.buffer.ptr + p.comp_offset, lvl))
+        src = blob.buffer.ptr + p.comp_offset + lvl
+        dst = ring.ptr + off + lvl
+        clen = p.comp_size - lvl
+        ulen = p.uncomp_size - lvl
+        if not p.is_compressed or codec == CODEC_UNCOMPRESSED:
+            copies.append((dst, src, clen))
+        elif codec == CODEC_ZSTD:
+            frames.append((src, clen, dst, ulen))
+        elif codec == CODEC_SNAPPY:
+            snappy.append((src, clen, dst, ulen))
         else:
             raise ValueError(f"GPU path supports ZSTD/SNAPPY/"
                              f"UNCOMPRESSED, got codec {codec}")

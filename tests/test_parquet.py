@@ -15,7 +15,8 @@ import pyarrow.parquet as pq  # noqa: E402
 from demodel_amd.engine.formats import parquet as pqf  # noqa: E402
 
 
-def _write_parquet(path, n_rows=20_000, codec="zstd"):
+def _write_parquet(path, n_rows=20_000, codec="zstd",
+                   page_version="1.0"):
     import numpy as np
 
     rng = np.random.default_rng(3)
@@ -28,7 +29,7 @@ def _write_parquet(path, n_rows=20_000, codec="zstd"):
         "score": rng.random(n_rows),
     })
     pq.write_table(table, path, compression=codec,
-                   data_page_version="1.0")
+                   data_page_version=page_version)
     return table
 
 
@@ -95,9 +96,33 @@ def test_uncompressed_parquet_pages(tmp_path):
         assert info.comp_size == info.uncomp_size
 
 
+def _page_plain(raw, codec_name, info):
+    """CPU reconstruction of one decompressed page (v1 or v2)."""
+    c = pa.Codec(codec_name)
+    lvl = info.lvl_bytes
+    levels = raw[info.comp_offset:info.comp_offset + lvl]
+    body = raw[info.comp_offset + lvl:info.comp_offset + info.comp_size]
+    if not info.is_compressed:
+        return levels + body
+    return levels + bytes(c.decompress(body, info.uncomp_size - lvl))
+
+
+def test_v2_pages_parse_and_decode_cpu(tmp_path):
+    p = tmp_path / "v2.parquet"
+    _write_parquet(str(p), codec="zstd", page_version="2.0")
+    raw, pages = pqf.file_pages(str(p))
+    assert any(info.page_type == 3 for _, info in pages)
+    for codec, info in pages:
+        if info.page_type == 3:
+            assert 0 <= info.lvl_bytes < info.comp_size
+        plain = _page_plain(raw, "zstd", info)
+        assert len(plain) == info.uncomp_size
+
+
 @pytest.mark.gpu
 @pytest.mark.parametrize("codec", ["zstd", "snappy"])
-def test_gpu_page_decompress(tmp_path, codec):
+@pytest.mark.parametrize("page_version", ["1.0", "2.0"])
+def test_gpu_page_decompress(tmp_path, codec, page_version):
     import ctypes
 
     from demodel_amd.engine.pipeline import Lander
@@ -105,7 +130,8 @@ def test_gpu_page_decompress(tmp_path, codec):
 
     assert have_gpu()
     p = tmp_path / "g.parquet"
-    _write_parquet(str(p), n_rows=50_000, codec=codec)
+    _write_parquet(str(p), n_rows=50_000, codec=codec,
+                   page_version=page_version)
     raw, pages = pqf.file_pages(str(p))
 
     pos = [0]
@@ -122,13 +148,11 @@ def test_gpu_page_decompress(tmp_path, codec):
 
     h = hip()
     s = h.Stream(0)
-    c = pa.Codec(codec)
     total = sum(sz for _, sz in spans)
     out = bytearray(total)
     addr = ctypes.addressof((ctypes.c_char * total).from_buffer(out))
     h.d2h_async(addr, ring.ptr, total, s.handle)
     s.sync()
-    for (off, sz), (codec, info) in zip(spans, pages):
-        payload = raw[info.comp_offset:info.comp_offset + info.comp_size]
-        want = bytes(c.decompress(payload, info.uncomp_size))
+    for (off, sz), (_, info) in zip(spans, pages):
+        want = _page_plain(raw, codec, info)
         assert bytes(out[off:off + sz]) == want
