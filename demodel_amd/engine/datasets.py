@@ -60,7 +60,7 @@ def _prep_zst_frames_gpu(blob, idx):
                        ring.ptr + off, fr["decompressed"]))
         spans.append((off, fr["decompressed"]))
         off += fr["decompressed"]
-    return frames, [], [], ring, spans
+    return frames, [], [], [], ring, spans
 
 
 def _decompress_cpu(blob, idx):
@@ -134,6 +134,7 @@ def stream_dataset(repo: str, endpoint: str | None = None,
         out = []
         all_frames = []
         all_snappy = []
+        all_deflate = []
         all_copies = []
         entries = []
         for f in shards:
@@ -142,20 +143,22 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                     raise RuntimeError(
                         "parquet streaming needs a GPU (CPU fallback "
                         "covers .zst shards)")
-                frames, snappy, copies, ring, spans = _prep_parquet_gpu(
-                    f.blob)
+                (frames, snappy, deflate, copies, ring,
+                 spans) = _prep_parquet_gpu(f.blob)
             else:
                 idx = _sidecar_idx(f.sidecar)
                 if not gpu:
                     data, spans = _decompress_cpu(f.blob, idx)
                     out.append((f, None, data, spans))
                     continue
-                frames, snappy, copies, ring, spans = \
-                    _prep_zst_frames_gpu(f.blob, idx)
+                (frames, snappy, deflate, copies, ring,
+                 spans) = _prep_zst_frames_gpu(f.blob, idx)
             entries.append((f, len(all_frames), len(frames),
-                            len(all_snappy), len(snappy), ring, spans))
+                            len(all_snappy), len(snappy),
+                            len(all_deflate), len(deflate), ring, spans))
             all_frames += frames
             all_snappy += snappy
+            all_deflate += deflate
             all_copies += copies
         if entries:
             h = hip()
@@ -165,9 +168,11 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                     h.d2d_async(dst, src, n, handle)
 
             job = ZstdJob(all_frames, pre_launch=pre, window=16 << 10,
-                          snappy_frames=all_snappy)
-            for f, lo, n, slo, sn, ring, spans in entries:
-                out.append((f, job.view(lo, n, slo, sn), ring, spans))
+                          snappy_frames=all_snappy,
+                          deflate_frames=all_deflate)
+            for f, lo, n, slo, sn, dlo, dn, ring, spans in entries:
+                out.append((f, job.view(lo, n, slo, sn, dlo, dn),
+                            ring, spans))
         return out
 
     def finish(item):

@@ -121,15 +121,30 @@ class ZstdJob:
     def __init__(self, frames: list[tuple[int, int, int, int]],
                  pre_launch=None, window: int = 0,
                  snappy_frames: list[tuple[int, int, int, int]] | None
+                 = None,
+                 deflate_frames: list[tuple[int, int, int, int]] | None
                  = None):
         from ...gpu import hip
 
         h = hip()
         n = self._n = len(frames)
         sn = self._sn = len(snappy_frames or ())
+        dn = self._dn = len(deflate_frames or ())
         self._s = h.Stream(0)
         if pre_launch is not None:
             pre_launch(self._s.handle)
+        if dn:
+            nd = dn * DESC_WORDS * 8
+            self._dpin = h.PinnedPool(nd, 1)
+            ddesc = self._dpin.slab_view(0)
+            for i, (src, slen, dst, cap) in enumerate(deflate_frames):
+                struct.pack_into("<8Q", ddesc, i * DESC_WORDS * 8,
+                                 src, slen, dst, cap, 0, 0, 0, 0)
+            self._ddbuf = h.DeviceBuffer(nd)
+            addr = self._dpin.slab_ptr(0)
+            h.h2d_async(self._ddbuf.ptr, addr, nd, self._s.handle)
+            h.inflate_streams(self._ddbuf.ptr, dn, self._s.handle)
+            h.d2h_async(addr, self._ddbuf.ptr, nd, self._s.handle)
         if sn:
             nd = sn * DESC_WORDS * 8
             self._spin = h.PinnedPool(nd, 1)
@@ -176,11 +191,14 @@ class ZstdJob:
             self.snappy_results = (_parse_results(
                 bytearray(self._spin.slab_view(0)), self._sn)
                 if self._sn else [])
+            self.deflate_results = (_parse_results(
+                bytearray(self._dpin.slab_view(0)), self._dn)
+                if self._dn else [])
         return self._results
 
-    def view(self, lo: int, n: int, slo: int = 0,
-             sn: int = 0) -> "ZstdJobView":
-        return ZstdJobView(self, lo, n, slo, sn)
+    def view(self, lo: int, n: int, slo: int = 0, sn: int = 0,
+             dlo: int = 0, dn: int = 0) -> "ZstdJobView":
+        return ZstdJobView(self, lo, n, slo, sn, dlo, dn)
 
 
 class ZstdJobView:
@@ -189,9 +207,10 @@ class ZstdJobView:
     this view's zstd results followed by its snappy results."""
 
     def __init__(self, job: ZstdJob, lo: int, n: int, slo: int = 0,
-                 sn: int = 0):
+                 sn: int = 0, dlo: int = 0, dn: int = 0):
         self._job, self._lo, self._vn = job, lo, n
         self._slo, self._svn = slo, sn
+        self._dlo, self._dvn = dlo, dn
 
     def done(self) -> bool:
         return self._job.done()
@@ -201,6 +220,9 @@ class ZstdJobView:
         if self._svn:
             res = res + self._job.snappy_results[
                 self._slo:self._slo + self._svn]
+        if self._dvn:
+            res = res + self._job.deflate_results[
+                self._dlo:self._dlo + self._dvn]
         return res
 
 

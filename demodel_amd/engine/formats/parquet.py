@@ -129,6 +129,10 @@ class PageInfo:
     # only when is_compressed)
     lvl_bytes: int = 0
     is_compressed: bool = True
+    # first bytes of the VALUES region (payload after any v2 level
+    # prefix), captured during the header walk — GZIP pages need the
+    # member header parsed on CPU (RFC 1952)
+    head: bytes = b""
 
 
 def parse_page_header(data, pos: int) -> tuple[PageInfo, int]:
@@ -184,6 +188,8 @@ def column_chunk_pages(data, start: int, total_compressed: int
     end = start + total_compressed
     while pos < end:
         info, payload = parse_page_header(data, pos)
+        lv = payload + info.lvl_bytes
+        info.head = bytes(data[lv:lv + 64])
         pages.append(info)
         pos = payload + info.comp_size
     if pos != end:
@@ -309,8 +315,14 @@ def blob_pages(blob) -> list[tuple[int, "PageInfo"]]:
                 win_off = pos
                 win = d2h(pos, 64 << 10)
             info, payload = parse_page_header(win, pos - win_off)
-            info = PageInfo(info.page_type, payload + win_off,
-                            info.comp_size, info.uncomp_size)
+            abs_payload = payload + win_off
+            lv = payload + info.lvl_bytes
+            head = bytes(win[lv:lv + 64])
+            if len(head) < min(64, info.comp_size - info.lvl_bytes):
+                head = d2h(abs_payload + info.lvl_bytes, 64)
+            info = PageInfo(info.page_type, abs_payload,
+                            info.comp_size, info.uncomp_size,
+                            info.lvl_bytes, info.is_compressed, head)
             out.append((cm.codec, info))
             pos = info.comp_offset + info.comp_size
         if pos != end:
@@ -347,16 +359,19 @@ def file_pages(path: str):
 
 def prep_pages_gpu(blob, pages, ring=None):
     """Build the decode plan for a landed parquet blob's pages: returns
-    (frames, snappy, copies, ring, spans) where frames feed the zstd
-    kernel, snappy the snappy kernel, and copies are (dst, src, n)
-    device copies for UNCOMPRESSED pages."""
+    (frames, snappy, deflate, copies, ring, spans) — zstd / snappy /
+    raw-DEFLATE (gzip pages, member header parsed from PageInfo.head)
+    kernel inputs plus (dst, src, n) device copies for UNCOMPRESSED
+    pages and v2 level prefixes."""
     from ...gpu import hip
+    from .compress import gzip_deflate_offset
 
     h = hip()
     total = sum(p.uncomp_size for _, p in pages)
     ring = ring or h.DeviceBuffer(max(total, 1))
     frames = []
     snappy = []
+    deflate = []
     copies = []
     spans = []
     off = 0
@@ -376,12 +391,17 @@ def prep_pages_gpu(blob, pages, ring=None):
             frames.append((src, clen, dst, ulen))
         elif codec == CODEC_SNAPPY:
             snappy.append((src, clen, dst, ulen))
+        elif codec == CODEC_GZIP:
+            gz = gzip_deflate_offset(p.head)
+            # raw DEFLATE stream between the member header and the
+            # 8-byte crc32+isize trailer
+            deflate.append((src + gz, clen - gz - 8, dst, ulen))
         else:
-            raise ValueError(f"GPU path supports ZSTD/SNAPPY/"
+            raise ValueError(f"GPU path supports ZSTD/SNAPPY/GZIP/"
                              f"UNCOMPRESSED, got codec {codec}")
         spans.append((off, p.uncomp_size))
         off += p.uncomp_size
-    return frames, snappy, copies, ring, spans
+    return frames, snappy, deflate, copies, ring, spans
 
 
 def launch_pages_gpu(blob, pages, ring=None):
@@ -393,8 +413,8 @@ def launch_pages_gpu(blob, pages, ring=None):
     from .compress import ZstdJob
 
     h = hip()
-    frames, snappy, copies, ring, spans = prep_pages_gpu(blob, pages,
-                                                         ring)
+    frames, snappy, deflate, copies, ring, spans = prep_pages_gpu(
+        blob, pages, ring)
 
     def pre(handle):
         for dst, src, n in copies:
@@ -403,7 +423,7 @@ def launch_pages_gpu(blob, pages, ring=None):
     # 16 KiB LDS window: page batches run as concurrent jobs in
     # stream_dataset, so occupancy beats far-match locality
     job = ZstdJob(frames, pre_launch=pre, window=16 << 10,
-                  snappy_frames=snappy)
+                  snappy_frames=snappy, deflate_frames=deflate)
     return job, ring, spans
 
 
@@ -411,7 +431,7 @@ def decompress_pages_gpu(blob, pages, ring=None):
     """Synchronous wrapper around launch_pages_gpu: returns
     (ring_buffer, [(out_offset, size)]) covering every page."""
     job, ring, spans = launch_pages_gpu(blob, pages, ring)
-    results = job.wait() + job.snappy_results
+    results = job.wait() + job.snappy_results + job.deflate_results
     bad = [(i, r) for i, r in enumerate(results) if not r.ok]
     if bad:
         raise IOError(f"GPU page decompress failed: {bad[:3]}")

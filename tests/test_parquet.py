@@ -120,7 +120,7 @@ def test_v2_pages_parse_and_decode_cpu(tmp_path):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("codec", ["zstd", "snappy"])
+@pytest.mark.parametrize("codec", ["zstd", "snappy", "gzip"])
 @pytest.mark.parametrize("page_version", ["1.0", "2.0"])
 def test_gpu_page_decompress(tmp_path, codec, page_version):
     import ctypes
