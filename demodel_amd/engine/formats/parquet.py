@@ -8,13 +8,12 @@ CPU walks the headers (this module — a minimal Thrift compact reader, no
 parquet library in the data path), and the GPU decompresses every page
 payload in one wave-parallel zstd/deflate launch into the HBM ring.
 
-Scope: v1 data pages + dictionary pages (what pyarrow writes by
-default); ZSTD and SNAPPY pages decompress on the GPU (csrc/
-zstd_kernel.hip, csrc/snappy.hip — snappy is parquet's default codec),
-UNCOMPRESSED pages device-copy, anything else fails loudly.  v2 data
-pages are handled: their rep+def levels (stored uncompressed at the
-front of the payload) device-copy and only the values region feeds the
-codec kernel.
+Scope: the full pyarrow codec matrix — ZSTD (csrc/zstd_kernel.hip),
+SNAPPY (csrc/snappy.hip, parquet's default codec), GZIP (member header
+parsed on CPU, raw DEFLATE on csrc/inflate.hip) and UNCOMPRESSED
+(device copy) — for BOTH v1 and v2 data pages (v2 rep+def level
+prefixes device-copy; only the values region feeds the codec kernel).
+Unknown codecs fail loudly.
 """
 
 from __future__ import annotations
