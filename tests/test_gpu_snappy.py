@@ -108,3 +108,60 @@ def test_snappy_corrupt_fails_loudly(hipmod):
     dst = h.DeviceBuffer(4096)
     res = snappy_gpu([(src.ptr, len(comp), dst.ptr, 4096)])[0]
     assert not res.ok
+
+
+def _varint(n: int) -> bytes:
+    out = b""
+    while True:
+        b = n & 0x7F
+        n >>= 7
+        if n:
+            out += bytes([b | 0x80])
+        else:
+            return out + bytes([b])
+
+
+def _lit(data: bytes) -> bytes:
+    n = len(data) - 1
+    if n < 60:
+        return bytes([n << 2]) + data
+    return bytes([60 << 2, n & 0xFF]) + data  # 1-byte extended length
+
+
+def test_snappy_handcrafted_copy_tags(hipmod):
+    """Streams pyarrow's encoder never emits: copy2/copy4 with large
+    offsets, max-length copies, back-to-back overlapping runs."""
+    from demodel_amd.engine.formats.compress import snappy_gpu
+
+    h = hipmod
+    s = h.Stream(0)
+    base = bytes(range(256)) * 300           # 76800 bytes of literals
+    # copy2: 64 bytes from offset 70000; copy4: 33 from offset 76000
+    stream = (_lit(base)
+              + bytes([(63 << 2) | 2, 70000 & 0xFF, 70000 >> 8])
+              + bytes([(32 << 2) | 3]) + (76000).to_bytes(4, "little")
+              + bytes([(7 << 2) | 1 | (0 << 5), 1]))   # copy1 d=1 l=11
+    expect = bytearray(base)
+    expect += expect[70000:70000 + 64]
+    expect += expect[76000:76000 + 33]
+    expect += bytes([expect[-1]]) * 11       # overlapping d=1 run
+    payload = _varint(len(expect)) + stream
+    src = _upload(h, payload, s)
+    dst = h.DeviceBuffer(len(expect))
+    res = snappy_gpu([(src.ptr, len(payload), dst.ptr, len(expect))])[0]
+    assert res.ok, res.status
+    assert res.written == len(expect)
+    assert _download(h, dst, len(expect), s) == bytes(expect)
+
+
+def test_snappy_rejects_bad_offset(hipmod):
+    from demodel_amd.engine.formats.compress import snappy_gpu
+
+    h = hipmod
+    s = h.Stream(0)
+    # copy1 with offset 5 after only 4 bytes of output
+    payload = _varint(15) + _lit(b"abcd") + bytes([(7 << 2) | 1, 5])
+    src = _upload(h, payload, s)
+    dst = h.DeviceBuffer(64)
+    res = snappy_gpu([(src.ptr, len(payload), dst.ptr, 64)])[0]
+    assert not res.ok
