@@ -125,7 +125,8 @@ def _lit(data: bytes) -> bytes:
     n = len(data) - 1
     if n < 60:
         return bytes([n << 2]) + data
-    return bytes([60 << 2, n & 0xFF]) + data  # 1-byte extended length
+    nb = (n.bit_length() + 7) // 8            # 1..4 extended length bytes
+    return bytes([(59 + nb) << 2]) + n.to_bytes(nb, "little") + data
 
 
 def test_snappy_handcrafted_copy_tags(hipmod):
