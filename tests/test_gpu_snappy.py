@@ -137,14 +137,15 @@ def test_snappy_handcrafted_copy_tags(hipmod):
     h = hipmod
     s = h.Stream(0)
     base = bytes(range(256)) * 300           # 76800 bytes of literals
-    # copy2: 64 bytes from offset 60000; copy4: 33 from offset 76000
+    # copy2: 64 bytes from distance 60000; copy4: 33 from distance
+    # 76000 (offsets are distances BACK from the write position)
     stream = (_lit(base)
               + bytes([(63 << 2) | 2, 60000 & 0xFF, 60000 >> 8])
               + bytes([(32 << 2) | 3]) + (76000).to_bytes(4, "little")
               + bytes([(7 << 2) | 1 | (0 << 5), 1]))   # copy1 d=1 l=11
     expect = bytearray(base)
-    expect += expect[60000:60000 + 64]
-    expect += expect[76000:76000 + 33]
+    expect += expect[len(expect) - 60000:len(expect) - 60000 + 64]
+    expect += expect[len(expect) - 76000:len(expect) - 76000 + 33]
     expect += bytes([expect[-1]]) * 11       # overlapping d=1 run
     payload = _varint(len(expect)) + stream
     src = _upload(h, payload, s)
