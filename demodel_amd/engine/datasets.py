@@ -92,7 +92,8 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                    verify: str = "chunked",
                    landers: LanderPool | None = None,
                    digest_map: dict | None = None,
-                   on_file=None, inflight: int = 4, eager: bool = True):
+                   on_file=None, inflight: int = 4, eager: bool = True,
+                   repo_type: str = "model"):
     """Yield ShardBatch per data shard of an HF dataset repo.
 
     inflight bounds how many decompressed shard rings can be in flight
@@ -114,7 +115,8 @@ def stream_dataset(repo: str, endpoint: str | None = None,
     _, names, gen = pull_hf_stream(
         repo, endpoint=endpoint, workers=workers, verify=verify,
         patterns=list(patterns), device_index=device_index,
-        landers=landers, digest_map=digest_map, batched=True)
+        landers=landers, digest_map=digest_map, batched=True,
+        repo_type=repo_type)
     expected = set(names)
     gpu = have_gpu()
 

@@ -468,7 +468,7 @@ def pull_hf_stream(repo: str, rev: str = "main",
                    slab_bytes: int = 32 << 20,
                    digest_map: dict[str, bytes] | None = None,
                    peer_verify: bool = False, batched: bool = False,
-                   on_range=None):
+                   on_range=None, repo_type: str = "model"):
     """Streaming pull: returns (info, names, generator) where the
     generator yields each PulledFile AS IT FINISHES landing, so a
     consumer (e.g. stream_dataset's GPU decompression) overlaps with the
@@ -479,7 +479,11 @@ def pull_hf_stream(repo: str, rev: str = "main",
     unstarted pulls."""
     endpoint = (endpoint or os.environ.get("HF_ENDPOINT")
                 or HF_DEFAULT_ENDPOINT).rstrip("/")
-    info = fetch.get_json(f"{endpoint}/api/models/{repo}/revision/{rev}",
+    # HF dataset repos live under /api/datasets and resolve their blobs
+    # at /datasets/{repo}/resolve/... (models have no path prefix)
+    api = "datasets" if repo_type == "dataset" else "models"
+    prefix = "datasets/" if repo_type == "dataset" else ""
+    info = fetch.get_json(f"{endpoint}/api/{api}/{repo}/revision/{rev}",
                           cafile=cafile, insecure=insecure)
     names = [s["rfilename"] for s in info.get("siblings", [])]
     if patterns:
@@ -491,8 +495,9 @@ def pull_hf_stream(repo: str, rev: str = "main",
     def peer_expected(n):
         if not peer_verify:
             return None
-        return fetch_peer_digests(endpoint, f"/{repo}/resolve/{rev}/{n}",
-                                  cafile=cafile, insecure=insecure)
+        return fetch_peer_digests(
+            endpoint, f"/{prefix}{repo}/resolve/{rev}/{n}",
+            cafile=cafile, insecure=insecure)
 
     def gen():
         seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
@@ -508,7 +513,7 @@ def pull_hf_stream(repo: str, rev: str = "main",
                         exp, vc = pd
                     futs[ex.submit(
                         _pull_blob, landers, n,
-                        f"{endpoint}/{repo}/resolve/{rev}/{n}",
+                        f"{endpoint}/{prefix}{repo}/resolve/{rev}/{n}",
                         None, verify, cafile, insecure, None,
                         exp, seg_ex, vc, on_range)] = n
                 try:
@@ -540,17 +545,19 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             landers: LanderPool | None = None,
             slab_bytes: int = 32 << 20,
             digest_map: dict[str, bytes] | None = None,
-            peer_verify: bool = False, on_range=None) -> PullResult:
+            peer_verify: bool = False, on_range=None,
+            repo_type: str = "model") -> PullResult:
     """peer_verify: when `endpoint` is another demodel node, fetch its
     recorded chunk digests per blob and GPU-verify the pull against them
-    (verified distribution).  on_range: progress hook, see _pull_blob."""
+    (verified distribution).  on_range: progress hook, see _pull_blob.
+    repo_type: "model" or "dataset" (different HF URL layout)."""
     t0 = time.perf_counter()
     info, _, gen = pull_hf_stream(
         repo, rev, endpoint=endpoint, device_index=device_index,
         workers=workers, verify=verify, cafile=cafile, insecure=insecure,
         patterns=patterns, landers=landers, slab_bytes=slab_bytes,
         digest_map=digest_map, peer_verify=peer_verify,
-        on_range=on_range)
+        on_range=on_range, repo_type=repo_type)
     result = PullResult(spec=f"hf://{repo}@{rev}")
     result.files = list(gen)
     result.files.sort(key=lambda f: f.name)

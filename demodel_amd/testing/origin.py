@@ -6,7 +6,8 @@ the protocol shapes come from the reference's worked example —
 CONTRIBUTING.md:127-153 for the Ollama manifest — and the public HF Hub
 HTTP API as exercised by huggingface_hub):
 
-* HF:  ``GET /api/models/{repo}[/revision/{rev}]`` (repo info + siblings),
+* HF:  ``GET /api/{models|datasets}/{repo}[/revision/{rev}]`` (repo
+       info + siblings),
        ``GET|HEAD /{repo}/resolve/{rev}/{file}`` with
        ``X-Repo-Commit`` / ``ETag`` / ``Content-Length`` headers and an
        optional 302 hop to a /cdn/ path (mirrors the hub->CDN redirect),
@@ -180,7 +181,8 @@ class FakeOrigin:
 
         # ---- HF api: /api/models/{repo}/tree/{rev} (paginated file list,
         # what huggingface_hub 1.x snapshot_download walks) ----
-        if parts[:2] == ["api", "models"] and "tree" in parts:
+        if parts[:2] in (["api", "models"], ["api", "datasets"]) \
+                and "tree" in parts:
             i = parts.index("tree")
             repo_id = "/".join(parts[2:i])
             repo = self.hf_repos.get(repo_id)
@@ -201,7 +203,7 @@ class FakeOrigin:
                 [("Content-Type", "application/json")], body)
 
         # ---- HF api: /api/models/{repo}[/revision/{rev}] ----
-        if parts[:2] == ["api", "models"]:
+        if parts[:2] in (["api", "models"], ["api", "datasets"]):
             rest = parts[2:]
             rev = "main"
             if "revision" in rest:
@@ -228,7 +230,11 @@ class FakeOrigin:
         # ---- HF resolve: /{repo}/resolve/{rev}/{path...} ----
         if "resolve" in parts:
             i = parts.index("resolve")
-            repo_id = "/".join(parts[:i])
+            lead = parts[:i]
+            # dataset blobs resolve under /datasets/{org}/{name}/...
+            if lead and lead[0] == "datasets":
+                lead = lead[1:]
+            repo_id = "/".join(lead)
             rev = parts[i + 1] if i + 1 < len(parts) else "main"
             fname = "/".join(parts[i + 2:])
             repo = self.hf_repos.get(repo_id)

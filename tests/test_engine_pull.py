@@ -232,3 +232,21 @@ def test_recycle_noop_on_cpu(stack, tmp_path):
     res = pull_mod.pull_hf("org/cyc", endpoint=stack.origin_base,
                            verify="chunked", workers=1, landers=landers)
     assert landers.recycle(res) == 0  # host blobs are not pooled
+
+
+def test_pull_hf_dataset_repo_type(stack, tmp_path):
+    """HF dataset repos use /api/datasets + /datasets/.../resolve."""
+    data = os.urandom(30_000)
+    p = tmp_path / "shard.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/dset", {"shard.bin": str(p)})
+    res = pull_mod.pull_hf("org/dset", endpoint=stack.origin_base,
+                           verify="chunked", workers=1,
+                           repo_type="dataset")
+    f = res.files[0]
+    assert f.url.endswith("/datasets/org/dset/resolve/main/shard.bin") \
+        or "/datasets/org/dset/" in f.url or f.nbytes == len(data)
+    assert bytes(f.blob.buffer) == data
+    reqs = [r for r in stack.origin.requests if "datasets" in r]
+    assert any("/api/datasets/org/dset" in r for r in reqs)
+    assert any("/datasets/org/dset/resolve/" in r for r in reqs)
