@@ -662,6 +662,11 @@ def pull_spec(spec: str, cfg: Config | None = None, endpoint=None,
         kw.setdefault("landers", LanderPool(0, gpu=gpu))
     if spec.startswith("hf://"):
         body = spec[len("hf://"):]
+        # hf://datasets/org/name pulls a DATASET repo (different HF URL
+        # layout); anything else is a model repo
+        if body.startswith("datasets/"):
+            body = body[len("datasets/"):]
+            kw.setdefault("repo_type", "dataset")
         repo, _, rev = body.partition("@")
         res = pull_hf(repo, rev or "main", endpoint=endpoint,
                       out_dir=out_dir, **kw)

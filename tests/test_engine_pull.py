@@ -250,3 +250,15 @@ def test_pull_hf_dataset_repo_type(stack, tmp_path):
     reqs = [r for r in stack.origin.requests if "datasets" in r]
     assert any("/api/datasets/org/dset" in r for r in reqs)
     assert any("/datasets/org/dset/resolve/" in r for r in reqs)
+
+
+def test_pull_spec_dataset_prefix(stack, tmp_path):
+    """hf://datasets/org/name routes through the dataset URL layout."""
+    p = tmp_path / "d.bin"
+    p.write_bytes(b"y" * 5000)
+    stack.origin.add_hf_repo("org/dsp", {"d.bin": str(p)})
+    out = pull_mod.pull_spec("hf://datasets/org/dsp",
+                             endpoint=stack.origin_base)
+    assert out["total_bytes"] == 5000
+    assert any("/api/datasets/org/dsp" in r
+               for r in stack.origin.requests)
