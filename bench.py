@@ -51,7 +51,8 @@ TINY_GEOM = {"hidden": 256, "inter": 688, "layers": 4, "heads": 8,
              "kv_heads": 4, "vocab": 32000}
 
 
-def make_model_files(model: str, data_dir: str, n_shards_override=None):
+def make_model_files(model: str, data_dir: str, n_shards_override=None,
+                     parquet_codec="zstd"):
     from demodel_amd.testing import synth
 
     if model == "tiny":
@@ -59,7 +60,8 @@ def make_model_files(model: str, data_dir: str, n_shards_override=None):
     if model == "dataset":
         return synth.write_dataset_shards(data_dir)
     if model == "parquet":
-        return synth.write_parquet_shards(data_dir)
+        return synth.write_parquet_shards(data_dir,
+                                          compression=parquet_codec)
     if model in ("gguf-8b", "gguf-tiny"):
         geom = synth.LLAMA3_8B if model == "gguf-8b" else TINY_GEOM
         path = os.path.join(data_dir, "model.gguf")
@@ -95,6 +97,9 @@ def main():
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--shards", type=int, default=None,
                     help="override safetensors shard count")
+    ap.add_argument("--parquet-codec", default="zstd",
+                    choices=["zstd", "snappy", "gzip"],
+                    help="page codec for --model parquet")
     ap.add_argument("--virtual", action="store_true",
                     help="serve blobs from memory (no disk) — for models "
                          "bigger than the box's disk, e.g. llama3-70b; "
@@ -129,7 +134,9 @@ def main():
     data_dir = args.data_dir or os.path.join(
         os.environ.get("TMPDIR", "/tmp"),
         f"demodel_bench_{args.model}"
-        + (f"_s{args.shards}" if args.shards else ""))
+        + (f"_s{args.shards}" if args.shards else "")
+        + (f"_{args.parquet_codec}" if args.model == "parquet"
+           and args.parquet_codec != "zstd" else ""))
     t = time.time()
     lt = LoopThread()
     if args.virtual:
@@ -147,11 +154,13 @@ def main():
     else:
         # rank 0 generates the shared synthetic files; others wait
         if rank == 0:
-            files = make_model_files(args.model, data_dir, args.shards)
+            files = make_model_files(args.model, data_dir, args.shards,
+                                     args.parquet_codec)
         if dist:
             dist.barrier()
         if rank != 0:
-            files = make_model_files(args.model, data_dir, args.shards)
+            files = make_model_files(args.model, data_dir, args.shards,
+                                     args.parquet_codec)
         origin = FakeOrigin(data_dir, redirect_blobs=True)
         origin.add_hf_repo("bench/model", files)
         total_bytes = sum(os.path.getsize(p) for p in files.values())
