@@ -164,7 +164,14 @@ async def iter_body(reader: asyncio.StreamReader, mode: str, length: int):
             size_line = await reader.readline()
             if not size_line:
                 raise ProtocolError("chunked body truncated")
-            size = int(size_line.split(b";")[0].strip() or b"0", 16)
+            s = size_line.split(b";")[0].strip()
+            if not s:
+                raise ProtocolError(f"bad chunk size line {size_line!r}")
+            try:
+                size = int(s, 16)
+            except ValueError as e:
+                raise ProtocolError(
+                    f"bad chunk size line {size_line!r}") from e
             if size == 0:
                 # trailers until blank line
                 while True:
@@ -178,6 +185,10 @@ async def iter_body(reader: asyncio.StreamReader, mode: str, length: int):
                     raise ProtocolError("chunked body truncated")
                 remaining -= len(data)
                 yield data
+            # the CRLF that terminates this chunk's data
+            term = await reader.readline()
+            if term not in (b"\r\n", b"\n"):
+                raise ProtocolError(f"bad chunk terminator {term!r}")
             crlf = await reader.readexactly(2)
             if crlf != b"\r\n":
                 raise ProtocolError("bad chunk terminator")
