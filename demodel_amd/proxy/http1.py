@@ -186,9 +186,6 @@ async def iter_body(reader: asyncio.StreamReader, mode: str, length: int):
                 remaining -= len(data)
                 yield data
             # the CRLF that terminates this chunk's data
-            term = await reader.readline()
-            if term not in (b"\r\n", b"\n"):
-                raise ProtocolError(f"bad chunk terminator {term!r}")
             crlf = await reader.readexactly(2)
             if crlf != b"\r\n":
                 raise ProtocolError("bad chunk terminator")
