@@ -262,3 +262,37 @@ def test_pull_spec_dataset_prefix(stack, tmp_path):
     assert out["total_bytes"] == 5000
     assert any("/api/datasets/org/dsp" in r
                for r in stack.origin.requests)
+
+
+def test_buffer_pool_exact_size_recycling():
+    from demodel_amd.engine.pipeline import BufferPool
+
+    pool = BufferPool()
+    a, b = object(), object()
+    pool.put(a, 100)
+    pool.put(b, 200)
+    assert pool.take(50) is None         # exact-size only
+    assert pool.take(100) is a
+    assert pool.take(100) is None        # drained
+    assert pool.take(200) is b
+    pool.put(a, 100)
+    pool.clear()
+    assert pool.take(100) is None
+
+
+def test_progressive_dequant_range_merging():
+    """Pure range bookkeeping: out-of-order segments merge and the
+    contiguous prefix only advances once byte 0 is covered."""
+    from demodel_amd.engine.formats.gguf import ProgressiveDequant
+
+    pd = ProgressiveDequant()
+    pd._add_range(100, 200)
+    assert pd._prefix == 0               # no coverage from 0 yet
+    pd._add_range(300, 400)
+    pd._add_range(0, 100)
+    assert pd._prefix == 200             # [0,200) now contiguous
+    pd._add_range(200, 300)
+    assert pd._prefix == 400             # all merged
+    assert pd._ranges == [(0, 400)]
+    pd._add_range(50, 150)               # duplicate/overlap is harmless
+    assert pd._ranges == [(0, 400)]
