@@ -6,7 +6,7 @@ see cmd/demodel/main.go:59) rebuilt from scratch MI355X-first:
 
   * TLS-MITM / HF_ENDPOINT / OLLAMA_HOST-compatible proxy front-end
     (asyncio + per-host leaf certs minted natively via libcrypto),
-  * byte-compatible on-disk response cache (.cache/{sha256} + .meta,
+  * layout-compatible on-disk response cache (.cache/{key} + .meta,
     bodies kept in original Content-Encoding — reference
     CONTRIBUTING.md:53-153),
   * a GPU landing pipeline: chunked blob downloads stream through a

@@ -51,6 +51,16 @@ def read_or_new_ca(use_ecdsa: bool = False) -> CA:
         with open(key_path) as f:
             key_pem = f.read()
         return CA(cert_pem, key_pem)
+    if os.path.exists(crt_path) != os.path.exists(key_path):
+        # Silently re-minting over a half-present pair would break every
+        # client that trusts the surviving cert; make the operator decide.
+        have, missing = ((crt_path, key_path)
+                         if os.path.exists(crt_path)
+                         else (key_path, crt_path))
+        raise FileExistsError(
+            f"CA pair is half-present: {have} exists but {missing} is "
+            f"missing. Restore the missing file, or delete {have} to "
+            f"mint a fresh CA (clients trusting the old one will break).")
 
     from . import _native
 

@@ -1,4 +1,6 @@
-"""On-disk response cache, byte-compatible with the reference layout.
+"""On-disk response cache, layout-compatible with the reference
+(bodies byte-exact in original Content-Encoding; 16-hex key + .meta
+sidecar — the KEY DERIVATION differs, see below).
 
 Layout contract (reference CONTRIBUTING.md:53-153):
 
@@ -12,8 +14,12 @@ Layout contract (reference CONTRIBUTING.md:53-153):
   metadata with cat" workflow honest.
 
 ``key`` is 16 lowercase hex chars (64 bits) derived from the canonical
-request URI — the reference example key is ``1b8c2ef6c820e0c0``
-(CONTRIBUTING.md:57).  We use the first 8 bytes of SHA-256(uri).
+request URI — the same SHAPE as the reference example key
+``1b8c2ef6c820e0c0`` (CONTRIBUTING.md:57).  We use the first 8 bytes of
+SHA-256(uri); the reference's derivation is undocumented (none of
+sha256/md5/sha1/sha512/blake2b/sha3 over the documented URI reproduces
+its example), so an existing reference cache directory is NOT hit —
+the compatibility claim is layout, not key-for-key.
 
 Extensions beyond the reference (which never shipped cache code in Go —
 SURVEY.md §0):

@@ -64,6 +64,11 @@ class Config:
     mitm_hosts: list[str] = field(default_factory=lambda: list(DEFAULT_MITM_HOSTS))
 
     # --- listener ---
+    # 0.0.0.0:8080 mirrors the reference (start.go:206) for drop-in
+    # parity, but note what it means: an UNAUTHENTICATED forward proxy
+    # with a CONNECT tunnel reachable on every interface.  Set
+    # DEMODEL_HOST=127.0.0.1 on any machine with an untrusted network
+    # path; ProxyServer.start() logs a warning on wildcard binds.
     host: str = "0.0.0.0"
     port: int = 8080
 

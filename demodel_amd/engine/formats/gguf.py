@@ -450,6 +450,17 @@ class ProgressiveDequant:
         return getattr(self, "_early", self._next if self._gg else 0)
 
 
+def _check_extent(gg: GGUFModel, t: GGUFTensor) -> None:
+    """A crafted GGUF from an untrusted registry must not drive an
+    out-of-bounds HBM read: validate the tensor's byte extent against
+    the landed blob before any kernel launch."""
+    end = gg.data_offset + t.offset + t.nbytes
+    if t.offset < 0 or end > gg.blob.nbytes:
+        raise ValueError(
+            f"GGUF tensor {t.name!r} extent [{gg.data_offset + t.offset},"
+            f" {end}) exceeds blob size {gg.blob.nbytes}")
+
+
 def dequant_all_gpu(gg: GGUFModel, stream=None) -> dict:
     """Dequantize every tensor of a landed GGUF blob -> {name: bf16
     torch tensor}.  One output arena, all launches async on one stream
@@ -462,6 +473,8 @@ def dequant_all_gpu(gg: GGUFModel, stream=None) -> dict:
     own = stream is None
     stream = stream or h.Stream(0)
     quants = [t for t in gg.tensors if t.type_id in (2, 8, 12, 14)]
+    for t in quants:
+        _check_extent(gg, t)
     out_bytes = sum(t.n_elems * 2 for t in quants)
     arena = h.DeviceBuffer(max(out_bytes, 1))
     offsets = {}
@@ -497,6 +510,7 @@ def dequant_tensor_gpu(gg: GGUFModel, t: GGUFTensor, stream=None):
     blob = gg.blob
     assert blob is not None and blob.device != "cpu", \
         "GPU dequant needs a GPU-landed blob"
+    _check_extent(gg, t)
     src = blob.buffer.ptr + gg.data_offset + t.offset
     own_stream = stream is None
     stream = stream or h.Stream(0)

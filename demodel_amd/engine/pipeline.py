@@ -251,6 +251,23 @@ class Lander:
         self.copy_stream.sync()
         self.verify_stream.sync()
 
+    def read_head(self, buf, nbytes: int) -> bytes:
+        """First min(nbytes, head_bytes) bytes of a landed buffer, read
+        back D2H — the authoritative head after Range-resumed landings
+        (a resumed stream starts past byte 0, so the head captured
+        in-flight is incomplete)."""
+        import ctypes
+
+        n = min(nbytes, self.head_bytes)
+        if n <= 0:
+            return b""
+        out = bytearray(n)
+        addr = ctypes.addressof((ctypes.c_char * n).from_buffer(out))
+        self.copy_stream.sync()
+        self._h.d2h_async(addr, buf.ptr, n, self.copy_stream.handle)
+        self.copy_stream.sync()
+        return bytes(out)
+
     def alloc(self, nbytes: int):
         if self.buffer_pool is not None:
             buf = self.buffer_pool.take(max(nbytes, 1))
@@ -408,6 +425,9 @@ class HostLander:
         return b"".join(
             hashlib.sha256(mv[o:o + vc]).digest()
             for o in range(0, nbytes, vc))
+
+    def read_head(self, buf, nbytes: int) -> bytes:
+        return bytes(memoryview(buf)[:min(nbytes, self.head_bytes)])
 
     def sync(self) -> None:
         pass

@@ -210,6 +210,12 @@ def _land_with_resume(lander, open_fn, nbytes: int, verify: bool,
                      landed, nbytes, attempt)
             fill = open_fn(landed)
     lander.sync()
+    # A resumed landing captured its head from the resume offset (or not
+    # at all — land_into drops its partial head when it raises); rebuild
+    # from the landed buffer so header parsing (safetensors/GGUF)
+    # survives mid-stream drops.
+    if attempt > 0 or len(head) != min(nbytes, lander.head_bytes):
+        head = lander.read_head(buf, nbytes)
     vc = verify_chunk or lander.verify_chunk
     device = ("cpu" if isinstance(buf, bytearray)
               else f"cuda:{lander.device_index}")
@@ -400,7 +406,7 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
                                     file_size=total, keep_head=True)
     except LandingError as e:
         # the probe stream died mid-segment-0: finish it with range GETs
-        head = bytearray()
+        head = None  # rebuilt from the landed buffer below
         at = e.landed
         attempt = 0
         while at < bounds[1]:
@@ -425,6 +431,8 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
             finally:
                 s.close()
     lander0.sync()
+    if head is None:
+        head = lander0.read_head(buf, total)
     if on_range is not None:
         on_range(name, 0, bounds[1], buf, bytes(head))
     for f in futs:

@@ -99,6 +99,12 @@ class ProxyServer:
         )
         self.port = self._server.sockets[0].getsockname()[1]
         log.info("listening on %s:%d", self.cfg.host, self.port)
+        if self.cfg.host in ("0.0.0.0", "::"):
+            log.warning(
+                "listening on %s: this is an unauthenticated forward "
+                "proxy reachable from every interface (reference-parity "
+                "default). Set DEMODEL_HOST=127.0.0.1 unless the network "
+                "is trusted.", self.cfg.host)
         return self.port
 
     async def close(self) -> None:

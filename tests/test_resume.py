@@ -36,12 +36,36 @@ def test_pull_resumes_after_drop(stack, tmp_path):
     assert f.blob.sha256 == hashlib.sha256(data).hexdigest()
     assert f.digest_ok is True
     assert bytes(f.blob.buffer) == data
+    # blob.head must survive the resume (it is rebuilt from the landed
+    # buffer — the in-flight capture restarts at the resume offset)
+    assert f.blob.head == data[:len(f.blob.head)]
+    assert len(f.blob.head) == len(data)  # head_bytes default > 2 MiB
     assert not stack.origin.drop_once  # fault fired
     # origin saw the blob requested twice (original + resume)
     blob_gets = [r for r in stack.origin.requests
                  if r.startswith("GET") and "big.bin" in r
                  and "/cdn/" in r]
     assert len(blob_gets) == 2
+
+
+def test_resume_past_head_keeps_head(stack, tmp_path):
+    """Drop AFTER head_bytes: the resumed stream never sees byte 0, yet
+    blob.head must still be the true file prefix (ADVICE round-1
+    medium: head was lost / captured from the resume offset)."""
+    from demodel_amd.engine.pull import LanderPool
+
+    data = os.urandom(2 << 20)
+    p = tmp_path / "late.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/late", {"late.bin": str(p)})
+    stack.origin.drop_once["late.bin"] = 1 << 20  # drop at 1 MiB
+    landers = LanderPool(0, slab_bytes=128 << 10, head_bytes=64 << 10)
+    res = pull_mod.pull_hf("org/late", endpoint=stack.origin_base,
+                           verify="digest", workers=1, landers=landers)
+    f = res.files[0]
+    assert f.blob.sha256 == hashlib.sha256(data).hexdigest()
+    assert f.blob.head == data[:64 << 10]
+    assert not stack.origin.drop_once
 
 
 def test_pull_fails_when_retries_exhausted(stack, tmp_path, monkeypatch):
@@ -103,6 +127,9 @@ class TestSegmentedResume:
         assert not stack.origin.drop_once  # fault fired
         got = bytes(f.blob.torch_u8().cpu().numpy().tobytes())
         assert got == data
+        # seg-0 fallback must not lose the head (rebuilt via D2H)
+        assert f.blob.head == data[:len(f.blob.head)]
+        assert len(f.blob.head) == len(data)  # < head_bytes default
         del res, f
 
     def test_drop_in_segment0(self, stack, tmp_path, monkeypatch):
