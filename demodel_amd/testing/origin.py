@@ -48,10 +48,14 @@ class FakeOrigin:
         self.port: int | None = None
         self._server = None
         self.requests: list[str] = []  # log of "<METHOD> <path>"
+        self.connections = 0           # accepted TCP connections
         # fault injection: path-substring -> serve N body bytes then drop
         # the connection (consumed on first match) — exercises the
         # engine's Range-resume path
         self.drop_once: dict[str, int] = {}
+        # fault injection: path-substrings that hang (no response bytes)
+        # until the client gives up — exercises proxy read timeouts
+        self.hang_once: set[str] = set()
 
     # ------------------------------------------------------------------ #
     # content registration
@@ -131,13 +135,18 @@ class FakeOrigin:
             await self._server.wait_closed()
 
     async def _handle(self, reader, writer):
+        self.connections += 1
         try:
             while True:
                 head = await http1.read_request_head(reader)
                 if head is None:
                     return
                 self.requests.append(f"{head.method} {head.target}")
-                await self._dispatch(head, writer)
+                for key in list(self.hang_once):
+                    if key in head.target:
+                        self.hang_once.discard(key)
+                        await asyncio.sleep(3600)  # until client drops
+                await self._dispatch(head, writer, reader)
                 if head.get("connection", "").lower() == "close":
                     return
         except (http1.ProtocolError, ConnectionResetError,
@@ -150,9 +159,25 @@ class FakeOrigin:
             except Exception:
                 pass
 
-    async def _dispatch(self, head: RequestHead, writer):
+    async def _dispatch(self, head: RequestHead, writer, reader=None):
         path = head.target.split("?")[0]
         parts = [p for p in path.split("/") if p]
+
+        # ---- POST/PUT echo (proxy pass-through tests): digest the
+        # streamed request body, reply with its size + sha256 ----
+        if parts == ["echo"] and head.method in ("POST", "PUT"):
+            mode, length = http1.body_mode(head, method=head.method)
+            h = hashlib.sha256()
+            n = 0
+            if reader is not None and mode != "none":
+                async for chunk in http1.iter_body(reader, mode, length):
+                    h.update(chunk)
+                    n += len(chunk)
+            body = json.dumps({"bytes": n,
+                               "sha256": h.hexdigest()}).encode()
+            return await self._reply(
+                writer, head, 200,
+                [("Content-Type", "application/json")], body)
 
         # ---- Ollama registry v2 ----
         if parts and parts[0] == "v2":

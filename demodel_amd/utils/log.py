@@ -28,14 +28,30 @@ def get_logger(name: str) -> logging.Logger:
 
 
 class TransferLog:
-    """Accumulates per-transfer byte counts / timings for metrics."""
+    """Per-transfer metrics: exact aggregate counters plus a BOUNDED
+    ring of recent records (a long-running proxy must not leak one dict
+    per request — round-1 ADVICE/VERDICT finding)."""
 
-    def __init__(self):
-        self.records: list[dict] = []
+    MAX_RECORDS = 4096
+
+    def __init__(self, max_records: int = MAX_RECORDS):
+        from collections import deque
+
+        self.records: "deque[dict]" = deque(maxlen=max_records)
+        # event -> (count, bytes); exact over the proxy's whole lifetime
+        self.counts: dict[str, int] = {}
+        self.bytes: dict[str, int] = {}
 
     def record(self, **kw) -> None:
         kw.setdefault("t", time.time())
+        ev = kw.get("event", "other")
+        self.counts[ev] = self.counts.get(ev, 0) + 1
+        self.bytes[ev] = self.bytes.get(ev, 0) + kw.get("bytes", 0)
         self.records.append(kw)
 
+    @property
+    def n_requests(self) -> int:
+        return sum(self.counts.values())
+
     def total_bytes(self) -> int:
-        return sum(r.get("bytes", 0) for r in self.records)
+        return sum(self.bytes.values())

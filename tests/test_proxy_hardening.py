@@ -1,0 +1,122 @@
+"""Round-2 proxy hardening: upstream keep-alive pooling, read timeouts,
+bounded transfer log, strict /__demodel routing, streaming PUT/POST
+pass-through, off-loop cache fills (VERDICT round-1 items 2 and 7)."""
+
+import hashlib
+import json
+import os
+import urllib.request
+
+import pytest
+
+from demodel_amd.utils.log import TransferLog
+from helpers import Stack
+
+
+@pytest.fixture()
+def stack(tmp_path):
+    s = Stack(tmp_path)
+    yield s
+    s.close()
+
+
+def _get(url, timeout=20, method="GET", data=None, headers=None):
+    req = urllib.request.Request(url, method=method, data=data,
+                                 headers=headers or {})
+    return urllib.request.urlopen(req, timeout=timeout)
+
+
+def test_transfer_log_is_bounded():
+    tl = TransferLog(max_records=8)
+    for i in range(100):
+        tl.record(event="miss", bytes=10)
+    for i in range(50):
+        tl.record(event="hit", bytes=5)
+    assert len(tl.records) == 8          # ring stays bounded
+    assert tl.n_requests == 150          # aggregates stay exact
+    assert tl.counts == {"miss": 100, "hit": 50}
+    assert tl.bytes == {"miss": 1000, "hit": 250}
+    assert tl.total_bytes() == 1250
+
+
+def test_strict_demodel_routing(stack):
+    with pytest.raises(urllib.error.HTTPError) as ei:
+        _get(f"{stack.endpoint}/__demodel/bogus")
+    assert ei.value.code == 404
+    with pytest.raises(urllib.error.HTTPError) as ei:
+        _get(f"{stack.endpoint}/__demodel/statsfoo")
+    assert ei.value.code == 404
+    with _get(f"{stack.endpoint}/__demodel/stats") as r:
+        obj = json.loads(r.read())
+    assert "upstream_pool" in obj
+
+
+def test_upstream_keepalive_reuse(stack, tmp_path):
+    """Sequential proxied requests reuse ONE upstream connection
+    (reference/goproxy behavior; round 1 opened one per request)."""
+    files = {}
+    for i in range(5):
+        p = tmp_path / f"f{i}.bin"
+        p.write_bytes(os.urandom(10_000))
+        files[f"f{i}.bin"] = str(p)
+    stack.origin.add_hf_repo("org/ka", files)
+    for i in range(5):
+        with _get(f"{stack.endpoint}/org/ka/resolve/main/f{i}.bin") as r:
+            r.read()
+    # each pull is 2 upstream requests (resolve 302 + cdn GET) but they
+    # should all ride a handful of pooled connections, not 10
+    assert stack.origin.connections <= 2, stack.origin.connections
+    with _get(f"{stack.endpoint}/__demodel/stats") as r:
+        obj = json.loads(r.read())
+    assert obj["upstream_pool"]["reused"] >= 8
+
+
+def test_upstream_read_timeout_504(stack, tmp_path, monkeypatch):
+    import demodel_amd.proxy.server as server_mod
+
+    monkeypatch.setattr(server_mod, "READ_TIMEOUT", 0.5)
+    p = tmp_path / "h.bin"
+    p.write_bytes(b"x" * 100)
+    stack.origin.add_hf_repo("org/hang", {"h.bin": str(p)})
+    stack.origin.hang_once.add("h.bin")
+    with pytest.raises(urllib.error.HTTPError) as ei:
+        _get(f"{stack.endpoint}/org/hang/resolve/main/h.bin", timeout=30)
+    assert ei.value.code == 504
+    # the fault is consumed; the next request succeeds
+    with _get(f"{stack.endpoint}/org/hang/resolve/main/h.bin") as r:
+        assert r.read() == b"x" * 100
+
+
+def test_streaming_post_passthrough(stack):
+    """A PUT/POST body larger than the buffer threshold streams through
+    the proxy to the origin without truncation."""
+    body = os.urandom(3 << 20)  # 3 MiB > _REQ_BUFFER_MAX
+    with _get(f"{stack.endpoint}/echo", method="POST", data=body,
+              headers={"Content-Type": "application/octet-stream"}) as r:
+        obj = json.loads(r.read())
+    assert obj["bytes"] == len(body)
+    assert obj["sha256"] == hashlib.sha256(body).hexdigest()
+
+
+def test_small_post_passthrough(stack):
+    body = b"hello world" * 10
+    with _get(f"{stack.endpoint}/echo", method="POST", data=body) as r:
+        obj = json.loads(r.read())
+    assert obj["bytes"] == len(body)
+    assert obj["sha256"] == hashlib.sha256(body).hexdigest()
+
+
+def test_cache_fill_still_correct_off_loop(stack, tmp_path):
+    """Cache fills now run on a worker thread; the entry must still be
+    byte-exact and replayable with the origin down."""
+    data = os.urandom(2 << 20)
+    p = tmp_path / "c.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/cw", {"c.bin": str(p)})
+    url = f"{stack.endpoint}/org/cw/resolve/main/c.bin"
+    with _get(url) as r:
+        assert r.read() == data
+    stack.stop_origin()
+    with _get(url) as r:
+        assert r.read() == data
+        assert r.headers["X-Demodel-Cache"] == "HIT"
