@@ -819,6 +819,13 @@ class ProxyServer:
             if cache_writer:
                 await cache_writer.abort()
             raise
+        # ---- hook 3: response hook ------------------------------------
+        # record BEFORE the cache finalize awaits its worker thread: the
+        # client has every byte already, and stats must reflect that
+        self.transfers.record(event="miss", uri=uri, status=resp.status,
+                              bytes=total)
+        log.info("MISS %s %s -> %d (%d bytes)", req.method, uri,
+                 resp.status, total)
         # HEAD responses carry no body: don't poison the cache with an
         # empty entry for a URI whose GET has content.
         if cache_writer:
@@ -827,11 +834,6 @@ class ProxyServer:
             else:
                 await cache_writer.finalize()
                 self._maybe_gc()
-        # ---- hook 3: response hook ------------------------------------
-        self.transfers.record(event="miss", uri=uri, status=resp.status,
-                              bytes=total)
-        log.info("MISS %s %s -> %d (%d bytes)", req.method, uri,
-                 resp.status, total)
         # Only fully-drained deterministic framings are safe to reuse
         # (HEAD responses carry no body by spec, so they qualify too).
         return (mode in ("none", "length", "chunked")
