@@ -148,6 +148,8 @@ class AsyncCacheWriter:
     Writes now queue to a dedicated worker thread; a bounded queue
     applies backpressure (awaited off-loop) instead of unbounded RAM."""
 
+    on_done = None  # set by ProxyServer._fill_writer (in-flight registry)
+
     def __init__(self, inner, max_queued: int = 64):
         from ..utils.netio import _pool
 
@@ -186,22 +188,31 @@ class AsyncCacheWriter:
     async def finalize(self) -> None:
         from ..utils.netio import _pool
 
-        await self._join()
-        if self._exc is not None:
-            log.warning("cache fill failed: %r; entry dropped", self._exc)
+        try:
+            await self._join()
+            if self._exc is not None:
+                log.warning("cache fill failed: %r; entry dropped",
+                            self._exc)
+                await asyncio.get_running_loop().run_in_executor(
+                    _pool(), self._inner.abort)
+                return
             await asyncio.get_running_loop().run_in_executor(
-                _pool(), self._inner.abort)
-            return
-        await asyncio.get_running_loop().run_in_executor(
-            _pool(), self._inner.finalize)
+                _pool(), self._inner.finalize)
+        finally:
+            if self.on_done:
+                self.on_done()
 
     async def abort(self) -> None:
         from ..utils.netio import _pool
 
-        self._exc = self._exc or asyncio.CancelledError()
-        await self._join()
-        await asyncio.get_running_loop().run_in_executor(
-            _pool(), self._inner.abort)
+        try:
+            self._exc = self._exc or asyncio.CancelledError()
+            await self._join()
+            await asyncio.get_running_loop().run_in_executor(
+                _pool(), self._inner.abort)
+        finally:
+            if self.on_done:
+                self.on_done()
 
 
 class ProxyServer:
@@ -221,6 +232,12 @@ class ProxyServer:
                                          digest_mode="async")
         self.transfers = TransferLog()
         self.upstreams = UpstreamPool()
+        # uri -> [refcount, Event]: cache fills in flight.  A lookup
+        # that misses but sees an in-flight fill WAITS for the commit
+        # instead of re-fetching upstream (dedups concurrent pulls of
+        # one blob, and makes "pull then re-pull hits the cache"
+        # deterministic now that fills commit off-loop).
+        self._filling: dict[str, list] = {}
         self._server: asyncio.AbstractServer | None = None
         self.port: int | None = None
         # reverse-mode routing table: path-prefix -> upstream base.
@@ -418,6 +435,43 @@ class ProxyServer:
                     u.hostname, port, is_tls, t
         raise ProtocolError(f"no reverse route for {t!r}")
 
+    def _fill_writer(self, uri: str, status: int, reason: str,
+                     headers) -> AsyncCacheWriter:
+        """Create an off-loop cache fill for `uri`, registered in the
+        in-flight table so concurrent lookups can wait on it."""
+        cw = AsyncCacheWriter(self.cache.writer(uri, status, reason,
+                                                headers))
+        slot = self._filling.get(uri)
+        if slot is None:
+            slot = [0, asyncio.Event()]
+            self._filling[uri] = slot
+        slot[0] += 1
+
+        def done():
+            slot[0] -= 1
+            if slot[0] <= 0:
+                slot[1].set()
+                if self._filling.get(uri) is slot:
+                    del self._filling[uri]
+
+        cw.on_done = done
+        return cw
+
+    async def _lookup_or_wait(self, lookup_uri: str):
+        """Cache lookup that waits for an in-flight fill of the same
+        URI to commit (bounded by READ_TIMEOUT) before giving up."""
+        hit = self.cache.lookup(lookup_uri)
+        if hit is not None:
+            return hit
+        slot = self._filling.get(lookup_uri)
+        if slot is None:
+            return None
+        try:
+            await asyncio.wait_for(slot[1].wait(), READ_TIMEOUT)
+        except asyncio.TimeoutError:
+            return None
+        return self.cache.lookup(lookup_uri)
+
     _gc_countdown = 0
 
     def _maybe_gc(self) -> None:
@@ -569,7 +623,7 @@ class ProxyServer:
         if head.method in ("GET", "HEAD"):
             lookup_uri = uri
             for _ in range(6):
-                hit = self.cache.lookup(lookup_uri)
+                hit = await self._lookup_or_wait(lookup_uri)
                 if hit is None:
                     break
                 if reverse_mode and 300 <= hit.status < 400:
@@ -720,8 +774,8 @@ class ProxyServer:
                     cw = None
                     if (head.method == "GET"
                             and self.cache.cacheable("GET", resp.status)):
-                        cw = AsyncCacheWriter(self.cache.writer(
-                            uri, resp.status, resp.reason, resp.headers))
+                        cw = self._fill_writer(uri, resp.status,
+                                               resp.reason, resp.headers)
                     try:
                         if mode != "none":
                             async for chunk in _timed_body(
@@ -774,8 +828,8 @@ class ProxyServer:
         cache_writer = None
         if (self.cache.cacheable(req.method, resp.status)
                 and req.get("range") is None):
-            cache_writer = AsyncCacheWriter(self.cache.writer(
-                uri, resp.status, resp.reason, resp.headers))
+            cache_writer = self._fill_writer(uri, resp.status,
+                                             resp.reason, resp.headers)
 
         out = ResponseHead("HTTP/1.1", resp.status, resp.reason,
                            [(k, v) for k, v in resp.headers
