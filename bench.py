@@ -118,12 +118,29 @@ def main():
         torch.cuda.set_device(local_rank)
 
     dist = None
-    if world > 1:
+    if world > 1 or args.mode != "dp":
+        # collective modes init the process group even at world 1 so the
+        # RCCL backend path (init, dtype, stream interaction with the
+        # landing pipeline) is exercised on a single-GPU box
+        import datetime
+
         import torch.distributed as dist_mod
 
         dist = dist_mod
+        # first-try-robust rendezvous on an unfamiliar node: loopback
+        # defaults (container hostnames may not resolve), explicit
+        # device binding for RCCL, and a bounded timeout so a wedged
+        # rank fails the job instead of hanging it
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29513")
+        os.environ.setdefault("RANK", "0")
+        os.environ.setdefault("WORLD_SIZE", str(world))
+        kw = {}
+        if have_gpu:
+            kw["device_id"] = torch.device("cuda", local_rank)
         dist.init_process_group(
-            backend="nccl" if have_gpu else "gloo")
+            backend="nccl" if have_gpu else "gloo",
+            timeout=datetime.timedelta(seconds=600), **kw)
 
     from demodel_amd.engine import pull as pull_mod
     from demodel_amd.engine.pull import LanderPool
@@ -181,8 +198,31 @@ def main():
             dist.barrier()
 
     digest_map: dict = {}
+    # pre-allocated scatter targets (name -> tensor), built during the
+    # first (untimed) step: the timed step then exercises the K4
+    # scatter_ranges kernel exactly as BASELINE config 2 words it
+    # ("GPU SHA256 + tensor scatter"), not just zero-copy views
+    scatter_targets: dict = {}
+
+    def _build_targets(res):
+        from demodel_amd.engine.formats import safetensors as st
+
+        dev = "cuda" if have_gpu else "cpu"
+        for f in res.files:
+            if not f.name.endswith(".safetensors"):
+                continue
+            hdr = st.parse_header(f.blob.head)
+            tg = {}
+            for t in hdr.tensors:
+                tg[t.name] = torch.empty(
+                    t.shape, dtype=getattr(torch, t.torch_dtype),
+                    device=dev)
+            scatter_targets[f.name] = (tg)
 
     def dp_step(record_digests=False):
+        from demodel_amd.engine.formats import safetensors as st
+        from demodel_amd.engine.loader import load_into
+
         res = pull_mod.pull_hf(
             "bench/model", endpoint=endpoint, workers=args.workers,
             verify=args.verify, landers=landers,
@@ -192,9 +232,19 @@ def main():
             for f in res.files:
                 if f.blob.digest_blob:
                     digest_map[f.name] = f.blob.digest_blob
-        # model-ready: materialize the tensor views (virtual payloads are
-        # patterned bytes, not parseable safetensors)
-        n_t = 0 if args.virtual else len(res.tensors())
+        # model-ready: tensors scattered into their final (preallocated)
+        # HBM addresses + zero-copy views materialized (virtual payloads
+        # are patterned bytes, not parseable safetensors)
+        n_t = 0
+        if not args.virtual:
+            if not scatter_targets:
+                _build_targets(res)
+            for f in res.files:
+                tg = scatter_targets.get(f.name)
+                if tg:
+                    hdr = st.parse_header(f.blob.head)
+                    n_t += len(load_into(f.blob, hdr, tg))
+            n_t += len(res.tensors())
         if have_gpu:
             torch.cuda.synchronize()
         return res, n_t
@@ -441,6 +491,7 @@ def main():
                 "bytes_per_model": total_bytes,
                 "files": len(files),
                 "verify": args.verify,
+                "scatter": bool(scatter_targets),
                 "seconds_to_ready": round(ms_per_step / 1000.0, 3),
                 "parallelism": (
                     f"independent-pull dp{world}" if args.mode == "dp"

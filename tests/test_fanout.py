@@ -26,6 +26,22 @@ def test_shard_assignment_balanced():
         assert p.owners == plans[0].owners
 
 
+def test_broadcast_order_deterministic_round_robin():
+    from demodel_amd.parallel.fanout import broadcast_order
+
+    files = [(f"f{i}", (7 * i + 3) % 11 * 1000 + 100) for i in range(9)]
+    plans = [shard_assignment(files, 4, r) for r in range(4)]
+    orders = [broadcast_order(p) for p in plans]
+    # identical on every rank (NCCL requires identical enqueue order)
+    assert all(o == orders[0] for o in orders)
+    assert sorted(orders[0]) == sorted(n for n, _ in files)
+    # round 0 has one file per owner that owns anything
+    owners = plans[0].owners
+    n_owners = len({r for r, _ in owners.values()})
+    head = [owners[n][0] for n in orders[0][:n_owners]]
+    assert len(set(head)) == n_owners
+
+
 def _free_port():
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
