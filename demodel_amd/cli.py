@@ -32,6 +32,8 @@ def _cmd_start(args) -> int:
     cfg = load_config()
     if args.port is not None:
         cfg.port = args.port
+    if getattr(args, "gpu_prefetch", None) is not None:
+        cfg.gpu_prefetch = args.gpu_prefetch
     try:
         asyncio.run(run_proxy(cfg))
     except KeyboardInterrupt:
@@ -115,6 +117,10 @@ def main(argv: list[str] | None = None) -> int:
 
     sp = sub.add_parser("start", help="run the caching proxy")
     sp.add_argument("--port", type=int, default=None)
+    sp.add_argument("--gpu-prefetch", default=None,
+                    choices=["off", "auto"],
+                    help="land proxy-cached blobs into HBM ahead of "
+                         "engine pulls (also DEMODEL_GPU_PREFETCH)")
     sp.set_defaults(fn=_cmd_start)
 
     ip = sub.add_parser("init", help="create or load the demodel CA")

@@ -85,6 +85,12 @@ class Config:
     pinned_slabs: int = 4                # pinned host ring depth per pull
     gpu_verify: str = "chain"            # "chain" | "chunked" | "off"
 
+    # --- proxy -> HBM pull-ahead ---
+    # "off": only POST /__demodel/prefetch lands blobs; "auto": every
+    # blob-looking response the proxy caches is landed into HBM
+    gpu_prefetch: str = "off"
+    gpu_cache_max_bytes: int | None = None  # HBM registry LRU budget
+
     @property
     def mitm_host_set(self) -> set[str]:
         return set(self.mitm_hosts)
@@ -120,6 +126,10 @@ def load_config(**overrides) -> Config:
         chunk_bytes=_env_int("DEMODEL_CHUNK_BYTES", 32 << 20),
         pinned_slabs=_env_int("DEMODEL_PINNED_SLABS", 4),
         gpu_verify=os.environ.get("DEMODEL_GPU_VERIFY", "chain"),
+        gpu_prefetch=os.environ.get("DEMODEL_GPU_PREFETCH", "off"),
+        gpu_cache_max_bytes=(
+            int(float(os.environ["DEMODEL_GPU_CACHE_MAX_GB"]) * 1e9)
+            if os.environ.get("DEMODEL_GPU_CACHE_MAX_GB") else None),
     )
     for k, v in overrides.items():
         setattr(cfg, k, v)

@@ -40,6 +40,7 @@ class LandedBlob:
     sha256: str | None = None         # exact whole-blob digest if computed
     head: bytes = b""                 # first bytes (for header parsing)
     timings: dict = field(default_factory=dict)
+    shared: bool = False              # registry-owned: recycle must skip
 
     @property
     def chunk_digests(self) -> list[str]:

@@ -123,6 +123,8 @@ class LanderPool:
             blob = getattr(f, "blob", f)
             if getattr(blob, "device", "cpu") == "cpu":
                 continue
+            if getattr(blob, "shared", False):
+                continue  # registry-owned (proxy prefetch): not ours
             buf = blob.buffer
             if buf is not None:
                 self.buffer_pool.put(buf, max(blob.nbytes, 1))
@@ -476,7 +478,8 @@ def pull_hf_stream(repo: str, rev: str = "main",
                    slab_bytes: int = 32 << 20,
                    digest_map: dict[str, bytes] | None = None,
                    peer_verify: bool = False, batched: bool = False,
-                   on_range=None, repo_type: str = "model"):
+                   on_range=None, repo_type: str = "model",
+                   registry=None):
     """Streaming pull: returns (info, names, generator) where the
     generator yields each PulledFile AS IT FINISHES landing, so a
     consumer (e.g. stream_dataset's GPU decompression) overlaps with the
@@ -507,13 +510,33 @@ def pull_hf_stream(repo: str, rev: str = "main",
             endpoint, f"/{prefix}{repo}/resolve/{rev}/{n}",
             cafile=cafile, insecure=insecure)
 
+    def reg_hit(n):
+        """HBM-registry short-circuit (proxy pull-ahead): the blob is
+        already landed and verified — serve it with zero fetch, zero
+        disk (engine/registry.py)."""
+        if registry is None:
+            return None
+        blob = registry.get(f"/{prefix}{repo}/resolve/{rev}/{n}")
+        if blob is None:
+            return None
+        if on_range is not None:
+            on_range(n, 0, blob.nbytes, blob.buffer, blob.head)
+        return PulledFile(
+            name=n, url=f"{endpoint}/{prefix}{repo}/resolve/{rev}/{n}",
+            nbytes=blob.nbytes, blob=blob, digest_ok=True, seconds=0.0)
+
     def gen():
         seg_ex = (cf.ThreadPoolExecutor(max_workers=max(workers, 4))
                   if have_gpu() else None)
         try:
             with cf.ThreadPoolExecutor(max_workers=workers) as ex:
                 futs = {}
+                hits = []
                 for n in names:
+                    pf = reg_hit(n)
+                    if pf is not None:
+                        hits.append(pf)
+                        continue
                     pd = peer_expected(n)
                     exp = (digest_map or {}).get(n)
                     vc = None
@@ -526,12 +549,15 @@ def pull_hf_stream(repo: str, rev: str = "main",
                         exp, seg_ex, vc, on_range)] = n
                 try:
                     if batched:
+                        if hits:
+                            yield hits
                         left = set(futs)
                         while left:
                             finished, left = cf.wait(
                                 left, return_when=cf.FIRST_COMPLETED)
                             yield [f.result() for f in finished]
                     else:
+                        yield from hits
                         for fut in cf.as_completed(futs):
                             yield fut.result()
                 except GeneratorExit:
@@ -554,18 +580,20 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             slab_bytes: int = 32 << 20,
             digest_map: dict[str, bytes] | None = None,
             peer_verify: bool = False, on_range=None,
-            repo_type: str = "model") -> PullResult:
+            repo_type: str = "model", registry=None) -> PullResult:
     """peer_verify: when `endpoint` is another demodel node, fetch its
     recorded chunk digests per blob and GPU-verify the pull against them
     (verified distribution).  on_range: progress hook, see _pull_blob.
-    repo_type: "model" or "dataset" (different HF URL layout)."""
+    repo_type: "model" or "dataset" (different HF URL layout).
+    registry: a BlobRegistry (proxy pull-ahead); files already landed in
+    HBM are served from it with zero fetch."""
     t0 = time.perf_counter()
     info, _, gen = pull_hf_stream(
         repo, rev, endpoint=endpoint, device_index=device_index,
         workers=workers, verify=verify, cafile=cafile, insecure=insecure,
         patterns=patterns, landers=landers, slab_bytes=slab_bytes,
         digest_map=digest_map, peer_verify=peer_verify,
-        on_range=on_range, repo_type=repo_type)
+        on_range=on_range, repo_type=repo_type, registry=registry)
     result = PullResult(spec=f"hf://{repo}@{rev}")
     result.files = list(gen)
     result.files.sort(key=lambda f: f.name)

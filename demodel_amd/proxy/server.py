@@ -217,9 +217,25 @@ class AsyncCacheWriter:
 
 class ProxyServer:
     def __init__(self, cfg: Config, leafs: LeafStore | None = None,
-                 cache: CacheStore | None = None):
+                 cache: CacheStore | None = None,
+                 prefetch_landers=None):
+        """prefetch_landers: a LanderPool; when set, blob responses the
+        proxy caches can be landed ahead into HBM (auto or via
+        POST /__demodel/prefetch) and registered in self.registry, so a
+        subsequent engine pull of the same path is HBM-warm with zero
+        upstream and zero disk reads in its hot path (the north star's
+        single proxy+GPU pipeline; reference hook contract
+        start.go:197-200)."""
         self.cfg = cfg
         self.leafs = leafs
+        self.prefetch_landers = prefetch_landers
+        self.registry = None
+        self._prefetching: set[str] = set()
+        if prefetch_landers is not None:
+            from ..engine.registry import BlobRegistry
+
+            self.registry = BlobRegistry(
+                max_bytes=getattr(cfg, "gpu_cache_max_bytes", None))
         # async digesting: blob fills must not run at hashlib speed; the
         # digests land in the meta a moment after commit.  64 KiB chunks:
         # a peer GPU-verifies pulls against this record with the
@@ -488,6 +504,132 @@ class ProxyServer:
         max_bytes = self.cfg.cache_max_bytes
         _pool().submit(lambda: self.cache.gc(max_bytes))
 
+    def _cached_entry_for_path(self, path: str):
+        """Resolve a request path to its cached entry, following the
+        cached redirect chain.  Returns (entry_or_None, final_uri)."""
+        try:
+            uri, *_ = self._canonical_uri(
+                RequestHead("GET", path, "HTTP/1.1", []), None)
+        except ProtocolError:
+            return None, None
+        hit = None
+        lookup_uri = uri
+        for _ in range(6):
+            hit = self.cache.lookup(lookup_uri)
+            if hit is None or not (300 <= hit.status < 400):
+                break
+            loc = dict((k.lower(), v) for k, v in hit.headers
+                       ).get("location")
+            if not loc:
+                break
+            lookup_uri = self._absolute_uri(loc, lookup_uri)
+        if hit is not None and 300 <= hit.status < 400:
+            return None, lookup_uri
+        return hit, lookup_uri
+
+    # ------------------------------------------------------------------ #
+    # GPU pull-ahead (proxy -> HBM unification)
+
+    _BLOB_SUFFIXES = (".safetensors", ".gguf", ".bin", ".pt", ".onnx",
+                      ".zst", ".parquet")
+
+    @classmethod
+    def _is_blob_path(cls, path: str) -> bool:
+        p = path.split("?")[0]
+        return p.endswith(cls._BLOB_SUFFIXES) or "/blobs/sha256:" in p
+
+    def _maybe_prefetch(self, path: str, force: bool = False) -> bool:
+        """Schedule HBM landing of a (cached) blob path; returns whether
+        a prefetch task was queued."""
+        if self.prefetch_landers is None or self.registry is None:
+            return False
+        if not force and getattr(self.cfg, "gpu_prefetch", "off") != "auto":
+            return False
+        if not force and not self._is_blob_path(path):
+            return False
+        if path in self._prefetching or path in self.registry:
+            return False
+        self._prefetching.add(path)
+        from ..utils.netio import _pool
+
+        loop = asyncio.get_running_loop()
+        fut = loop.run_in_executor(_pool(), self._prefetch_land, path)
+
+        def _done(f):
+            self._prefetching.discard(path)
+            e = f.exception()
+            if e:
+                log.warning("prefetch of %s failed: %r", path, e)
+
+        fut.add_done_callback(_done)
+        return True
+
+    def _prefetch_land(self, path: str) -> None:
+        """Worker thread: land a cached body into HBM (verified against
+        the cache's recorded chunk digests when present) and register
+        it.  Not-yet-cached paths are first pulled through our own
+        front door (which tees them into the cache)."""
+        hit, _ = self._cached_entry_for_path(path)
+        if hit is None:
+            from ..engine import fetch
+
+            src = fetch.http_get(f"http://127.0.0.1:{self.port}{path}")
+            try:
+                if src.status != 200:
+                    raise FileNotFoundError(
+                        f"prefetch GET {path} -> {src.status}")
+                sink = memoryview(bytearray(1 << 20))
+                while src.fill(sink) > 0:
+                    pass
+            finally:
+                src.close()
+            hit, _ = self._cached_entry_for_path(path)
+        if hit is None or hit.status != 200:
+            raise FileNotFoundError(f"{path} not cacheable for prefetch")
+        expected = None
+        vc = None
+        if hit.chunk_sha256 and hit.chunk_bytes:
+            expected = bytes.fromhex("".join(hit.chunk_sha256))
+            vc = hit.chunk_bytes
+        lander = self.prefetch_landers.get()
+        with hit.open_body() as f:
+            blob = lander.land(f.readinto, hit.body_size, verify=True,
+                               expected_digests=expected,
+                               verify_chunk=vc)
+        self.registry.put(path, blob)
+        log.info("prefetched %s -> %s (%d bytes, verified=%s)",
+                 path, blob.device, blob.nbytes, expected is not None)
+
+    async def _serve_prefetch(self, head: RequestHead, reader,
+                              writer) -> bool:
+        import json as _json
+
+        if self.registry is None:
+            return await self._simple(
+                writer, head, 503,
+                b'{"error": "prefetch landers not configured"}')
+        if head.method == "POST":
+            mode, length = http1.body_mode(head, method=head.method)
+            parts = []
+            async for chunk in http1.iter_body(reader, mode, length):
+                parts.append(chunk)
+            try:
+                obj = _json.loads(b"".join(parts) or b"{}")
+                paths = list(obj.get("paths", []))
+            except (ValueError, AttributeError):
+                return await self._simple(writer, head, 400,
+                                          b'{"error": "bad body"}')
+            queued = [p for p in paths if self._maybe_prefetch(
+                p, force=True)]
+            body = _json.dumps({"queued": queued}).encode()
+            return await self._simple(writer, head, 202, body)
+        body = _json.dumps({
+            "registered": self.registry.keys(),
+            "in_flight": sorted(self._prefetching),
+            **self.registry.stats(),
+        }).encode()
+        return await self._simple(writer, head, 200, body)
+
     async def _serve_digests(self, head: RequestHead, writer) -> bool:
         """GET /__demodel/digests/<path> — the cache's recorded per-chunk
         sha256 digests for the entry <path> resolves to (following the
@@ -532,8 +674,9 @@ class ProxyServer:
 
     async def _simple(self, writer, head, status: int,
                       body: bytes) -> bool:
-        out = ResponseHead("HTTP/1.1", status,
-                           "OK" if status == 200 else "Error",
+        reason = {200: "OK", 202: "Accepted", 404: "Not Found"}.get(
+            status, "Error")
+        out = ResponseHead("HTTP/1.1", status, reason,
                            [("Content-Type", "application/json"),
                             ("Content-Length", str(len(body))),
                             ("Connection", "keep-alive")])
@@ -595,6 +738,8 @@ class ProxyServer:
                 return await self._serve_digests(head, writer)
             if head.target.rstrip("/") == "/__demodel/stats":
                 return await self._serve_stats(head, writer)
+            if head.target.rstrip("/") == "/__demodel/prefetch":
+                return await self._serve_prefetch(head, reader, writer)
             return await self._simple(
                 writer, head, 404, b'{"error": "unknown endpoint"}')
         uri, host, port, is_tls, path = self._canonical_uri(head, tls_host)
@@ -643,6 +788,7 @@ class ProxyServer:
 
         redirects = 0
         carry: dict[str, str] = {}
+        orig_path = head.target  # pre-redirect path = the registry key
         while True:
             result = await self._forward_once(
                 head, req_body, uri, host, port, is_tls, path, writer,
@@ -650,6 +796,10 @@ class ProxyServer:
                 carry=carry, req_stream=req_stream, req_mode=req_mode,
             )
             if result is None:
+                # pull-ahead: a blob this proxy just cached can land in
+                # HBM now, so a later engine pull is served GPU-warm
+                if reverse_mode and head.method == "GET":
+                    self._maybe_prefetch(orig_path)
                 # a request body we never finished forwarding leaves the
                 # client connection desynced — close it
                 return client_wants_close or (
@@ -961,7 +1111,14 @@ async def run_proxy(cfg: Config) -> None:
 
     ca = read_or_new_ca(cfg.ca_use_ecdsa)
     leafs = LeafStore(ca)
-    srv = ProxyServer(cfg, leafs=leafs)
+    landers = None
+    if getattr(cfg, "gpu_prefetch", "off") != "off":
+        from ..engine.pull import LanderPool
+        from ..gpu import have_gpu
+
+        # HBM pull-ahead on GPU boxes; host-RAM registry elsewhere
+        landers = LanderPool(0, gpu=True if have_gpu() else False)
+    srv = ProxyServer(cfg, leafs=leafs, prefetch_landers=landers)
     await srv.start()
     assert srv._server is not None
     async with srv._server:
