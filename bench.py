@@ -100,6 +100,14 @@ def main():
     ap.add_argument("--parquet-codec", default="zstd",
                     choices=["zstd", "snappy", "gzip"],
                     help="page codec for --model parquet")
+    ap.add_argument("--via", default="direct",
+                    choices=["direct", "proxy", "proxy-miss"],
+                    help="direct: engine pulls from the origin (the "
+                         "headline path); proxy: pulls go THROUGH the "
+                         "demodel proxy with a primed cache (measures "
+                         "the cache-HIT data plane a real client sees); "
+                         "proxy-miss: caching disabled, measures the "
+                         "origin->proxy->client relay")
     ap.add_argument("--virtual", action="store_true",
                     help="serve blobs from memory (no disk) — for models "
                          "bigger than the box's disk, e.g. llama3-70b; "
@@ -186,6 +194,45 @@ def main():
 
     port = lt.call(origin.start())
     endpoint = f"http://127.0.0.1:{port}"
+
+    proxy = None
+    if args.via != "direct":
+        # the client-facing data plane (reference hot loop,
+        # start.go:201-204): engine pulls go THROUGH the proxy
+        from demodel_amd.config import Config
+        from demodel_amd.proxy.server import ProxyServer
+
+        pcfg = Config(host="127.0.0.1", port=0,
+                      cache_dir=os.path.join(data_dir,
+                                             f"proxycache_r{rank}"))
+        proxy = ProxyServer(pcfg)
+        proxy.reverse_routes = [("/", endpoint)]
+        if args.via == "proxy-miss":
+            proxy.cache.cacheable = lambda *a, **k: False
+        pport = lt.call(proxy.start())
+        endpoint = f"http://127.0.0.1:{pport}"
+        if args.via == "proxy":
+            # prime the cache (untimed) so timed pulls measure the HIT
+            # path: cache file -> sendfile -> engine -> HBM
+            import concurrent.futures as cf
+
+            from demodel_amd.engine import fetch
+
+            def prime(name):
+                src = fetch.http_get(
+                    f"{endpoint}/bench/model/resolve/main/{name}")
+                try:
+                    assert src.status == 200, (name, src.status)
+                    sink = memoryview(bytearray(8 << 20))
+                    while src.fill(sink) > 0:
+                        pass
+                finally:
+                    src.close()
+
+            t = time.time()
+            with cf.ThreadPoolExecutor(max_workers=8) as ex:
+                list(ex.map(prime, files))
+            log(f"proxy cache primed in {time.time() - t:.1f}s")
 
     landers = LanderPool(local_rank if have_gpu else 0,
                          slab_bytes=args.slab_mib << 20,
@@ -491,6 +538,7 @@ def main():
                 "bytes_per_model": total_bytes,
                 "files": len(files),
                 "verify": args.verify,
+                "via": args.via,
                 "scatter": bool(scatter_targets),
                 "seconds_to_ready": round(ms_per_step / 1000.0, 3),
                 "parallelism": (
@@ -501,6 +549,8 @@ def main():
         }
         print(json.dumps(out), flush=True)
 
+    if proxy is not None:
+        lt.call(proxy.close())
     lt.call(origin.close())
     lt.stop()
     if dist:

@@ -43,6 +43,32 @@ def test_bench_single_process(tmp_path):
     _check_line(line[0], world=1)
 
 
+def test_bench_via_proxy(tmp_path):
+    """--via proxy: engine pulls THROUGH the demodel proxy (primed
+    cache) — the client-facing data-plane measurement."""
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--model", "tiny", "--steps", "1",
+         "--warmup", "1", "--via", "proxy",
+         "--data-dir", str(tmp_path / "d")],
+        cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+    out = _check_line(line[0], world=1)
+    assert out["config"]["via"] == "proxy"
+
+
+def test_bench_via_proxy_miss(tmp_path):
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--model", "tiny", "--steps", "1",
+         "--warmup", "1", "--via", "proxy-miss",
+         "--data-dir", str(tmp_path / "d")],
+        cwd=REPO, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    line = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+    out = _check_line(line[0], world=1)
+    assert out["config"]["via"] == "proxy-miss"
+
+
 def test_bench_torchrun_world2(tmp_path):
     env = dict(os.environ)
     env["MASTER_ADDR"] = "127.0.0.1"
