@@ -208,44 +208,83 @@ class Lander:
         off = 0
         i = 0
         fill_s = 0.0
-        while off < nbytes:
-            slab = i % n_slabs
-            if self._slab_busy[slab]:
-                self._slab_events[slab].sync()
-            want = min(self.slab_bytes, nbytes - off)
-            view = self.pool.slab_view(slab)[:want]
-            tf = time.perf_counter()
-            got = 0
-            while got < want:
-                try:
-                    n = fill(view[got:])
-                except Exception as e:
-                    raise LandingError(off, e) from e
-                if n <= 0:
-                    raise LandingError(off)
-                got += n
-            fill_s += time.perf_counter() - tf
-            if keep_head and len(head) < self.head_bytes:
-                take = min(want, self.head_bytes - len(head))
-                head += bytes(view[:take])
-            if chain is not None:
-                chain.update(view)
-            h.h2d_async(buf.ptr + base_off + off, self.pool.slab_ptr(slab),
-                        want, self.copy_stream.handle)
-            self._slab_events[slab].record(self.copy_stream.handle)
-            self._slab_busy[slab] = True
-            if gpu_state is not None:
-                # chain over the landed region (whole 64B blocks only; the
-                # ragged tail is folded in at finalize)
-                self._slab_events[slab].wait(self.verify_stream.handle)
-                abs_off = base_off + off
-                nblk = (want if abs_off + want < file_size
-                        else file_size - (file_size % 64) - abs_off) // 64
-                h.sha256_chain_update(gpu_state.ptr, buf.ptr + abs_off,
-                                      max(nblk, 0),
-                                      self.verify_stream.handle)
-            off += want
-            i += 1
+        # Exact-digest mode: the sequential host SHA-256 chain runs on a
+        # DEDICATED thread, gated per slab, so network receive of slab
+        # k+1 overlaps hashing of slab k (hashlib releases the GIL on
+        # >2 KiB updates).  Round 1 hashed inline on the fill thread,
+        # serializing recv with the chain.
+        hasher = hq = hash_done = None
+        if chain is not None:
+            import queue as _q
+            import threading as _t
+
+            hq = _q.Queue()
+            hash_done = [_t.Event() for _ in range(n_slabs)]
+            for e in hash_done:
+                e.set()
+
+            def _hash_loop():
+                while True:
+                    item = hq.get()
+                    if item is None:
+                        return
+                    s_i, v = item
+                    chain.update(v)
+                    hash_done[s_i].set()
+
+            hasher = _t.Thread(target=_hash_loop, daemon=True,
+                               name="sha256-chain")
+            hasher.start()
+        try:
+            while off < nbytes:
+                slab = i % n_slabs
+                if self._slab_busy[slab]:
+                    self._slab_events[slab].sync()
+                if hash_done is not None:
+                    hash_done[slab].wait()
+                want = min(self.slab_bytes, nbytes - off)
+                view = self.pool.slab_view(slab)[:want]
+                tf = time.perf_counter()
+                got = 0
+                while got < want:
+                    try:
+                        n = fill(view[got:])
+                    except Exception as e:
+                        raise LandingError(off, e) from e
+                    if n <= 0:
+                        raise LandingError(off)
+                    got += n
+                fill_s += time.perf_counter() - tf
+                if keep_head and len(head) < self.head_bytes:
+                    take = min(want, self.head_bytes - len(head))
+                    head += bytes(view[:take])
+                if chain is not None:
+                    hash_done[slab].clear()
+                    hq.put((slab, view))
+                h.h2d_async(buf.ptr + base_off + off,
+                            self.pool.slab_ptr(slab),
+                            want, self.copy_stream.handle)
+                self._slab_events[slab].record(self.copy_stream.handle)
+                self._slab_busy[slab] = True
+                if gpu_state is not None:
+                    # chain over the landed region (whole 64B blocks only;
+                    # the ragged tail is folded in at finalize)
+                    self._slab_events[slab].wait(self.verify_stream.handle)
+                    abs_off = base_off + off
+                    nblk = (want if abs_off + want < file_size
+                            else file_size - (file_size % 64) - abs_off
+                            ) // 64
+                    h.sha256_chain_update(gpu_state.ptr, buf.ptr + abs_off,
+                                          max(nblk, 0),
+                                          self.verify_stream.handle)
+                off += want
+                i += 1
+        finally:
+            if hasher is not None:
+                # drain: completed slabs hash before we return/raise, so
+                # the chain state always equals the landed byte count
+                hq.put(None)
+                hasher.join()
         return head, fill_s
 
     def sync(self) -> None:
