@@ -27,6 +27,7 @@ GEOMS = {
     "llama3-70b": ("LLAMA3_70B", 16),
     "tiny": (None, 2),
     "gguf-8b": ("LLAMA3_8B", 0),     # Ollama q4_K pull + GPU dequant
+    "gguf-70b": ("LLAMA3_70B", 0),   # nameplate llama3:70b (use --virtual)
     "gguf-tiny": (None, 0),
     "dataset": (None, 0),            # zstd frame shards into HBM ring
     "parquet": (None, 0),            # real c4-like parquet, ZSTD pages
@@ -168,13 +169,22 @@ def main():
         from demodel_amd.testing import synth
 
         geom_name, n_shards = GEOMS[args.model]
-        assert geom_name, f"--virtual unsupported for {args.model}"
-        geom = getattr(synth, geom_name)
-        sizes = synth.shard_sizes(geom, args.shards or n_shards)
+        geom = getattr(synth, geom_name) if geom_name else TINY_GEOM
+        n_shards = n_shards or 2
+        prefixes = None
+        if args.model.startswith("gguf"):
+            # REAL GGUF header prefix + patterned quant payload: the
+            # 41 GB llama3:70b q4_K config without disk backing
+            prefix, total = synth.gguf_virtual(geom)
+            sizes = {"model.gguf": total}
+            prefixes = {"model.gguf": prefix}
+        else:
+            sizes = synth.shard_sizes(geom, args.shards or n_shards)
         files = {n: None for n in sizes}
         os.makedirs(data_dir, exist_ok=True)
         origin = FakeOrigin(data_dir, redirect_blobs=True)
-        origin.add_hf_repo_virtual("bench/model", sizes)
+        origin.add_hf_repo_virtual("bench/model", sizes,
+                                   prefixes=prefixes)
         total_bytes = sum(sizes.values())
     else:
         # rank 0 generates the shared synthetic files; others wait
@@ -303,7 +313,8 @@ def main():
         from demodel_amd.engine.formats import gguf
         from demodel_amd.gpu import have_gpu as _hg, hip
 
-        pd = gguf.ProgressiveDequant() if _hg() else None
+        pd = (gguf.ProgressiveDequant(
+            buffer_pool=landers.buffer_pool) if _hg() else None)
         res = pull_mod.pull_hf(
             "bench/model", endpoint=endpoint, workers=args.workers,
             verify=args.verify, landers=landers,
@@ -322,7 +333,9 @@ def main():
             n_t = len(tensors)
             log(f"dequant overlap: {pd.launched_early}/{n_t} tensors "
                 f"launched before pull completion")
-            del tensors, pd
+            del tensors
+            pd.recycle_arena()  # views dropped; reuse at 70B scale
+            del pd
         else:
             gg = gguf.parse(blob)
             out_bytes = sum(t.n_elems * 2 for t in gg.tensors

@@ -148,13 +148,10 @@ def parse(blob) -> GGUFModel:
 # ------------------------------------------------------------------ #
 # synthesis (tests + synthetic benchmarks)
 
-def build_file(path: str, tensors: list[tuple[str, tuple[int, ...], int]],
-               kv: dict | None = None, rng=None) -> GGUFModel:
-    """Write a synthetic GGUF file: (name, dims, type_id) with random
-    quant payloads."""
-    import numpy as np
-
-    rng = rng or np.random.default_rng(0)
+def _build_header(tensors: list[tuple[str, tuple[int, ...], int]],
+                  kv: dict | None = None):
+    """Serialize a GGUF v3 header for (name, dims, type_id) tensors.
+    Returns (head_bytes, infos, data_offset, align, payload_bytes)."""
     kv = dict(kv or {})
     kv.setdefault("general.architecture", "llama")
     kv.setdefault("general.alignment", 32)
@@ -191,6 +188,31 @@ def build_file(path: str, tensors: list[tuple[str, tuple[int, ...], int]],
             head += struct.pack("<Q", d)
         head += struct.pack("<IQ", t.type_id, t.offset)
     data_offset = (len(head) + align - 1) // align * align
+    return head, infos, data_offset, align, off
+
+
+def build_virtual(tensors: list[tuple[str, tuple[int, ...], int]],
+                  kv: dict | None = None) -> tuple[bytes, int]:
+    """A GGUF blob for virtual origins: (prefix_bytes, total_size).
+
+    The prefix is the REAL serialized header (padded to data_offset);
+    payloads are left to the origin's tiled pattern — any bytes are
+    numerically valid quant blocks, so dequant-at-scale benchmarks
+    (llama3:70b, BASELINE config 4 at nameplate size) run without a
+    41 GB file on disk."""
+    head, infos, data_offset, align, payload = _build_header(tensors, kv)
+    prefix = head + b"\0" * (data_offset - len(head))
+    return prefix, data_offset + payload
+
+
+def build_file(path: str, tensors: list[tuple[str, tuple[int, ...], int]],
+               kv: dict | None = None, rng=None) -> GGUFModel:
+    """Write a synthetic GGUF file: (name, dims, type_id) with random
+    quant payloads."""
+    import numpy as np
+
+    rng = rng or np.random.default_rng(0)
+    head, infos, data_offset, align, _ = _build_header(tensors, kv)
     # payloads tile a fixed random block so multi-GB synthesis is IO-bound,
     # not RNG-bound
     tile = rng.integers(0, 256, size=16 << 20, dtype=np.uint8).tobytes()
@@ -343,9 +365,11 @@ class ProgressiveDequant:
 
     QUANT_IDS = (2, 8, 12, 14)
 
-    def __init__(self, device_index: int = 0):
+    def __init__(self, device_index: int = 0, buffer_pool=None):
         import threading
 
+        self._pool = buffer_pool
+        self._arena_bytes = 0
         self._lock = threading.Lock()
         self._ranges: list = []     # merged completed [lo, hi)
         self._prefix = 0            # contiguous bytes landed from 0
@@ -384,7 +408,11 @@ class ProgressiveDequant:
         self._quants = [t for t in self._gg.tensors
                         if t.type_id in self.QUANT_IDS]
         out_bytes = sum(t.n_elems * 2 for t in self._quants)
-        self._arena = self._h.DeviceBuffer(max(out_bytes, 1))
+        self._arena_bytes = max(out_bytes, 1)
+        self._arena = (self._pool.take(self._arena_bytes)
+                       if self._pool is not None else None)
+        if self._arena is None:
+            self._arena = self._h.DeviceBuffer(self._arena_bytes)
         off = 0
         for t in self._quants:
             self._offsets[t.name] = off
@@ -448,6 +476,15 @@ class ProgressiveDequant:
     def launched_early(self) -> int:
         """How many tensors launched before finish() (observability)."""
         return getattr(self, "_early", self._next if self._gg else 0)
+
+    def recycle_arena(self) -> None:
+        """Hand the bf16 output arena back to the buffer pool (near
+        device capacity a fresh hipMalloc of just-freed pages costs
+        SECONDS of driver page reclaim — BufferPool doc).  Caller must
+        have dropped every tensor view returned by finish()."""
+        if self._pool is not None and self._arena is not None:
+            self._pool.put(self._arena, self._arena_bytes)
+        self._arena = None
 
 
 def _check_extent(gg: GGUFModel, t: GGUFTensor) -> None:

@@ -67,13 +67,20 @@ class FakeOrigin:
         return commit
 
     def add_hf_repo_virtual(self, repo_id: str, sizes: dict[str, int],
-                            commit: str | None = None) -> str:
+                            commit: str | None = None,
+                            prefixes: dict[str, bytes] | None = None
+                            ) -> str:
         """Register a repo whose blobs are served from memory (tiled
         deterministic pattern) — no disk, no page cache; for benchmarks
-        bigger than the box's disk (e.g. the 141 GB Llama-3-70B set)."""
+        bigger than the box's disk (e.g. the 141 GB Llama-3-70B set).
+
+        prefixes: optional per-file leading bytes (e.g. a REAL GGUF
+        header) served before the pattern, so format parsers work on
+        virtual blobs."""
         commit = commit or hashlib.sha1(repo_id.encode()).hexdigest()
         self.hf_repos[repo_id] = {"sha": commit, "files": {},
-                                  "virtual": dict(sizes)}
+                                  "virtual": dict(sizes),
+                                  "vprefix": dict(prefixes or {})}
         return commit
 
     _pattern: bytes | None = None
@@ -277,8 +284,9 @@ class FakeOrigin:
                              ("Content-Type", "text/plain")],
                     b"redirect")
             if fpath is None:
-                return await self._serve_virtual(writer, head, size,
-                                                 etag, extra=extra)
+                return await self._serve_virtual(
+                    writer, head, size, etag, extra=extra,
+                    prefix=repo.get("vprefix", {}).get(fname))
             return await self._serve_file(writer, head, fpath, etag=etag,
                                           extra=extra)
 
@@ -291,7 +299,9 @@ class FakeOrigin:
                 return await self._error(writer, 404)
             size, etag, fpath = self._file_info(repo, fname)
             if fpath is None:
-                return await self._serve_virtual(writer, head, size, etag)
+                return await self._serve_virtual(
+                    writer, head, size, etag,
+                    prefix=repo.get("vprefix", {}).get(fname))
             return await self._serve_file(writer, head, fpath, etag=etag)
 
         return await self._error(writer, 404)
@@ -311,7 +321,8 @@ class FakeOrigin:
             f"virtual:{name}:{size}".encode()).hexdigest(), None
 
     async def _serve_virtual(self, writer, req: RequestHead, size: int,
-                             etag: str, extra=None):
+                             etag: str, extra=None,
+                             prefix: bytes | None = None):
         from ..utils.netio import send_pattern_threaded
 
         start, end = 0, size - 1
@@ -338,8 +349,15 @@ class FakeOrigin:
         await writer.drain()
         if req.method == "HEAD":
             return
-        await send_pattern_threaded(writer, self._virtual_pattern(),
-                                    start, length)
+        sent = 0
+        if prefix and start < len(prefix):
+            piece = prefix[start:min(end + 1, len(prefix))]
+            writer.write(piece)
+            await writer.drain()
+            sent = len(piece)
+        if length - sent > 0:
+            await send_pattern_threaded(writer, self._virtual_pattern(),
+                                        start + sent, length - sent)
 
     _etag_cache: dict[tuple[str, float], str] = {}
 

@@ -202,6 +202,26 @@ def write_parquet_shards(out_dir: str, n_shards: int = 8,
     return files
 
 
+def gguf_tensor_specs(geom: dict, qtype: int = 12):
+    """(name, ggml_dims, type_id) specs for a llama-shaped GGUF."""
+    tensors = []
+    for name, shape in llama_tensor_table(geom):
+        if len(shape) == 1:
+            tensors.append((name, shape, 0))       # f32 norms
+        else:
+            d0, d1 = shape[1], shape[0]
+            tensors.append((name, (d0, d1), qtype))
+    return tensors
+
+
+def gguf_virtual(geom: dict, qtype: int = 12) -> tuple[bytes, int]:
+    """(header_prefix, total_size) for a virtual-origin GGUF — the 70B
+    nameplate config (41 GB q4_K) without disk backing."""
+    from ..engine.formats import gguf
+
+    return gguf.build_virtual(gguf_tensor_specs(geom, qtype))
+
+
 def write_gguf_model(path: str, geom: dict, qtype: int = 12,
                      reuse: bool = True):
     """Synthetic GGUF (default q4_K) with llama-shaped 2-D tensors."""
@@ -209,13 +229,6 @@ def write_gguf_model(path: str, geom: dict, qtype: int = 12,
 
     if reuse and os.path.exists(path) and os.path.getsize(path) > 0:
         return gguf.parse_bytes(open(path, "rb").read(8 << 20))
-    tensors = []
-    for name, shape in llama_tensor_table(geom):
-        if len(shape) == 1:
-            tensors.append((name, shape, 0))       # f32 norms
-        else:
-            # ggml dims are reversed (fastest first) and dim0 must divide
-            # the quant block size
-            d0, d1 = shape[1], shape[0]
-            tensors.append((name, (d0, d1), qtype))
-    return gguf.build_file(path, tensors)
+    # ggml dims are reversed (fastest first) and dim0 must divide the
+    # quant block size — gguf_tensor_specs handles both
+    return gguf.build_file(path, gguf_tensor_specs(geom, qtype))
