@@ -101,6 +101,33 @@ def test_explicit_prefetch_endpoint(stack, tmp_path):
     assert st["entries"] >= 1
 
 
+@pytest.mark.gpu
+def test_auto_prefetch_lands_in_hbm(stack, tmp_path):
+    """On a GPU box the pull-ahead lands in HBM (cuda:0) and the engine
+    pull reuses the device-resident blob."""
+    stack.proxy.prefetch_landers = LanderPool(0, gpu=True)
+    data = _mk_repo(stack, tmp_path, "org/gpupf", "model.safetensors",
+                    nbytes=8 << 20)
+    urllib.request.urlopen(
+        f"{stack.endpoint}/api/models/org/gpupf/revision/main",
+        timeout=20).read()
+    urllib.request.urlopen(
+        f"{stack.endpoint}/org/gpupf/resolve/main/model.safetensors",
+        timeout=30).read()
+    path = "/org/gpupf/resolve/main/model.safetensors"
+    assert _wait_registered(stack.proxy, path, timeout=60)
+    blob = stack.proxy.registry.get(path)
+    assert blob.device.startswith("cuda")
+    n_origin = len(stack.origin.requests)
+    res = pull_mod.pull_hf("org/gpupf", endpoint=stack.endpoint,
+                           registry=stack.proxy.registry)
+    assert len(stack.origin.requests) == n_origin
+    f = res.files[0]
+    assert f.blob is blob
+    got = bytes(f.blob.torch_u8().cpu().numpy().tobytes())
+    assert got == data
+
+
 def test_registry_lru_eviction():
     from demodel_amd.engine.pipeline import LandedBlob
     from demodel_amd.engine.registry import BlobRegistry
