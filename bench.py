@@ -296,12 +296,16 @@ def main():
         if not args.virtual:
             if not scatter_targets:
                 _build_targets(res)
+            t_sc = time.perf_counter()
             for f in res.files:
                 tg = scatter_targets.get(f.name)
                 if tg:
                     hdr = st.parse_header(f.blob.head)
                     n_t += len(load_into(f.blob, hdr, tg))
+            t_sc2 = time.perf_counter()
             n_t += len(res.tensors())
+            log(f"scatter {t_sc2 - t_sc:.3f}s views "
+                f"{time.perf_counter() - t_sc2:.3f}s")
         if have_gpu:
             torch.cuda.synchronize()
         return res, n_t

@@ -434,11 +434,18 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
                 s.close()
     lander0.sync()
     if head is None:
-        head = lander0.read_head(buf, total)
+        # seg-0 fallback dropped its in-flight head capture: rebuild the
+        # SEGMENT-0 portion now (other segments may still be landing)
+        head = lander0.read_head(buf, bounds[1])
     if on_range is not None:
         on_range(name, 0, bounds[1], buf, bytes(head))
     for f in futs:
         f.result()
+    # head spans min(total, head_bytes), which can extend past segment 0
+    # (small files under a small SEGMENT_MIN): complete it now that every
+    # segment's lander has synced
+    if len(head) < min(total, lander0.head_bytes):
+        head = lander0.read_head(buf, total)
     vc = verify_chunk or lander0.verify_chunk
     blob = LandedBlob(nbytes=total, device=f"cuda:{lander0.device_index}",
                       buffer=buf, verify_chunk=vc, head=bytes(head))

@@ -178,6 +178,11 @@ class AsyncCacheWriter:
             await asyncio.get_running_loop().run_in_executor(
                 _pool(), self._q.put, chunk)
 
+    def thread_write(self, chunk: bytes) -> None:
+        """Enqueue from a worker thread (threaded relay tee); blocking
+        put is fine off-loop."""
+        self._q.put(chunk)
+
     async def _join(self):
         from ..utils.netio import _pool
 
@@ -958,7 +963,8 @@ class ProxyServer:
                     if k not in present:
                         resp.headers.append((k, v))
             reusable = await self._stream_response(head, resp, up_r,
-                                                   writer, uri)
+                                                   writer, uri,
+                                                   up_w=up_w)
             return None
         finally:
             if reusable:
@@ -967,7 +973,8 @@ class ProxyServer:
                 UpstreamPool._close(up_w)
 
     async def _stream_response(self, req: RequestHead, resp: ResponseHead,
-                               up_r, writer, uri: str) -> bool:
+                               up_r, writer, uri: str,
+                               up_w=None) -> bool:
         """Stream the upstream body to the client (+ cache tee).  Returns
         True when the upstream connection is reusable (deterministic
         body framing, fully drained, no Connection: close)."""
@@ -1009,7 +1016,24 @@ class ProxyServer:
             else:
                 out.replace("Connection", "keep-alive")
                 writer.write(http1.serialize_response(out))
-                if req.method != "HEAD":
+                relayed = False
+                if (req.method != "HEAD" and mode == "length"
+                        and length >= (1 << 20) and up_w is not None):
+                    # blob bodies: socket->socket pump on a worker
+                    # thread (splice when no cache tee) — the MISS-path
+                    # equivalent of the HIT path's sendfile
+                    from ..utils.netio import relay_body_threaded
+
+                    try:
+                        await relay_body_threaded(
+                            up_r, up_w, writer, length,
+                            tee=(cache_writer.thread_write
+                                 if cache_writer else None))
+                        total = length
+                        relayed = True
+                    except NotImplementedError:
+                        relayed = False
+                if req.method != "HEAD" and not relayed:
                     async for chunk in _timed_body(
                             http1.iter_body(up_r, mode, length),
                             READ_TIMEOUT):

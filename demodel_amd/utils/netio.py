@@ -110,6 +110,121 @@ async def send_pattern_threaded(writer: asyncio.StreamWriter,
             pass
 
 
+def _sendall_nb(fd: int, view) -> None:
+    """sendall on a non-blocking socket fd (select-backed)."""
+    off = 0
+    while off < len(view):
+        try:
+            n = os.write(fd, view[off:])
+        except BlockingIOError:
+            sel.select([], [fd], [], 10)
+            continue
+        if n == 0:
+            raise ConnectionResetError("peer went away")
+        off += n
+
+
+async def relay_body_threaded(up_r: asyncio.StreamReader,
+                              up_w: asyncio.StreamWriter,
+                              writer: asyncio.StreamWriter,
+                              length: int, tee=None) -> None:
+    """Relay a length-delimited body upstream->client on a worker
+    thread — the proxy MISS hot loop (reference: goproxy's io.Copy,
+    start.go:201-204).  The asyncio relay pays event-loop wakeups and
+    userspace copies per 256 KiB chunk and tops out ~3 GB/s aggregate;
+    this path moves the socket->socket pump off-loop (os.splice
+    kernel-side when there is no cache tee, else an 8 MiB recv/send
+    loop) and scales with cores like the sendfile HIT path.
+
+    Raises NotImplementedError (before moving any bytes) when a
+    precondition fails — TLS on either side, no raw socket — so the
+    caller falls back to the asyncio loop.
+    tee(bytes) is called on the worker thread for every chunk, in
+    order (cache fill).
+    """
+    up_sock = up_w.transport.get_extra_info("socket")
+    cl_sock = writer.transport.get_extra_info("socket")
+    if (up_sock is None or cl_sock is None
+            or up_w.transport.get_extra_info("sslcontext") is not None
+            or writer.transport.get_extra_info("sslcontext") is not None):
+        raise NotImplementedError
+    await writer.drain()
+    up_w.transport.pause_reading()
+    up_fd = up_sock.fileno()
+    cl_fd = cl_sock.fileno()
+    # bytes asyncio already consumed past the response head
+    buf = up_r._buffer  # CPython StreamReader internal (3.10)
+    take = min(len(buf), length)
+    pre = bytes(buf[:take])
+    del buf[:take]
+    can_splice = hasattr(os, "splice") and tee is None
+
+    def run():
+        if pre:
+            if tee:
+                tee(pre)
+            _sendall_nb(cl_fd, memoryview(pre))
+        left = length - len(pre)
+        if left <= 0:
+            return
+        if can_splice:
+            rfd, wfd = os.pipe()
+            # a splice bigger than the pipe would BLOCK with the pipe
+            # full (we drain only after it returns): grow the pipe and
+            # never ask for more than its capacity
+            pipe_sz = 1 << 16
+            try:
+                import fcntl
+
+                pipe_sz = fcntl.fcntl(wfd, 1031, 1 << 20)  # F_SETPIPE_SZ
+            except OSError:
+                pass
+            try:
+                while left > 0:
+                    try:
+                        n = os.splice(up_fd, wfd, min(left, pipe_sz))
+                    except BlockingIOError:
+                        sel.select([up_fd], [], [], 10)
+                        continue
+                    if n == 0:
+                        raise ConnectionResetError("upstream EOF")
+                    m = 0
+                    while m < n:
+                        try:
+                            m += os.splice(rfd, cl_fd, n - m)
+                        except BlockingIOError:
+                            sel.select([], [cl_fd], [], 10)
+                    left -= n
+            finally:
+                os.close(rfd)
+                os.close(wfd)
+            return
+        big = bytearray(8 << 20)
+        mv = memoryview(big)
+        while left > 0:
+            want = min(left, len(big))
+            try:
+                n = os.readv(up_fd, [mv[:want]])
+            except BlockingIOError:
+                sel.select([up_fd], [], [], 10)
+                continue
+            if n == 0:
+                raise ConnectionResetError("upstream EOF")
+            if tee:
+                tee(bytes(mv[:n]))
+            _sendall_nb(cl_fd, mv[:n])
+            left -= n
+
+    loop = asyncio.get_running_loop()
+    try:
+        await loop.run_in_executor(_pool(), run)
+    finally:
+        try:
+            up_w.transport.resume_reading()
+        except Exception:
+            pass
+
+
 async def sendfile_threaded(writer: asyncio.StreamWriter, f,
                             start: int, length: int) -> None:
     """Send [start, start+length) of file f on writer's socket.
