@@ -601,9 +601,50 @@ class ProxyServer:
             blob = lander.land(f.readinto, hit.body_size, verify=True,
                                expected_digests=expected,
                                verify_chunk=vc)
+        # cached bodies keep their original Content-Encoding
+        # (CONTRIBUTING.md:116); a gzip body is stored compressed, so
+        # the HBM-resident copy must be the DECODED bytes — inflate on
+        # the DEFLATE kernel (K2) for GPU landings, zlib on host ones
+        enc = next((v for k, v in hit.headers
+                    if k.lower() == "content-encoding"), "").lower()
+        if "gzip" in enc and blob.nbytes > 18:
+            blob = self._gunzip_blob(blob)
         self.registry.put(path, blob)
         log.info("prefetched %s -> %s (%d bytes, verified=%s)",
                  path, blob.device, blob.nbytes, expected is not None)
+
+    @staticmethod
+    def _gunzip_blob(blob):
+        from ..engine.pipeline import LandedBlob
+
+        if blob.device == "cpu":
+            import gzip as _gzip
+
+            raw = _gzip.decompress(bytes(blob.buffer))
+            out = LandedBlob(nbytes=len(raw), device="cpu",
+                             buffer=bytearray(raw),
+                             head=raw[:8 << 20])
+            return out
+        from ..engine.formats.compress import gunzip_blob_gpu
+        from ..engine.pipeline import VERIFY_CHUNK
+
+        dst, res = gunzip_blob_gpu(blob)
+        out = LandedBlob(nbytes=res.written, device=blob.device,
+                         buffer=dst, verify_chunk=VERIFY_CHUNK)
+        # head for downstream format parsing (D2H of the prefix)
+        from ..gpu import hip
+
+        import ctypes as _ct
+
+        h = hip()
+        n = min(res.written, 8 << 20)
+        hb = bytearray(n)
+        addr = _ct.addressof((_ct.c_char * n).from_buffer(hb))
+        s = h.Stream(0)
+        h.d2h_async(addr, dst.ptr, n, s.handle)
+        s.sync()
+        out.head = bytes(hb)
+        return out
 
     async def _serve_prefetch(self, head: RequestHead, reader,
                               writer) -> bool:

@@ -128,6 +128,37 @@ def test_auto_prefetch_lands_in_hbm(stack, tmp_path):
     assert got == data
 
 
+def test_prefetch_decodes_gzip_bodies(stack, tmp_path):
+    """Cached bodies keep their original Content-Encoding (the
+    reference's worked example is a gzip Ollama manifest,
+    CONTRIBUTING.md:116); the registry copy must be the DECODED
+    bytes."""
+    import json as _json
+
+    blob = tmp_path / "layer.bin"
+    blob.write_bytes(os.urandom(4096))
+    manifest = stack.origin.add_ollama_model(
+        "library/tiny", "latest",
+        [("application/vnd.ollama.image.model", str(blob))])
+    # client pulls the manifest through the proxy (cached gzip'd)
+    got = urllib.request.urlopen(
+        f"{stack.endpoint}/v2/library/tiny/manifests/latest",
+        timeout=20).read()
+    # urllib doesn't auto-decode; body is gzip per origin config
+    import gzip as _gzip
+
+    assert _json.loads(_gzip.decompress(got)) == manifest
+    path = "/v2/library/tiny/manifests/latest"
+    req = urllib.request.Request(
+        f"{stack.endpoint}/__demodel/prefetch", method="POST",
+        data=json.dumps({"paths": [path]}).encode())
+    with urllib.request.urlopen(req, timeout=20) as r:
+        assert json.loads(r.read())["queued"] == [path]
+    assert _wait_registered(stack.proxy, path)
+    reg = stack.proxy.registry.get(path)
+    assert _json.loads(bytes(reg.buffer)) == manifest  # decoded!
+
+
 def test_registry_lru_eviction():
     from demodel_amd.engine.pipeline import LandedBlob
     from demodel_amd.engine.registry import BlobRegistry
