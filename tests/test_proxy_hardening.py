@@ -106,6 +106,37 @@ def test_small_post_passthrough(stack):
     assert obj["sha256"] == hashlib.sha256(body).hexdigest()
 
 
+def test_upstream_pool_unit(stack, tmp_path):
+    """Pool mechanics: reuse, idle-TTL expiry, max-idle cap."""
+    import asyncio
+
+    from demodel_amd.proxy.server import UpstreamPool
+
+    host, port = "127.0.0.1", stack.origin_port
+
+    async def scenario():
+        pool = UpstreamPool(max_idle_per_key=1, idle_ttl=0.05)
+        r1, w1, reused = await pool.acquire(host, port, None)
+        assert reused is False
+        pool.release(host, port, False, r1, w1)
+        r2, w2, reused = await pool.acquire(host, port, None)
+        assert reused is True and w2 is w1
+        # max_idle=1: releasing two keeps only one
+        r3, w3, _ = await pool.acquire(host, port, None)
+        pool.release(host, port, False, r2, w2)
+        pool.release(host, port, False, r3, w3)
+        assert len(pool._idle[(host, port, False)]) == 1
+        assert w3.is_closing()  # the overflow one was closed
+        # TTL expiry: after the idle window the conn is discarded
+        await asyncio.sleep(0.1)
+        r4, w4, reused = await pool.acquire(host, port, None)
+        assert reused is False
+        pool.close_all()
+        w4.close()
+
+    stack.lt.call(scenario())
+
+
 def test_cache_fill_still_correct_off_loop(stack, tmp_path):
     """Cache fills now run on a worker thread; the entry must still be
     byte-exact and replayable with the origin down."""
