@@ -17,6 +17,9 @@
 
 #include <hip/hip_runtime.h>
 
+#include <stdlib.h>
+#include <string.h>
+
 #include "zstd_common.h"
 
 using namespace zstd_core;
@@ -32,10 +35,21 @@ namespace {
 // matches, raw/RLE blocks) maintains it.  Matches farther back than the
 // window take a rare global-read path behind an explicit vmcnt drain.
 
+// LDS same-wave ordering fence.  GCN/CDNA processes a wave's LDS
+// instructions in issue order, so a ds_read after a may-alias ds_write
+// needs NO architectural fence — but the asm memory clobber also stops
+// the compiler from caching LDS-resident values (FSE table entries)
+// across the call, which costs register pressure + reloads in the
+// per-sequence loop.  FENCE=false relies on in-order LDS + compiler
+// alias analysis; selectable at runtime for A/B (DEMODEL_ZSTD_MODE).
+template <bool FENCE>
+__device__ __forceinline__ void lds_fence() {
+  if (FENCE) asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
 // one sequence's copies: literal run from the (global) literal buffer,
-// then the match via the LDS window (or the rare far-global path).
-// `lane0`/`stride` select full-wave (0,64) or half-wave (f*32,32) use.
-template <int ZWIN>
+// then the match via the LDS window (or the rare far-global path)
+template <int ZWIN, bool FENCE>
 __device__ __forceinline__ void exec_seq(
     uint8_t* __restrict__ out, uint8_t* win, uint64_t wfrom,
     const uint8_t* lit_src, uint64_t p0, uint32_t ll, uint32_t len,
@@ -48,7 +62,7 @@ __device__ __forceinline__ void exec_seq(
   }
   uint64_t mp = p0 + ll;
   if (dist <= ZWIN - 128 && mp - dist >= wfrom) {
-    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    lds_fence<FENCE>();
     if (dist >= len) {
       for (uint32_t k = lane; k < len; k += 64) {
         uint8_t v = win[(mp + k - dist) & ZWMASK];
@@ -65,12 +79,14 @@ __device__ __forceinline__ void exec_seq(
           out[mp + copied + k] = v;
           win[(mp + copied + k) & ZWMASK] = v;
         }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        lds_fence<FENCE>();
         copied += n;
       }
     }
   } else {
     // far match: read old output from HBM; drain our stores first
+    // (global stores have NO same-address ordering guarantee, unlike
+    // LDS — this fence is load-bearing and stays in both modes)
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     for (uint32_t k = lane; k < len; k += 64) {
       uint8_t v = out[mp + k - dist];
@@ -440,7 +456,7 @@ __device__ __forceinline__ void huf_stream_decode(ZShared& sh, int k) {
 // caps residency at ~5 (16K) / 2 (64K) workgroups per CU, so granting the
 // full register budget is free — without it the compiler spills the
 // sequence loop into AGPRs/scratch (1626 v_accvgpr_read in the ISA).
-template <int ZWIN>
+template <int ZWIN, bool FENCE>
 __global__ void __launch_bounds__(64, ZWIN <= 16 * 1024 ? 2 : 1)
 zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
   constexpr int ZWMASK = ZWIN - 1;
@@ -506,7 +522,7 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           seq_decode(s, sh);
           if (s.err) break;
           if (s.have) {
-            exec_seq<ZWIN>(out, win, s.wfrom, s.lit_base + s.lit_used,
+            exec_seq<ZWIN, FENCE>(out, win, s.wfrom, s.lit_base + s.lit_used,
                            s.pos_r, s.c_ll, s.c_len, s.c_dist, lane);
             s.pos_r += s.c_ll + s.c_len;
             s.lit_used += s.c_ll;
@@ -514,7 +530,7 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
           seq_commit(s, sh, d->dst_cap);
         }
         if (!s.err && s.have) {
-          exec_seq<ZWIN>(out, win, s.wfrom, s.lit_base + s.lit_used,
+          exec_seq<ZWIN, FENCE>(out, win, s.wfrom, s.lit_base + s.lit_used,
                          s.pos_r, s.c_ll, s.c_len, s.c_dist, lane);
           s.pos_r += s.c_ll + s.c_len;
           s.lit_used += s.c_ll;
@@ -570,7 +586,7 @@ zstd_kernel(ZstdDesc* __restrict__ descs, int n_streams) {
 // payloads; executing TWO frames' chains from one instruction stream
 // lets the scheduler overlap their latencies.  Each frame keeps its own
 // ZShared + LDS window; copies for both frames issue back-to-back.
-template <int ZWIN>
+template <int ZWIN, bool FENCE>
 __global__ void __launch_bounds__(64, 1)
 zstd_kernel_x2(ZstdDesc* __restrict__ descs, int n_streams) {
   constexpr int ZWMASK = ZWIN - 1;
@@ -673,7 +689,7 @@ zstd_kernel_x2(ZstdDesc* __restrict__ descs, int n_streams) {
           if (contB) seq_decode(B, sh[1]);
           if (contA && !A.err) {
             if (A.have) {
-              exec_seq<ZWIN>(out[0], win[0], A.wfrom,
+              exec_seq<ZWIN, FENCE>(out[0], win[0], A.wfrom,
                              A.lit_base + A.lit_used, A.pos_r,
                              A.c_ll, A.c_len, A.c_dist, lane);
               A.pos_r += A.c_ll + A.c_len;
@@ -683,7 +699,7 @@ zstd_kernel_x2(ZstdDesc* __restrict__ descs, int n_streams) {
           }
           if (contB && !B.err) {
             if (B.have) {
-              exec_seq<ZWIN>(out[1], win[1], B.wfrom,
+              exec_seq<ZWIN, FENCE>(out[1], win[1], B.wfrom,
                              B.lit_base + B.lit_used, B.pos_r,
                              B.c_ll, B.c_len, B.c_dist, lane);
               B.pos_r += B.c_ll + B.c_len;
@@ -696,14 +712,14 @@ zstd_kernel_x2(ZstdDesc* __restrict__ descs, int n_streams) {
         }
         // drain the last pending sequence of each frame
         if (litsA && !A.err && A.have) {
-          exec_seq<ZWIN>(out[0], win[0], A.wfrom,
+          exec_seq<ZWIN, FENCE>(out[0], win[0], A.wfrom,
                          A.lit_base + A.lit_used, A.pos_r,
                          A.c_ll, A.c_len, A.c_dist, lane);
           A.pos_r += A.c_ll + A.c_len;
           A.lit_used += A.c_ll;
         }
         if (litsB && !B.err && B.have) {
-          exec_seq<ZWIN>(out[1], win[1], B.wfrom,
+          exec_seq<ZWIN, FENCE>(out[1], win[1], B.wfrom,
                          B.lit_base + B.lit_used, B.pos_r,
                          B.c_ll, B.c_len, B.c_dist, lane);
           B.pos_r += B.c_ll + B.c_len;
@@ -811,19 +827,37 @@ extern "C" void launch_zstd_frames(const uint64_t* desc, int n_frames,
   // LDS is shared chip-wide, so 64 KiB windows would cap residency at
   // ~2 workgroups/CU across all their launches combined.
   bool small = window == 16 * 1024 || (window == 0 && n_frames >= 768);
-  if (small && n_frames >= 2) {
-    // paired kernel: 2 frames per wave, sequence chains interleaved
+  // DEMODEL_ZSTD_MODE: "x1" (default) / "x2" (paired interleave),
+  // optional "nf" suffix drops the LDS exec fences ("x1nf", "x2nf").
+  // Measured on MI355X (profiles/zstd_x2_ab.md): x2 LOSES on word
+  // salad (5.9 vs 8.9 GB/s) — the doubled live state parks in AGPRs
+  // and occupancy halves — so x1 stays the default.
+  const char* mode = getenv("DEMODEL_ZSTD_MODE");
+  bool use_x2 = mode && strstr(mode, "x2");
+  bool fence = !(mode && strstr(mode, "nf"));
+  int blocks = n_frames < 4096 ? n_frames : 4096;
+  if (use_x2 && small && n_frames >= 2) {
     int pairs = (n_frames + 1) / 2;
-    int blocks = pairs < 4096 ? pairs : 4096;
-    hipLaunchKernelGGL(zstd_kernel_x2<16 * 1024>, dim3(blocks), dim3(64),
-                       0, stream, (ZstdDesc*)desc, n_frames);
+    int pblocks = pairs < 4096 ? pairs : 4096;
+    if (fence)
+      hipLaunchKernelGGL((zstd_kernel_x2<16 * 1024, true>), dim3(pblocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
+    else
+      hipLaunchKernelGGL((zstd_kernel_x2<16 * 1024, false>), dim3(pblocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
   } else if (small) {
-    int blocks = n_frames < 4096 ? n_frames : 4096;
-    hipLaunchKernelGGL(zstd_kernel<16 * 1024>, dim3(blocks), dim3(64), 0,
-                       stream, (ZstdDesc*)desc, n_frames);
+    if (fence)
+      hipLaunchKernelGGL((zstd_kernel<16 * 1024, true>), dim3(blocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
+    else
+      hipLaunchKernelGGL((zstd_kernel<16 * 1024, false>), dim3(blocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
   } else {
-    int blocks = n_frames < 4096 ? n_frames : 4096;
-    hipLaunchKernelGGL(zstd_kernel<64 * 1024>, dim3(blocks), dim3(64), 0,
-                       stream, (ZstdDesc*)desc, n_frames);
+    if (fence)
+      hipLaunchKernelGGL((zstd_kernel<64 * 1024, true>), dim3(blocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
+    else
+      hipLaunchKernelGGL((zstd_kernel<64 * 1024, false>), dim3(blocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
   }
 }

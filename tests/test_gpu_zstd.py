@@ -105,3 +105,29 @@ def test_zstd_rejects_garbage(hipmod):
     dst = h.DeviceBuffer(1024)
     res = zstd_gpu([(src.ptr, len(junk), dst.ptr, 1024)])[0]
     assert not res.ok
+
+
+@pytest.mark.parametrize("mode", ["x2", "x1nf", "x2nf"])
+def test_zstd_kernel_variants(hipmod, mode, monkeypatch):
+    """The paired-interleave (x2) and fence-free (nf) kernel variants
+    (DEMODEL_ZSTD_MODE) must be byte-exact vs pyarrow — they share the
+    decode machinery but different scheduling/synchronization."""
+    from demodel_amd.engine.formats.compress import zstd_gpu
+
+    monkeypatch.setenv("DEMODEL_ZSTD_MODE", mode)
+    h = hipmod
+    s = h.Stream(0)
+    codec = pa.Codec("zstd", compression_level=3)
+    payloads = list(_payloads().values()) * 3  # odd count: unpaired tail
+    frames, dsts = [], []
+    for data in payloads:
+        f = bytes(codec.compress(data))
+        src = _upload(h, f, s)
+        dst = h.DeviceBuffer(max(len(data), 1))
+        dsts.append((src, dst, data))
+        frames.append((src.ptr, len(f), dst.ptr, max(len(data), 1)))
+    results = zstd_gpu(frames, window=16 * 1024)
+    for i, (res, (_, dst, data)) in enumerate(zip(results, dsts)):
+        assert res.ok, (mode, i, res.error)
+        assert res.written == len(data), (mode, i)
+        assert _download(h, dst, len(data), s) == data, (mode, i)
