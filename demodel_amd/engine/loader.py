@@ -16,6 +16,8 @@ import struct
 from ..gpu import hip
 from .formats.safetensors import SafetensorsHeader
 
+_PIECE = 1 << 20  # scatter descriptor granularity (see load_into)
+
 
 def load_into(blob, header: SafetensorsHeader, targets: dict,
               strict: bool = True, stream=None) -> list[str]:
@@ -74,8 +76,16 @@ def load_into(blob, header: SafetensorsHeader, targets: dict,
             raise ValueError(f"{name}: target must be contiguous")
         src_off = header.data_offset + info.begin
         if dst.dtype == getattr(torch, info.torch_dtype):
-            desc += struct.pack("<4Q", src_off, dst.data_ptr(),
-                                info.nbytes, 0)
+            # one descriptor per <=1 MiB piece: the kernel maps one
+            # workgroup per descriptor, so an unsplit 1 GB tensor would
+            # crawl on a single WG (~4 GB/s; measured 256 ms for the
+            # 16 GB model).  1 MiB pieces give 16k WGs across 256 CUs.
+            off = 0
+            while off < info.nbytes:
+                n = min(_PIECE, info.nbytes - off)
+                desc += struct.pack("<4Q", src_off + off,
+                                    dst.data_ptr() + off, n, 0)
+                off += n
         elif info.torch_dtype == "float32" and dst.dtype == torch.bfloat16:
             casts.append((src_off, dst.data_ptr(), info.nbytes // 4))
         else:
