@@ -98,6 +98,27 @@ def _cmd_stats(args) -> int:
     return 0
 
 
+def _cmd_prefetch(args) -> int:
+    """Ask a running proxy to land blob paths into its HBM registry
+    (POST /__demodel/prefetch); with no paths, print prefetch status."""
+    import urllib.request
+
+    base = (args.endpoint or "http://127.0.0.1:8080").rstrip("/")
+    url = base + "/__demodel/prefetch"
+    if not args.paths:
+        with urllib.request.urlopen(url, timeout=30) as r:
+            print(json.dumps(json.loads(r.read()), indent=1))
+        return 0
+    req = urllib.request.Request(
+        url, method="POST",
+        data=json.dumps({"paths": args.paths}).encode(),
+        headers={"Content-Type": "application/json"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        out = json.loads(r.read())
+    print(json.dumps(out, indent=1))
+    return 0 if out.get("queued") is not None else 1
+
+
 def _cmd_gc(args) -> int:
     from .cache import CacheStore
 
@@ -159,6 +180,17 @@ def main(argv: list[str] | None = None) -> int:
                                    "budget")
     gp.add_argument("--max-gb", type=float, required=True)
     gp.set_defaults(fn=_cmd_gc)
+
+    pf = sub.add_parser("prefetch",
+                        help="land blob paths into a running proxy's "
+                             "HBM registry (POST /__demodel/prefetch); "
+                             "no paths = show status")
+    pf.add_argument("paths", nargs="*",
+                    help="request paths, e.g. "
+                         "/org/repo/resolve/main/model.safetensors")
+    pf.add_argument("--endpoint", default=None,
+                    help="proxy base URL (default http://127.0.0.1:8080)")
+    pf.set_defaults(fn=_cmd_prefetch)
 
     args = p.parse_args(argv)
     if not hasattr(args, "fn"):

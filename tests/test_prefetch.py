@@ -159,6 +159,24 @@ def test_prefetch_decodes_gzip_bodies(stack, tmp_path):
     assert _json.loads(bytes(reg.buffer)) == manifest  # decoded!
 
 
+def test_prefetch_cli(stack, tmp_path, capsys):
+    from demodel_amd.cli import main as cli_main
+
+    data = _mk_repo(stack, tmp_path, "org/cli", "w.safetensors",
+                    nbytes=100_000)
+    path = "/org/cli/resolve/main/w.safetensors"
+    rc = cli_main(["prefetch", path, "--endpoint", stack.endpoint])
+    assert rc == 0
+    out = json.loads(capsys.readouterr().out)
+    assert out["queued"] == [path]
+    assert _wait_registered(stack.proxy, path)
+    assert bytes(stack.proxy.registry.get(path).buffer) == data
+    rc = cli_main(["prefetch", "--endpoint", stack.endpoint])
+    assert rc == 0
+    st = json.loads(capsys.readouterr().out)
+    assert path in st["registered"]
+
+
 def test_registry_lru_eviction():
     from demodel_amd.engine.pipeline import LandedBlob
     from demodel_amd.engine.registry import BlobRegistry
