@@ -835,8 +835,18 @@ extern "C" void launch_zstd_frames(const uint64_t* desc, int n_frames,
   const char* mode = getenv("DEMODEL_ZSTD_MODE");
   bool use_x2 = mode && strstr(mode, "x2");
   bool fence = !(mode && strstr(mode, "nf"));
+  // "w8": 8 KiB window — LDS/WG drops 28.9->20.7 KB so CU residency
+  // rises 5->7 workgroups; pays more far-match fallbacks.  A/B lever.
+  bool w8 = (mode && strstr(mode, "w8")) || window == 8 * 1024;
   int blocks = n_frames < 4096 ? n_frames : 4096;
-  if (use_x2 && small && n_frames >= 2) {
+  if (w8 && small && !use_x2) {
+    if (fence)
+      hipLaunchKernelGGL((zstd_kernel<8 * 1024, true>), dim3(blocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
+    else
+      hipLaunchKernelGGL((zstd_kernel<8 * 1024, false>), dim3(blocks),
+                         dim3(64), 0, stream, (ZstdDesc*)desc, n_frames);
+  } else if (use_x2 && small && n_frames >= 2) {
     int pairs = (n_frames + 1) / 2;
     int pblocks = pairs < 4096 ? pairs : 4096;
     if (fence)
