@@ -343,7 +343,7 @@ def main():
         else:
             gg = gguf.parse(blob)
             out_bytes = sum(t.n_elems * 2 for t in gg.tensors
-                            if t.type_id in (2, 8, 12, 14))
+                            if t.type_id in gguf.ProgressiveDequant.QUANT_IDS)
             n_t = len(gg.tensors)
         if have_gpu:
             torch.cuda.synchronize()

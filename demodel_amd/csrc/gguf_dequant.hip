@@ -161,31 +161,205 @@ __global__ void dequant_q6_K(const uint8_t* __restrict__ src,
   }
 }
 
+// ---- q5_0 / q5_1: one lane per 32-elem block ----------------------------
+// q5_0 (22 B): f16 d, u32 qh (5th bits), 16 B nibbles; v = d*(q-16)
+// q5_1 (24 B): f16 d, f16 m, u32 qh, 16 B nibbles;     v = d*q + m
+
+__global__ void dequant_q5_0(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_blocks) {
+  int64_t b0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t b = b0; b < n_blocks; b += stride) {
+    const uint8_t* q = src + b * 22;
+    float d = f16_to_f32(q);
+    uint32_t qh = (uint32_t)q[2] | ((uint32_t)q[3] << 8) |
+                  ((uint32_t)q[4] << 16) | ((uint32_t)q[5] << 24);
+    uint16_t* o = dst + b * 32;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      uint8_t byte = q[6 + j];
+      int x0 = (int)((byte & 0xF) | (((qh >> j) << 4) & 0x10)) - 16;
+      int x1 = (int)((byte >> 4) | ((qh >> (j + 12)) & 0x10)) - 16;
+      o[j] = f32_to_bf16(d * (float)x0);
+      o[j + 16] = f32_to_bf16(d * (float)x1);
+    }
+  }
+}
+
+__global__ void dequant_q5_1(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_blocks) {
+  int64_t b0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t b = b0; b < n_blocks; b += stride) {
+    const uint8_t* q = src + b * 24;
+    float d = f16_to_f32(q);
+    float m = f16_to_f32(q + 2);
+    uint32_t qh = (uint32_t)q[4] | ((uint32_t)q[5] << 8) |
+                  ((uint32_t)q[6] << 16) | ((uint32_t)q[7] << 24);
+    uint16_t* o = dst + b * 32;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      uint8_t byte = q[8 + j];
+      int x0 = (int)((byte & 0xF) | (((qh >> j) << 4) & 0x10));
+      int x1 = (int)((byte >> 4) | ((qh >> (j + 12)) & 0x10));
+      o[j] = f32_to_bf16(d * (float)x0 + m);
+      o[j + 16] = f32_to_bf16(d * (float)x1 + m);
+    }
+  }
+}
+
+// ---- q5_K: one wave per 256-elem superblock -----------------------------
+// 176 B: f16 d, f16 dmin, 12 B 6-bit scales (q4_K packing), 32 B qh
+// (5th bits, bit `sub` of qh[l]), 128 B nibbles.
+
+__global__ void dequant_q5_K(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_sblocks) {
+  int64_t sb0 = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  int64_t stride = (int64_t)gridDim.x * (blockDim.x / 64);
+  int lane = threadIdx.x & 63;
+  for (int64_t sb = sb0; sb < n_sblocks; sb += stride) {
+    const uint8_t* blk = src + sb * 176;
+    float d = f16_to_f32(blk);
+    float dmin = f16_to_f32(blk + 2);
+    const uint8_t* scales = blk + 4;
+    const uint8_t* qh = blk + 16;
+    const uint8_t* qs = blk + 48;
+    uint16_t* o = dst + sb * 256;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int j = r * 64 + lane;
+      int sub = j >> 5;
+      uint8_t sc, mn;
+      scale_min_k4(sub, scales, &sc, &mn);
+      int l = j & 31;
+      int pair = j >> 6;
+      uint8_t byte = qs[pair * 32 + l];
+      int nib = (sub & 1) ? (byte >> 4) : (byte & 0xF);
+      int q = nib + (((qh[l] >> sub) & 1) << 4);
+      o[j] = f32_to_bf16(d * (float)sc * (float)q -
+                         dmin * (float)mn);
+    }
+  }
+}
+
+// ---- q3_K: one wave per superblock --------------------------------------
+// 110 B: 32 B hmask, 64 B 2-bit qs, 12 B packed 6-bit scales, f16 d.
+
+__global__ void dequant_q3_K(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_sblocks) {
+  int64_t sb0 = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  int64_t stride = (int64_t)gridDim.x * (blockDim.x / 64);
+  int lane = threadIdx.x & 63;
+  for (int64_t sb = sb0; sb < n_sblocks; sb += stride) {
+    const uint8_t* blk = src + sb * 110;
+    const uint8_t* hmask = blk;
+    const uint8_t* qs = blk + 32;
+    const uint8_t* sp = blk + 96;
+    float d = f16_to_f32(blk + 108);
+    // 12 packed bytes -> 16 6-bit scales (ggml kmask unpack)
+    uint32_t w0 = (uint32_t)sp[0] | ((uint32_t)sp[1] << 8) |
+                  ((uint32_t)sp[2] << 16) | ((uint32_t)sp[3] << 24);
+    uint32_t w1 = (uint32_t)sp[4] | ((uint32_t)sp[5] << 8) |
+                  ((uint32_t)sp[6] << 16) | ((uint32_t)sp[7] << 24);
+    uint32_t w2 = (uint32_t)sp[8] | ((uint32_t)sp[9] << 8) |
+                  ((uint32_t)sp[10] << 16) | ((uint32_t)sp[11] << 24);
+    uint32_t aux[4];
+    aux[0] = (w0 & 0x0f0f0f0fu) | (((w2 >> 0) & 0x03030303u) << 4);
+    aux[1] = (w1 & 0x0f0f0f0fu) | (((w2 >> 2) & 0x03030303u) << 4);
+    aux[2] = ((w0 >> 4) & 0x0f0f0f0fu) | (((w2 >> 4) & 0x03030303u) << 4);
+    aux[3] = ((w1 >> 4) & 0x0f0f0f0fu) | (((w2 >> 6) & 0x03030303u) << 4);
+    uint16_t* o = dst + sb * 256;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int j = r * 64 + lane;
+      int nh = j >> 7;             // 128-half
+      int rr = j & 127;
+      int jj = rr >> 5;            // shift group 0..3
+      int l = rr & 31;
+      int is = nh * 8 + jj * 2 + (l >> 4);
+      int sc = (int)((aux[is >> 2] >> ((is & 3) * 8)) & 0xFF) - 32;
+      int shift = jj * 2;
+      int bit = nh * 4 + jj;
+      int qv = (int)((qs[nh * 32 + l] >> shift) & 3) -
+               (((hmask[l] >> bit) & 1) ? 0 : 4);
+      o[j] = f32_to_bf16(d * (float)sc * (float)qv);
+    }
+  }
+}
+
+// ---- q2_K: one wave per superblock --------------------------------------
+// 84 B: 16 B scales (lo nib = scale, hi nib = min), 64 B 2-bit qs,
+// f16 d, f16 dmin.
+
+__global__ void dequant_q2_K(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_sblocks) {
+  int64_t sb0 = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x / 64);
+  int64_t stride = (int64_t)gridDim.x * (blockDim.x / 64);
+  int lane = threadIdx.x & 63;
+  for (int64_t sb = sb0; sb < n_sblocks; sb += stride) {
+    const uint8_t* blk = src + sb * 84;
+    const uint8_t* scales = blk;
+    const uint8_t* qs = blk + 16;
+    float d = f16_to_f32(blk + 80);
+    float dmin = f16_to_f32(blk + 82);
+    uint16_t* o = dst + sb * 256;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int j = r * 64 + lane;
+      int nh = j >> 7;
+      int rr = j & 127;
+      int jj = rr >> 5;
+      int l = rr & 31;
+      uint8_t sc = scales[nh * 8 + jj * 2 + (l >> 4)];
+      int qv = (qs[nh * 32 + l] >> (jj * 2)) & 3;
+      o[j] = f32_to_bf16(d * (float)(sc & 0xF) * (float)qv -
+                         dmin * (float)(sc >> 4));
+    }
+  }
+}
+
 }  // namespace
 
-// qtype ids follow GGML: 2=q4_0, 8=q8_0, 12=q4_K, 14=q6_K
+// qtype ids follow GGML: 2=q4_0, 6=q5_0, 7=q5_1, 8=q8_0, 10=q2_K,
+// 11=q3_K, 12=q4_K, 13=q5_K, 14=q6_K
 extern "C" void launch_gguf_dequant(int qtype, const void* src,
                                     uint16_t* dst, int64_t n_blocks,
                                     hipStream_t stream) {
   if (n_blocks <= 0) return;
   const uint8_t* s = (const uint8_t*)src;
-  if (qtype == 2 || qtype == 8) {
+  if (qtype == 2 || qtype == 6 || qtype == 7 || qtype == 8) {
     int64_t want = (n_blocks + 255) / 256;
     int blocks = want > 8192 ? 8192 : (int)want;
     if (qtype == 2)
       hipLaunchKernelGGL(dequant_q4_0, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else if (qtype == 6)
+      hipLaunchKernelGGL(dequant_q5_0, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else if (qtype == 7)
+      hipLaunchKernelGGL(dequant_q5_1, dim3(blocks), dim3(256), 0, stream,
                          s, dst, n_blocks);
     else
       hipLaunchKernelGGL(dequant_q8_0, dim3(blocks), dim3(256), 0, stream,
                          s, dst, n_blocks);
     return;
   }
-  if (qtype == 12 || qtype == 14) {
+  if (qtype == 10 || qtype == 11 || qtype == 12 || qtype == 13 ||
+      qtype == 14) {
     // 4 waves per 256-thread workgroup, one superblock per wave
     int64_t want = (n_blocks + 3) / 4;
     int blocks = want > 8192 ? 8192 : (int)want;
-    if (qtype == 12)
+    if (qtype == 10)
+      hipLaunchKernelGGL(dequant_q2_K, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else if (qtype == 11)
+      hipLaunchKernelGGL(dequant_q3_K, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else if (qtype == 12)
       hipLaunchKernelGGL(dequant_q4_K, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else if (qtype == 13)
+      hipLaunchKernelGGL(dequant_q5_K, dim3(blocks), dim3(256), 0, stream,
                          s, dst, n_blocks);
     else
       hipLaunchKernelGGL(dequant_q6_K, dim3(blocks), dim3(256), 0, stream,

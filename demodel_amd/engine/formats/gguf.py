@@ -24,8 +24,13 @@ GGML_TYPES = {
     0: ("f32", 1, 4),
     1: ("f16", 1, 2),
     2: ("q4_0", 32, 18),
+    6: ("q5_0", 32, 22),
+    7: ("q5_1", 32, 24),
     8: ("q8_0", 32, 34),
+    10: ("q2_K", 256, 84),
+    11: ("q3_K", 256, 110),
     12: ("q4_K", 256, 144),
+    13: ("q5_K", 256, 176),
     14: ("q6_K", 256, 210),
     30: ("bf16", 1, 2),
 }
@@ -309,6 +314,116 @@ def _dequant_cpu_impl(type_id: int, raw: bytes, n_elems: int):
                 d[:, None] * sc[:, None] * nib.astype(np.float32)
                 - dmin[:, None] * mn[:, None])
         return out.reshape(-1)
+    if type_id == 6:  # q5_0
+        nb = n_elems // 32
+        blk = b[:nb * 22].reshape(nb, 22)
+        d = blk[:, 0:2].copy().view(np.float16).astype(np.float32)
+        qh = blk[:, 2:6].copy().view(np.uint32)[:, 0]
+        qs = blk[:, 6:22]
+        j = np.arange(16)
+        xh0 = ((qh[:, None] >> j) << 4) & 0x10
+        xh1 = (qh[:, None] >> (j + 12)) & 0x10
+        lo = ((qs & 0xF) | xh0).astype(np.int32) - 16
+        hi = ((qs >> 4) | xh1).astype(np.int32) - 16
+        out = np.concatenate([lo, hi], axis=1).astype(np.float32)
+        return (out * d).reshape(-1)
+    if type_id == 7:  # q5_1
+        nb = n_elems // 32
+        blk = b[:nb * 24].reshape(nb, 24)
+        d = blk[:, 0:2].copy().view(np.float16).astype(np.float32)
+        m = blk[:, 2:4].copy().view(np.float16).astype(np.float32)
+        qh = blk[:, 4:8].copy().view(np.uint32)[:, 0]
+        qs = blk[:, 8:24]
+        j = np.arange(16)
+        xh0 = ((qh[:, None] >> j) << 4) & 0x10
+        xh1 = (qh[:, None] >> (j + 12)) & 0x10
+        lo = ((qs & 0xF) | xh0).astype(np.float32)
+        hi = ((qs >> 4) | xh1).astype(np.float32)
+        out = np.concatenate([lo, hi], axis=1)
+        return (out * d + m).reshape(-1)
+    if type_id == 13:  # q5_K
+        nb = n_elems // 256
+        blk = b[:nb * 176].reshape(nb, 176)
+        d = blk[:, 0:2].copy().view(np.float16).astype(np.float32)[:, 0]
+        dmin = blk[:, 2:4].copy().view(np.float16).astype(np.float32)[:, 0]
+        scales = blk[:, 4:16]
+        qh = blk[:, 16:48]
+        qs = blk[:, 48:176]
+        out = np.empty((nb, 256), np.float32)
+        for j in range(8):  # 32-elem sub-blocks
+            if j < 4:
+                sc = (scales[:, j] & 63).astype(np.float32)
+                mn = (scales[:, j + 4] & 63).astype(np.float32)
+            else:
+                sc = ((scales[:, j + 4] & 0xF)
+                      | ((scales[:, j - 4] >> 6) << 4)).astype(np.float32)
+                mn = ((scales[:, j + 4] >> 4)
+                      | ((scales[:, j] >> 6) << 4)).astype(np.float32)
+            pair = j // 2
+            hi_nib = j % 2
+            q8 = qs[:, pair * 32:(pair + 1) * 32]
+            nib = (q8 >> 4) if hi_nib else (q8 & 0xF)
+            q5 = nib.astype(np.float32) + (
+                ((qh >> j) & 1) << 4).astype(np.float32)
+            out[:, j * 32:(j + 1) * 32] = (
+                d[:, None] * sc[:, None] * q5
+                - dmin[:, None] * mn[:, None])
+        return out.reshape(-1)
+    if type_id == 11:  # q3_K
+        nb = n_elems // 256
+        blk = b[:nb * 110].reshape(nb, 110)
+        hmask = blk[:, 0:32]
+        qs = blk[:, 32:96]
+        sp = blk[:, 96:108].copy().view(np.uint32)  # (nb, 3)
+        d = blk[:, 108:110].copy().view(np.float16).astype(np.float32)[:, 0]
+        km1, km2 = 0x03030303, 0x0f0f0f0f
+        w0, w1, w2 = sp[:, 0], sp[:, 1], sp[:, 2]
+        aux = np.stack([
+            (w0 & km2) | (((w2 >> 0) & km1) << 4),
+            (w1 & km2) | (((w2 >> 2) & km1) << 4),
+            ((w0 >> 4) & km2) | (((w2 >> 4) & km1) << 4),
+            ((w1 >> 4) & km2) | (((w2 >> 6) & km1) << 4),
+        ], axis=1)  # (nb, 4) u32 -> 16 bytes = 16 scales
+        sc16 = aux.view(np.uint8).reshape(nb, 16).astype(np.int32) - 32
+        out = np.empty((nb, 256), np.float32)
+        for nh in range(2):
+            for jj in range(4):
+                byte = qs[:, nh * 32:(nh + 1) * 32]
+                q2 = ((byte >> (jj * 2)) & 3).astype(np.int32)
+                bit = nh * 4 + jj
+                hm = ((hmask >> bit) & 1).astype(np.int32)
+                qv = q2 - np.where(hm == 1, 0, 4)
+                for half16 in range(2):
+                    is_ = nh * 8 + jj * 2 + half16
+                    sl = slice(half16 * 16, half16 * 16 + 16)
+                    col = nh * 128 + jj * 32 + half16 * 16
+                    out[:, col:col + 16] = (
+                        d[:, None] * sc16[:, is_][:, None]
+                        * qv[:, sl].astype(np.float32))
+        return out.reshape(-1)
+    if type_id == 10:  # q2_K
+        nb = n_elems // 256
+        blk = b[:nb * 84].reshape(nb, 84)
+        scales = blk[:, 0:16]
+        qs = blk[:, 16:80]
+        d = blk[:, 80:82].copy().view(np.float16).astype(np.float32)[:, 0]
+        dmin = blk[:, 82:84].copy().view(np.float16).astype(np.float32)[:, 0]
+        out = np.empty((nb, 256), np.float32)
+        for nh in range(2):
+            for jj in range(4):
+                byte = qs[:, nh * 32:(nh + 1) * 32]
+                q2 = ((byte >> (jj * 2)) & 3).astype(np.float32)
+                for half16 in range(2):
+                    is_ = nh * 8 + jj * 2 + half16
+                    sc = scales[:, is_]
+                    sl = slice(half16 * 16, half16 * 16 + 16)
+                    col = nh * 128 + jj * 32 + half16 * 16
+                    out[:, col:col + 16] = (
+                        d[:, None] * (sc & 0xF)[:, None].astype(np.float32)
+                        * q2[:, sl]
+                        - dmin[:, None]
+                        * (sc >> 4)[:, None].astype(np.float32))
+        return out.reshape(-1)
     if type_id == 14:  # q6_K
         nb = n_elems // 256
         blk = b[:nb * 210].reshape(nb, 210)
@@ -363,7 +478,7 @@ class ProgressiveDequant:
     never depends on the overlap.
     """
 
-    QUANT_IDS = (2, 8, 12, 14)
+    QUANT_IDS = (2, 6, 7, 8, 10, 11, 12, 13, 14)
 
     def __init__(self, device_index: int = 0, buffer_pool=None):
         import threading
@@ -509,7 +624,8 @@ def dequant_all_gpu(gg: GGUFModel, stream=None) -> dict:
     h = hip()
     own = stream is None
     stream = stream or h.Stream(0)
-    quants = [t for t in gg.tensors if t.type_id in (2, 8, 12, 14)]
+    quants = [t for t in gg.tensors
+              if t.type_id in ProgressiveDequant.QUANT_IDS]
     for t in quants:
         _check_extent(gg, t)
     out_bytes = sum(t.n_elems * 2 for t in quants)
@@ -527,7 +643,7 @@ def dequant_all_gpu(gg: GGUFModel, stream=None) -> dict:
     u8 = torch.from_dlpack(arena.to_dlpack())
     out = {}
     for t in gg.tensors:
-        if t.type_id in (2, 8, 12, 14):
+        if t.type_id in ProgressiveDequant.QUANT_IDS:
             o = offsets[t.name]
             out[t.name] = (u8[o:o + t.n_elems * 2]
                            .view(torch.bfloat16).view(t.dims[::-1]))
