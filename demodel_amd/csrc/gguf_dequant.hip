@@ -161,6 +161,27 @@ __global__ void dequant_q6_K(const uint8_t* __restrict__ src,
   }
 }
 
+// ---- q4_1: one lane per 32-elem block (20 B: f16 d, f16 m, 16 B nibbles;
+// v = d*q + m) -------------------------------------------------------------
+
+__global__ void dequant_q4_1(const uint8_t* __restrict__ src,
+                             uint16_t* __restrict__ dst, int64_t n_blocks) {
+  int64_t b0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t b = b0; b < n_blocks; b += stride) {
+    const uint8_t* q = src + b * 20;
+    float d = f16_to_f32(q);
+    float m = f16_to_f32(q + 2);
+    uint16_t* o = dst + b * 32;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      uint8_t byte = q[4 + j];
+      o[j] = f32_to_bf16(d * (float)(byte & 0xF) + m);
+      o[j + 16] = f32_to_bf16(d * (float)(byte >> 4) + m);
+    }
+  }
+}
+
 // ---- q5_0 / q5_1: one lane per 32-elem block ----------------------------
 // q5_0 (22 B): f16 d, u32 qh (5th bits), 16 B nibbles; v = d*(q-16)
 // q5_1 (24 B): f16 d, f16 m, u32 qh, 16 B nibbles;     v = d*q + m
@@ -320,18 +341,22 @@ __global__ void dequant_q2_K(const uint8_t* __restrict__ src,
 
 }  // namespace
 
-// qtype ids follow GGML: 2=q4_0, 6=q5_0, 7=q5_1, 8=q8_0, 10=q2_K,
+// qtype ids follow GGML: 2=q4_0, 3=q4_1, 6=q5_0, 7=q5_1, 8=q8_0, 10=q2_K,
 // 11=q3_K, 12=q4_K, 13=q5_K, 14=q6_K
 extern "C" void launch_gguf_dequant(int qtype, const void* src,
                                     uint16_t* dst, int64_t n_blocks,
                                     hipStream_t stream) {
   if (n_blocks <= 0) return;
   const uint8_t* s = (const uint8_t*)src;
-  if (qtype == 2 || qtype == 6 || qtype == 7 || qtype == 8) {
+  if (qtype == 2 || qtype == 3 || qtype == 6 || qtype == 7 ||
+      qtype == 8) {
     int64_t want = (n_blocks + 255) / 256;
     int blocks = want > 8192 ? 8192 : (int)want;
     if (qtype == 2)
       hipLaunchKernelGGL(dequant_q4_0, dim3(blocks), dim3(256), 0, stream,
+                         s, dst, n_blocks);
+    else if (qtype == 3)
+      hipLaunchKernelGGL(dequant_q4_1, dim3(blocks), dim3(256), 0, stream,
                          s, dst, n_blocks);
     else if (qtype == 6)
       hipLaunchKernelGGL(dequant_q5_0, dim3(blocks), dim3(256), 0, stream,

@@ -24,6 +24,7 @@ GGML_TYPES = {
     0: ("f32", 1, 4),
     1: ("f16", 1, 2),
     2: ("q4_0", 32, 18),
+    3: ("q4_1", 32, 20),
     6: ("q5_0", 32, 22),
     7: ("q5_1", 32, 24),
     8: ("q8_0", 32, 34),
@@ -314,6 +315,16 @@ def _dequant_cpu_impl(type_id: int, raw: bytes, n_elems: int):
                 d[:, None] * sc[:, None] * nib.astype(np.float32)
                 - dmin[:, None] * mn[:, None])
         return out.reshape(-1)
+    if type_id == 3:  # q4_1
+        nb = n_elems // 32
+        blk = b[:nb * 20].reshape(nb, 20)
+        d = blk[:, 0:2].copy().view(np.float16).astype(np.float32)
+        m = blk[:, 2:4].copy().view(np.float16).astype(np.float32)
+        qs = blk[:, 4:20]
+        lo = (qs & 0xF).astype(np.float32)
+        hi = (qs >> 4).astype(np.float32)
+        out = np.concatenate([lo, hi], axis=1)
+        return (out * d + m).reshape(-1)
     if type_id == 6:  # q5_0
         nb = n_elems // 32
         blk = b[:nb * 22].reshape(nb, 22)
@@ -478,7 +489,7 @@ class ProgressiveDequant:
     never depends on the overlap.
     """
 
-    QUANT_IDS = (2, 6, 7, 8, 10, 11, 12, 13, 14)
+    QUANT_IDS = (2, 3, 6, 7, 8, 10, 11, 12, 13, 14)
 
     def __init__(self, device_index: int = 0, buffer_pool=None):
         import threading

@@ -113,7 +113,7 @@ def test_cast_f32_to_bf16(hipmod):
     assert torch.equal(got, want)
 
 
-@pytest.mark.parametrize("qtype,n_sup", [(2, 1000), (6, 1000), (7, 1000),
+@pytest.mark.parametrize("qtype,n_sup", [(2, 1000), (3, 1000), (6, 1000), (7, 1000),
                                          (8, 1000), (10, 64), (11, 64),
                                          (12, 64), (13, 64), (14, 64)])
 def test_gguf_dequant_matches_cpu(hipmod, qtype, n_sup):

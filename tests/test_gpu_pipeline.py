@@ -173,7 +173,7 @@ def test_gpu_dequant_all(stack, tmp_path):
             t.type_id, raw[gg.data_offset + t.offset:
                            gg.data_offset + t.offset + t.nbytes],
             t.n_elems))
-        if t.type_id in (2, 6, 7, 8, 10, 11, 12, 13, 14):
+        if t.type_id in (2, 3, 6, 7, 8, 10, 11, 12, 13, 14):
             want = want.to(torch.bfloat16).float()
         else:
             want = want.to(torch.bfloat16).float()
