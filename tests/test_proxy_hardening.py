@@ -106,6 +106,35 @@ def test_small_post_passthrough(stack):
     assert obj["sha256"] == hashlib.sha256(body).hexdigest()
 
 
+def test_relay_upstream_drop_mid_body(stack, tmp_path):
+    """Origin dies mid-blob during the threaded relay: the client must
+    see a hard failure (truncated/closed), the cache entry must be
+    dropped, and the next request must succeed end-to-end."""
+    import http.client
+
+    data = os.urandom(4 << 20)
+    p = tmp_path / "drop.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/drop", {"drop.bin": str(p)})
+    stack.origin.drop_once["drop.bin"] = 1 << 20
+    url = f"{stack.endpoint}/org/drop/resolve/main/drop.bin"
+    got = None
+    try:
+        with _get(url) as r:
+            got = r.read()
+    except (http.client.IncompleteRead, ConnectionResetError,
+            urllib.error.HTTPError, OSError):
+        got = None
+    assert got != data  # truncated or errored, never silently complete
+    # the aborted fill must not have poisoned the cache
+    with _get(url) as r:
+        assert r.read() == data
+    stack.stop_origin()  # and the good body IS cached
+    with _get(url) as r:
+        assert r.read() == data
+        assert r.headers["X-Demodel-Cache"] == "HIT"
+
+
 def test_upstream_pool_unit(stack, tmp_path):
     """Pool mechanics: reuse, idle-TTL expiry, max-idle cap."""
     import asyncio
