@@ -102,13 +102,16 @@ def main():
                     choices=["zstd", "snappy", "gzip"],
                     help="page codec for --model parquet")
     ap.add_argument("--via", default="direct",
-                    choices=["direct", "proxy", "proxy-miss"],
+                    choices=["direct", "proxy", "proxy-miss", "peer"],
                     help="direct: engine pulls from the origin (the "
                          "headline path); proxy: pulls go THROUGH the "
                          "demodel proxy with a primed cache (measures "
                          "the cache-HIT data plane a real client sees); "
                          "proxy-miss: caching disabled, measures the "
-                         "origin->proxy->client relay")
+                         "origin->proxy->client relay; peer: verified "
+                         "peer distribution — pull from a primed peer "
+                         "node, every chunk GPU-verified against the "
+                         "peer's digest record (/__demodel/digests)")
     ap.add_argument("--virtual", action="store_true",
                     help="serve blobs from memory (no disk) — for models "
                          "bigger than the box's disk, e.g. llama3-70b; "
@@ -221,7 +224,7 @@ def main():
             proxy.cache.cacheable = lambda *a, **k: False
         pport = lt.call(proxy.start())
         endpoint = f"http://127.0.0.1:{pport}"
-        if args.via == "proxy":
+        if args.via in ("proxy", "peer"):
             # prime the cache (untimed) so timed pulls measure the HIT
             # path: cache file -> sendfile -> engine -> HBM
             import concurrent.futures as cf
@@ -243,6 +246,21 @@ def main():
             with cf.ThreadPoolExecutor(max_workers=8) as ex:
                 list(ex.map(prime, files))
             log(f"proxy cache primed in {time.time() - t:.1f}s")
+            if args.via == "peer":
+                # wait for the peer's async chunk-digest records: the
+                # verified-distribution pull compares every chunk
+                # against them
+                from demodel_amd.engine.pull import fetch_peer_digests
+
+                t = time.time()
+                for n in files:
+                    while fetch_peer_digests(
+                            endpoint,
+                            f"/bench/model/resolve/main/{n}") is None:
+                        assert time.time() - t < 120, \
+                            f"peer digests for {n} never appeared"
+                        time.sleep(0.1)
+                log(f"peer digest records ready in {time.time() - t:.1f}s")
 
     landers = LanderPool(local_rank if have_gpu else 0,
                          slab_bytes=args.slab_mib << 20,
@@ -283,7 +301,8 @@ def main():
         res = pull_mod.pull_hf(
             "bench/model", endpoint=endpoint, workers=args.workers,
             verify=args.verify, landers=landers,
-            digest_map=digest_map or None)
+            digest_map=digest_map or None,
+            peer_verify=(args.via == "peer"))
         assert res.total_bytes == total_bytes, res.total_bytes
         if record_digests:
             for f in res.files:

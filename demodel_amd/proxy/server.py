@@ -558,7 +558,8 @@ class ProxyServer:
         from ..utils.netio import _pool
 
         loop = asyncio.get_running_loop()
-        fut = loop.run_in_executor(_pool(), self._prefetch_land, path)
+        fut = loop.run_in_executor(_pool(), self._prefetch_land, path,
+                                   force)
 
         def _done(f):
             self._prefetching.discard(path)
@@ -569,12 +570,18 @@ class ProxyServer:
         fut.add_done_callback(_done)
         return True
 
-    def _prefetch_land(self, path: str) -> None:
+    def _prefetch_land(self, path: str,
+                       fetch_missing: bool = True) -> None:
         """Worker thread: land a cached body into HBM (verified against
         the cache's recorded chunk digests when present) and register
         it.  Not-yet-cached paths are first pulled through our own
-        front door (which tees them into the cache)."""
+        front door (which tees them into the cache) — explicit
+        prefetches only; auto pull-ahead lands only what is already
+        cached (a MITM'd host may not match the reverse route, so a
+        self-GET could hit the wrong upstream)."""
         hit, _ = self._cached_entry_for_path(path)
+        if hit is None and not fetch_missing:
+            return
         if hit is None:
             from ..engine import fetch
 
@@ -844,7 +851,9 @@ class ProxyServer:
             if result is None:
                 # pull-ahead: a blob this proxy just cached can land in
                 # HBM now, so a later engine pull is served GPU-warm
-                if reverse_mode and head.method == "GET":
+                # (reverse AND MITM'd traffic — both cache under URIs
+                # the path-keyed lookup can resolve)
+                if head.method == "GET":
                     self._maybe_prefetch(orig_path)
                 # a request body we never finished forwarding leaves the
                 # client connection desynced — close it
