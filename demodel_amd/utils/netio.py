@@ -21,8 +21,13 @@ _POOL: cf.ThreadPoolExecutor | None = None
 def _pool() -> cf.ThreadPoolExecutor:
     global _POOL
     if _POOL is None:
-        _POOL = cf.ThreadPoolExecutor(max_workers=32,
-                                      thread_name_prefix="sendfile")
+        # sized for the worst co-residency: origin sendfile bodies +
+        # proxy relay pumps + cache-writer threads can all be in flight
+        # at once in one process (the bench runs all three); threads
+        # here spend their lives in syscalls, so over-provisioning is
+        # cheap and starvation is not
+        _POOL = cf.ThreadPoolExecutor(max_workers=96,
+                                      thread_name_prefix="netio")
     return _POOL
 
 
