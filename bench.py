@@ -99,7 +99,7 @@ def main():
     ap.add_argument("--shards", type=int, default=None,
                     help="override safetensors shard count")
     ap.add_argument("--parquet-codec", default="zstd",
-                    choices=["zstd", "snappy", "gzip"],
+                    choices=["zstd", "snappy", "gzip", "lz4"],
                     help="page codec for --model parquet")
     ap.add_argument("--via", default="direct",
                     choices=["direct", "proxy", "proxy-miss", "peer"],

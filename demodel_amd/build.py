@@ -71,6 +71,7 @@ HIP_SRCS = [
     "inflate.hip",
     "zstd_kernel.hip",
     "snappy.hip",
+    "lz4.hip",
 ]
 
 

@@ -57,6 +57,8 @@ void launch_zstd_frames(const uint64_t* desc, int n_frames, int* status,
                         hipStream_t stream, int window);
 void launch_snappy_streams(const uint64_t* desc, int n_streams,
                            hipStream_t stream);
+void launch_lz4_streams(const uint64_t* desc, int n_streams,
+                        hipStream_t stream);
 }
 
 // ------------------------------------------------------------------------
@@ -401,4 +403,13 @@ PYBIND11_MODULE(_hip, m) {
         "snappy-decompress n_streams descriptors (8 u64 each: src, "
         "src_len, dst, dst_cap, written, status, consumed, pad), wave "
         "per stream (parquet's default page codec)");
+  m.def("lz4_streams",
+        [](uintptr_t desc, int n_streams, uintptr_t stream) {
+          launch_lz4_streams((const uint64_t*)desc, n_streams,
+                             (hipStream_t)stream);
+        },
+        py::call_guard<py::gil_scoped_release>(),
+        "LZ4 raw-block decompress n_streams descriptors (8 u64 each: "
+        "src, src_len, dst, dst_cap, written, status, consumed, pad), "
+        "wave per stream (parquet LZ4/LZ4_RAW page codecs)");
 }

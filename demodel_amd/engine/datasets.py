@@ -60,7 +60,7 @@ def _prep_zst_frames_gpu(blob, idx):
                        ring.ptr + off, fr["decompressed"]))
         spans.append((off, fr["decompressed"]))
         off += fr["decompressed"]
-    return frames, [], [], [], ring, spans
+    return frames, [], [], [], [], ring, spans
 
 
 def _decompress_cpu(blob, idx):
@@ -137,6 +137,7 @@ def stream_dataset(repo: str, endpoint: str | None = None,
         all_frames = []
         all_snappy = []
         all_deflate = []
+        all_lz4 = []
         all_copies = []
         entries = []
         for f in shards:
@@ -145,7 +146,7 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                     raise RuntimeError(
                         "parquet streaming needs a GPU (CPU fallback "
                         "covers .zst shards)")
-                (frames, snappy, deflate, copies, ring,
+                (frames, snappy, deflate, lz4, copies, ring,
                  spans) = _prep_parquet_gpu(f.blob)
             else:
                 idx = _sidecar_idx(f.sidecar)
@@ -153,14 +154,16 @@ def stream_dataset(repo: str, endpoint: str | None = None,
                     data, spans = _decompress_cpu(f.blob, idx)
                     out.append((f, None, data, spans))
                     continue
-                (frames, snappy, deflate, copies, ring,
+                (frames, snappy, deflate, lz4, copies, ring,
                  spans) = _prep_zst_frames_gpu(f.blob, idx)
             entries.append((f, len(all_frames), len(frames),
                             len(all_snappy), len(snappy),
-                            len(all_deflate), len(deflate), ring, spans))
+                            len(all_deflate), len(deflate),
+                            len(all_lz4), len(lz4), ring, spans))
             all_frames += frames
             all_snappy += snappy
             all_deflate += deflate
+            all_lz4 += lz4
             all_copies += copies
         if entries:
             h = hip()
@@ -171,9 +174,12 @@ def stream_dataset(repo: str, endpoint: str | None = None,
 
             job = ZstdJob(all_frames, pre_launch=pre, window=16 << 10,
                           snappy_frames=all_snappy,
-                          deflate_frames=all_deflate)
-            for f, lo, n, slo, sn, dlo, dn, ring, spans in entries:
-                out.append((f, job.view(lo, n, slo, sn, dlo, dn),
+                          deflate_frames=all_deflate,
+                          lz4_frames=all_lz4)
+            for (f, lo, n, slo, sn, dlo, dn, llo, ln, ring,
+                 spans) in entries:
+                out.append((f, job.view(lo, n, slo, sn, dlo, dn,
+                                        llo, ln),
                             ring, spans))
         return out
 

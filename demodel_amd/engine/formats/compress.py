@@ -123,6 +123,8 @@ class ZstdJob:
                  snappy_frames: list[tuple[int, int, int, int]] | None
                  = None,
                  deflate_frames: list[tuple[int, int, int, int]] | None
+                 = None,
+                 lz4_frames: list[tuple[int, int, int, int]] | None
                  = None):
         from ...gpu import hip
 
@@ -130,6 +132,7 @@ class ZstdJob:
         n = self._n = len(frames)
         sn = self._sn = len(snappy_frames or ())
         dn = self._dn = len(deflate_frames or ())
+        ln = self._ln = len(lz4_frames or ())
         self._s = h.Stream(0)
         if pre_launch is not None:
             pre_launch(self._s.handle)
@@ -145,6 +148,18 @@ class ZstdJob:
             h.h2d_async(self._ddbuf.ptr, addr, nd, self._s.handle)
             h.inflate_streams(self._ddbuf.ptr, dn, self._s.handle)
             h.d2h_async(addr, self._ddbuf.ptr, nd, self._s.handle)
+        if ln:
+            nd = ln * DESC_WORDS * 8
+            self._lpin = h.PinnedPool(nd, 1)
+            ldesc = self._lpin.slab_view(0)
+            for i, (src, slen, dst, cap) in enumerate(lz4_frames):
+                struct.pack_into("<8Q", ldesc, i * DESC_WORDS * 8,
+                                 src, slen, dst, cap, 0, 0, 0, 0)
+            self._ldbuf = h.DeviceBuffer(nd)
+            addr = self._lpin.slab_ptr(0)
+            h.h2d_async(self._ldbuf.ptr, addr, nd, self._s.handle)
+            h.lz4_streams(self._ldbuf.ptr, ln, self._s.handle)
+            h.d2h_async(addr, self._ldbuf.ptr, nd, self._s.handle)
         if sn:
             nd = sn * DESC_WORDS * 8
             self._spin = h.PinnedPool(nd, 1)
@@ -194,11 +209,15 @@ class ZstdJob:
             self.deflate_results = (_parse_results(
                 bytearray(self._dpin.slab_view(0)), self._dn)
                 if self._dn else [])
+            self.lz4_results = (_parse_results(
+                bytearray(self._lpin.slab_view(0)), self._ln)
+                if self._ln else [])
         return self._results
 
     def view(self, lo: int, n: int, slo: int = 0, sn: int = 0,
-             dlo: int = 0, dn: int = 0) -> "ZstdJobView":
-        return ZstdJobView(self, lo, n, slo, sn, dlo, dn)
+             dlo: int = 0, dn: int = 0, llo: int = 0,
+             ln: int = 0) -> "ZstdJobView":
+        return ZstdJobView(self, lo, n, slo, sn, dlo, dn, llo, ln)
 
 
 class ZstdJobView:
@@ -207,10 +226,12 @@ class ZstdJobView:
     this view's zstd results followed by its snappy results."""
 
     def __init__(self, job: ZstdJob, lo: int, n: int, slo: int = 0,
-                 sn: int = 0, dlo: int = 0, dn: int = 0):
+                 sn: int = 0, dlo: int = 0, dn: int = 0,
+                 llo: int = 0, ln: int = 0):
         self._job, self._lo, self._vn = job, lo, n
         self._slo, self._svn = slo, sn
         self._dlo, self._dvn = dlo, dn
+        self._llo, self._lvn = llo, ln
 
     def done(self) -> bool:
         return self._job.done()
@@ -223,6 +244,9 @@ class ZstdJobView:
         if self._dvn:
             res = res + self._job.deflate_results[
                 self._dlo:self._dlo + self._dvn]
+        if self._lvn:
+            res = res + self._job.lz4_results[
+                self._llo:self._llo + self._lvn]
         return res
 
 
@@ -287,6 +311,15 @@ def gunzip_blob_gpu(blob, out_size: int | None = None):
             continue
         raise IOError(f"GPU inflate failed: {res.error}")
     raise IOError("GPU inflate: capacity growth exhausted")
+
+
+def lz4_gpu(streams: list[tuple[int, int, int, int]]
+            ) -> list[InflateResult]:
+    """Decompress raw LZ4 blocks on the GPU (csrc/lz4.hip) — parquet's
+    LZ4/LZ4_RAW page codecs."""
+    job = ZstdJob([], lz4_frames=streams)
+    job.wait()
+    return job.lz4_results
 
 
 def snappy_gpu(streams: list[tuple[int, int, int, int]]

@@ -175,7 +175,9 @@ def parse_page_header(data, pos: int) -> tuple[PageInfo, int]:
 CODEC_UNCOMPRESSED = 0
 CODEC_SNAPPY = 1
 CODEC_GZIP = 2
+CODEC_LZ4 = 5
 CODEC_ZSTD = 6
+CODEC_LZ4_RAW = 7
 
 
 def column_chunk_pages(data, start: int, total_compressed: int
@@ -340,7 +342,9 @@ def file_pages(path: str):
     out = []
     codec_names = {"UNCOMPRESSED": CODEC_UNCOMPRESSED,
                    "SNAPPY": CODEC_SNAPPY, "GZIP": CODEC_GZIP,
-                   "ZSTD": CODEC_ZSTD}
+                   "ZSTD": CODEC_ZSTD,
+                   # arrow writes both as raw LZ4 blocks in pages
+                   "LZ4": CODEC_LZ4, "LZ4_RAW": CODEC_LZ4_RAW}
     for rg in range(meta.num_row_groups):
         for col in range(meta.num_columns):
             cc = meta.row_group(rg).column(col)
@@ -371,6 +375,7 @@ def prep_pages_gpu(blob, pages, ring=None):
     frames = []
     snappy = []
     deflate = []
+    lz4 = []
     copies = []
     spans = []
     off = 0
@@ -390,17 +395,19 @@ def prep_pages_gpu(blob, pages, ring=None):
             frames.append((src, clen, dst, ulen))
         elif codec == CODEC_SNAPPY:
             snappy.append((src, clen, dst, ulen))
+        elif codec in (CODEC_LZ4, CODEC_LZ4_RAW):
+            lz4.append((src, clen, dst, ulen))
         elif codec == CODEC_GZIP:
             gz = gzip_deflate_offset(p.head)
             # raw DEFLATE stream between the member header and the
             # 8-byte crc32+isize trailer
             deflate.append((src + gz, clen - gz - 8, dst, ulen))
         else:
-            raise ValueError(f"GPU path supports ZSTD/SNAPPY/GZIP/"
+            raise ValueError(f"GPU path supports ZSTD/SNAPPY/GZIP/LZ4/"
                              f"UNCOMPRESSED, got codec {codec}")
         spans.append((off, p.uncomp_size))
         off += p.uncomp_size
-    return frames, snappy, deflate, copies, ring, spans
+    return frames, snappy, deflate, lz4, copies, ring, spans
 
 
 def launch_pages_gpu(blob, pages, ring=None):
@@ -412,7 +419,7 @@ def launch_pages_gpu(blob, pages, ring=None):
     from .compress import ZstdJob
 
     h = hip()
-    frames, snappy, deflate, copies, ring, spans = prep_pages_gpu(
+    frames, snappy, deflate, lz4, copies, ring, spans = prep_pages_gpu(
         blob, pages, ring)
 
     def pre(handle):
@@ -422,7 +429,8 @@ def launch_pages_gpu(blob, pages, ring=None):
     # 16 KiB LDS window: page batches run as concurrent jobs in
     # stream_dataset, so occupancy beats far-match locality
     job = ZstdJob(frames, pre_launch=pre, window=16 << 10,
-                  snappy_frames=snappy, deflate_frames=deflate)
+                  snappy_frames=snappy, deflate_frames=deflate,
+                  lz4_frames=lz4)
     return job, ring, spans
 
 
@@ -430,7 +438,8 @@ def decompress_pages_gpu(blob, pages, ring=None):
     """Synchronous wrapper around launch_pages_gpu: returns
     (ring_buffer, [(out_offset, size)]) covering every page."""
     job, ring, spans = launch_pages_gpu(blob, pages, ring)
-    results = job.wait() + job.snappy_results + job.deflate_results
+    results = (job.wait() + job.snappy_results + job.deflate_results
+               + job.lz4_results)
     bad = [(i, r) for i, r in enumerate(results) if not r.ok]
     if bad:
         raise IOError(f"GPU page decompress failed: {bad[:3]}")

@@ -98,7 +98,9 @@ def test_uncompressed_parquet_pages(tmp_path):
 
 def _page_plain(raw, codec_name, info):
     """CPU reconstruction of one decompressed page (v1 or v2)."""
-    c = pa.Codec(codec_name)
+    # parquet "lz4" pages are raw LZ4 blocks (arrow behavior), while
+    # pa.Codec("lz4") is the FRAME codec — use the raw-block one
+    c = pa.Codec("lz4_raw" if codec_name == "lz4" else codec_name)
     lvl = info.lvl_bytes
     levels = raw[info.comp_offset:info.comp_offset + lvl]
     body = raw[info.comp_offset + lvl:info.comp_offset + info.comp_size]
@@ -120,7 +122,7 @@ def test_v2_pages_parse_and_decode_cpu(tmp_path):
 
 
 @pytest.mark.gpu
-@pytest.mark.parametrize("codec", ["zstd", "snappy", "gzip"])
+@pytest.mark.parametrize("codec", ["zstd", "snappy", "gzip", "lz4"])
 @pytest.mark.parametrize("page_version", ["1.0", "2.0"])
 def test_gpu_page_decompress(tmp_path, codec, page_version):
     import ctypes
