@@ -129,6 +129,29 @@ def test_torchrun_bench_collective_modes_world1(mode, tmp_path):
     assert out["n_gpus"] == 1
 
 
+def test_torchrun_bench_world2_nccl(tmp_path):
+    """Real 2-GPU RCCL run of the allgather mode — auto-skips on 1-GPU
+    boxes (collective tests gated on visible devices)."""
+    if torch.cuda.device_count() < 2:
+        pytest.skip("needs >=2 GPUs")
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(_free_port()),
+         os.path.join(REPO, "bench.py"),
+         "--model", "tiny", "--mode", "allgather", "--steps", "1",
+         "--warmup", "1",
+         "--data-dir", str(tmp_path / "tinydata")],
+        cwd=REPO, capture_output=True, text=True, timeout=600, env=env)
+    assert r.returncode == 0, (r.stdout[-2000:], r.stderr[-2000:])
+    out = json.loads([ln for ln in r.stdout.splitlines()
+                      if ln.startswith("{")][-1])
+    assert out["n_gpus"] == 2 and out["value"] > 0
+
+
 def test_bench_dp_world1_direct(tmp_path):
     """Plain `python bench.py` (the driver's N=1 BENCH call) with
     scatter wired into the timed step."""
