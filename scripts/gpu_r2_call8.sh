@@ -9,6 +9,8 @@ export TMPDIR=/tmp
 # PMC run (no trace domains alongside --pmc, pool rule): lz4 + q5_K
 # dequant probes under counters
 cat > /tmp/pmc_probe.py <<'PYEOF'
+import sys
+sys.path.insert(0, "/root/repo")
 import ctypes
 import numpy as np
 import pyarrow as pa
