@@ -75,8 +75,11 @@ def _cmd_pull(args) -> int:
               file=sys.stderr)
         return 1
     gpu = True if args.gpu else (False if args.cpu else None)
+    kw = {}
+    if getattr(args, "peer_verify", False):
+        kw["peer_verify"] = True
     result = pull_spec(args.spec, endpoint=args.endpoint, out_dir=args.out,
-                       gpu=gpu)
+                       gpu=gpu, **kw)
     print(json.dumps(result, indent=2, default=str))
     return 0
 
@@ -164,6 +167,10 @@ def main(argv: list[str] | None = None) -> int:
     pp.add_argument("--cpu", action="store_true",
                     help="force host-RAM landing even with a GPU present")
     pp.add_argument("--out", default=None, help="materialize files here")
+    pp.add_argument("--peer-verify", action="store_true",
+                    help="when --endpoint is another demodel node, "
+                         "GPU-verify every chunk against the peer's "
+                         "digest record (hf:// specs)")
     pp.set_defaults(fn=_cmd_pull)
 
     vp = sub.add_parser("verify", help="re-verify cached blobs")

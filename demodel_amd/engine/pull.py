@@ -715,6 +715,7 @@ def pull_spec(spec: str, cfg: Config | None = None, endpoint=None,
                       out_dir=out_dir, **kw)
         return res.summary()
     if spec.startswith("ollama://"):
+        kw.pop("peer_verify", None)  # hf-layout only
         body = spec[len("ollama://"):]
         name, _, tag = body.partition(":")
         res = pull_ollama(name, tag or "latest", endpoint=endpoint,
