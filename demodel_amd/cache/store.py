@@ -172,8 +172,14 @@ class CacheStore:
             "chunk_sha256": entry.chunk_sha256,
             "created": entry.created or time.time(),
         }
-        tmp_meta = entry.body_path + ".meta.part"
-        with open(tmp_meta, "w") as f:
+        # UNIQUE temp name: concurrent fills of the same URI (two cold
+        # misses racing before either registered in-flight) both commit;
+        # a shared ".meta.part" made the loser's os.replace crash with
+        # FileNotFoundError (surfaced by the 32-thread soak test)
+        fd, tmp_meta = tempfile.mkstemp(
+            prefix=f".{os.path.basename(entry.body_path)}.",
+            suffix=".meta.part", dir=self.root)
+        with os.fdopen(fd, "w") as f:
             json.dump(meta, f, indent=1)
         os.replace(tmp_meta, entry.body_path + ".meta")
 
