@@ -106,13 +106,24 @@ def test_small_post_passthrough(stack):
     assert obj["sha256"] == hashlib.sha256(body).hexdigest()
 
 
-def test_concurrent_mixed_load_soak(stack, tmp_path):
+@pytest.mark.parametrize("prefetch", [False, True])
+def test_concurrent_mixed_load_soak(stack, tmp_path, prefetch):
     """32 threads hammer the proxy with mixed hits/misses/ranges of
     several blobs at once — exercises the upstream pool, the in-flight
     fill registry (thundering herd on a cold URI), and the threaded
-    relay under contention.  Everything must come back byte-exact."""
+    relay under contention.  Everything must come back byte-exact.
+    With prefetch=True, auto pull-ahead landings run concurrently with
+    the serving load (registry + lander contention)."""
     import concurrent.futures as cf
     import random
+
+    if prefetch:
+        from demodel_amd.engine.pull import LanderPool
+        from demodel_amd.engine.registry import BlobRegistry
+
+        stack.cfg.gpu_prefetch = "auto"
+        stack.proxy.prefetch_landers = LanderPool(0, gpu=False)
+        stack.proxy.registry = BlobRegistry()
 
     blobs = {}
     paths = {}
