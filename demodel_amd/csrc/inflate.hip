@@ -298,6 +298,25 @@ inflate_kernel(InflateDesc* __restrict__ descs, int n_streams) {
       LGKM0();
 
       // ---- symbol loop (all lanes lockstep, zero barriers) ---------
+      // Literal batching: the decode is redundant-lockstep, so every
+      // lane sees every literal; lane (n & 63) parks literal n in a
+      // register and 64 accumulate into ONE coalesced 64 B store
+      // (round-1 wrote each literal via lane 0 with 63 lanes idle —
+      // the first wall on literal-heavy payloads).  Pending bytes are
+      // flushed before any match (matches read the window) and at
+      // block end.
+      uint64_t lit_base = 0;
+      uint32_t lit_n = 0;
+      uint8_t lit_pend = 0;
+      auto flush_lits = [&]() {
+        if (lit_n) {
+          if (lane < (int)lit_n) {
+            out[lit_base + lane] = lit_pend;
+            sh.win[(lit_base + lane) & DWMASK] = lit_pend;
+          }
+          lit_n = 0;
+        }
+      };
       while (true) {
         int sym = huff_decode_lut(&br, sh.litlen_lut, &sh.litlen_t);
         if (sym < 0 || br.overran()) {
@@ -306,11 +325,11 @@ inflate_kernel(InflateDesc* __restrict__ descs, int n_streams) {
         }
         if (sym < 256) {
           if (pos >= d->dst_cap) { status = INF_ERR_OVERFLOW; break; }
-          if (lane == 0) {
-            out[pos] = (uint8_t)sym;
-            sh.win[pos & DWMASK] = (uint8_t)sym;
-          }
+          if (lit_n == 0) lit_base = pos;
+          if (lane == (int)lit_n) lit_pend = (uint8_t)sym;
+          ++lit_n;
           ++pos;
+          if (lit_n == 64) flush_lits();
           continue;
         }
         if (sym == 256) break;  // end of block
@@ -322,6 +341,7 @@ inflate_kernel(InflateDesc* __restrict__ descs, int n_streams) {
         uint32_t dist = kDistBase[dsym] + br.bits(kDistExtra[dsym]);
         if (dist > pos) { status = INF_ERR_FORMAT; break; }
         if (pos + mlen > d->dst_cap) { status = INF_ERR_OVERFLOW; break; }
+        flush_lits();  // the match may read the just-written window
         // every source byte is within DWIN -> always LDS
         LGKM0();
         if (dist >= mlen) {
@@ -345,6 +365,7 @@ inflate_kernel(InflateDesc* __restrict__ descs, int n_streams) {
         }
         pos += mlen;
       }
+      flush_lits();  // trailing literals of the block
       if (status != INF_OK) break;
       if (bfinal) done = true;
     }
