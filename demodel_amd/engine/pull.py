@@ -166,8 +166,12 @@ def _etag_sha256(etag: str | None) -> str | None:
     return None
 
 
-SEGMENT_MIN = int(os.environ.get("DEMODEL_SEGMENT_MIN_MB", "512")) << 20
-MAX_SEGMENTS = int(os.environ.get("DEMODEL_MAX_SEGMENTS", "8"))
+# Segment defaults from the call-10 hardware sweep: a single-file
+# 4.5 GB pull runs 23.3 GB/s at 8x512 MiB segments, 25.3 at 16x256 MiB,
+# and REGRESSES to 17.1 at 24 segments + 16 workers (thread/conn
+# oversubscription); multi-file pulls are insensitive (31.0 vs 31.7).
+SEGMENT_MIN = int(os.environ.get("DEMODEL_SEGMENT_MIN_MB", "256")) << 20
+MAX_SEGMENTS = int(os.environ.get("DEMODEL_MAX_SEGMENTS", "16"))
 RESUME_RETRIES = 4
 
 
