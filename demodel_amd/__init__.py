@@ -12,13 +12,17 @@ see cmd/demodel/main.go:59) rebuilt from scratch MI355X-first:
   * a GPU landing pipeline: chunked blob downloads stream through a
     pinned host ring into HBM3E via hipMemcpyAsync on side streams,
     with hand-written CDNA4 (gfx950) HIP kernels for SHA-256 verify,
-    gzip/zstd inflate, safetensors tensor-scatter and GGUF q4 dequant,
+    gzip/zstd/snappy/LZ4 decompression, safetensors tensor-scatter and
+    GGUF dequantization (q4_0..q8_0 + q2_K..q6_K -> bf16),
+  * proxy -> HBM pull-ahead: blobs the proxy caches can land into a
+    device-resident registry so later engine pulls are GPU-warm
+    (POST /__demodel/prefetch or DEMODEL_GPU_PREFETCH=auto),
   * RCCL-over-xGMI fan-out: broadcast of a pulled model to all GPUs of
     a node, and sharded pulls reassembled with all-gather overlapped
     with the next chunk's download.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from .config import Config, load_config  # noqa: F401
 
