@@ -135,3 +135,38 @@ def test_mitm_all_and_no_mitm_flags(tmp_path):
     cfg = Config()
     assert cfg.should_mitm("huggingface.co:443")  # bug-fixed default list
     assert not cfg.should_mitm("example.com:443")
+
+
+def test_mitm_concurrent_soak(mitm_stack, tmp_path):
+    """16 threads of MITM'd HTTPS pulls at once: leaf minting memoizes
+    under concurrency, the TLS relay (asyncio fallback path — no raw
+    socket) stays byte-exact, and upstream TLS conns pool."""
+    import concurrent.futures as cf
+    import random
+
+    s = mitm_stack
+    blobs = {}
+    paths = {}
+    for i in range(4):
+        data = os.urandom((i + 1) * 200_000)
+        p = tmp_path / f"m{i}.bin"
+        p.write_bytes(data)
+        blobs[f"m{i}.bin"] = data
+        paths[f"m{i}.bin"] = str(p)
+    s.origin.add_hf_repo("org/mitm", paths)
+    rng = random.Random(3)
+
+    def one(k):
+        name = rng.choice(list(blobs))
+        url = (f"https://127.0.0.1:{s.origin_port}"
+               f"/org/mitm/resolve/main/{name}")
+        status, headers, body = _https_via_proxy(
+            url, s.proxy_port, s.client_cafile)
+        assert status == 200
+        assert body == blobs[name], name
+        return True
+
+    with cf.ThreadPoolExecutor(max_workers=16) as ex:
+        assert all(ex.map(one, range(48)))
+    # one leaf minted total, not one per connection
+    assert len(s.proxy.leafs._pems) == 1
