@@ -151,7 +151,9 @@ async def relay_body_threaded(up_r: asyncio.StreamReader,
     cl_sock = writer.transport.get_extra_info("socket")
     if (up_sock is None or cl_sock is None
             or up_w.transport.get_extra_info("sslcontext") is not None
-            or writer.transport.get_extra_info("sslcontext") is not None):
+            or writer.transport.get_extra_info("sslcontext") is not None
+            # CPython StreamReader internal; absent -> asyncio fallback
+            or not isinstance(getattr(up_r, "_buffer", None), bytearray)):
         raise NotImplementedError
     await writer.drain()
     up_w.transport.pause_reading()
