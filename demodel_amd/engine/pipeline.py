@@ -319,6 +319,50 @@ class Lander:
                       verify_chunk: int | None = None) -> bytes:
         return self._gpu_chunk_digests(buf, nbytes, verify_chunk)
 
+    def collect_digests(self, dig_dev, n_chunks: int, events) -> bytes:
+        """Wait the given hash events on this verify stream, read the
+        digest array back, and return canonical sha256 bytes."""
+        import ctypes
+
+        import numpy as np
+
+        for e in events:
+            e.wait(self.verify_stream.handle)
+        host = bytearray(n_chunks * 32)
+        addr = ctypes.addressof(
+            (ctypes.c_char * len(host)).from_buffer(host))
+        self._h.d2h_async(addr, dig_dev.ptr, n_chunks * 32,
+                          self.verify_stream.handle)
+        self.verify_stream.sync()
+        return np.frombuffer(bytes(host), dtype="<u4").byteswap().tobytes()
+
+    def hash_range_into(self, buf, lo: int, hi: int, vc: int, dig_dev,
+                        file_size: int):
+        """Launch chunk digests for the chunk-ALIGNED part of [lo, hi)
+        into dig_dev (32 B per chunk, indexed by absolute chunk id), on
+        this lander's verify stream, ordered after its copies.
+
+        Incremental verification: a segment hashes WHILE other segments
+        still download, so the end-of-blob verify tail shrinks to the
+        few boundary chunks + one D2H (round-1 hashed the whole buffer
+        after the last byte landed).  Returns (event, lo_chunk,
+        hi_chunk) or None when no full chunk lies inside."""
+        h = self._h
+        lo_c = (lo + vc - 1) // vc
+        hi_c = (file_size + vc - 1) // vc if hi >= file_size else hi // vc
+        if hi_c <= lo_c:
+            return None
+        done = h.Event()
+        done.record(self.copy_stream.handle)
+        done.wait(self.verify_stream.handle)
+        span = min(file_size, hi_c * vc) - lo_c * vc
+        h.sha256_batch(buf.ptr + lo_c * vc, span, vc,
+                       dig_dev.ptr + lo_c * 32, hi_c - lo_c,
+                       self.verify_stream.handle)
+        ev = h.Event()
+        ev.record(self.verify_stream.handle)
+        return ev, lo_c, hi_c
+
     def _gpu_chunk_digests(self, buf, nbytes: int,
                            verify_chunk: int | None = None) -> bytes:
         h = self._h

@@ -369,10 +369,19 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
 
     from .pipeline import LandingError
 
+    # Incremental verification: each segment hashes its chunk-ALIGNED
+    # range the moment it finishes landing (on that lander's verify
+    # stream), so the end-of-blob verify work shrinks to the few
+    # boundary chunks + one digest D2H — round 1 hashed the whole
+    # buffer serially after the last byte.
+    vc = verify_chunk or lander0.verify_chunk
+    n_chunks = (total + vc - 1) // vc
+    dig_dev = lander0._h.DeviceBuffer(n_chunks * 32)
+
     def land_range(i):
         lo, hi = bounds[i], bounds[i + 1]
         if hi <= lo:
-            return 0.0
+            return None
         lander = landers.get()
         at = lo
         attempt = 0
@@ -400,9 +409,10 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
             finally:
                 s.close()
         lander.sync()
+        hashed = lander.hash_range_into(buf, lo, hi, vc, dig_dev, total)
         if on_range is not None:
             on_range(name, lo, hi, buf, None)
-        return 0.0
+        return hashed
 
     futs = [seg_executor.submit(land_range, i) for i in range(1, n_segs)]
     # this thread lands segment 0 from the open stream (it was asked for
@@ -437,6 +447,8 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
             finally:
                 s.close()
     lander0.sync()
+    covered = [lander0.hash_range_into(buf, 0, bounds[1], vc, dig_dev,
+                                       total)]
     if head is None:
         # seg-0 fallback dropped its in-flight head capture: rebuild the
         # SEGMENT-0 portion now (other segments may still be landing)
@@ -444,17 +456,35 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     if on_range is not None:
         on_range(name, 0, bounds[1], buf, bytes(head))
     for f in futs:
-        f.result()
+        covered.append(f.result())
     # head spans min(total, head_bytes), which can extend past segment 0
     # (small files under a small SEGMENT_MIN): complete it now that every
     # segment's lander has synced
     if len(head) < min(total, lander0.head_bytes):
         head = lander0.read_head(buf, total)
-    vc = verify_chunk or lander0.verify_chunk
     blob = LandedBlob(nbytes=total, device=f"cuda:{lander0.device_index}",
                       buffer=buf, verify_chunk=vc, head=bytes(head))
     blob._ranged = True  # on_range already fired per segment
-    blob.digest_blob = lander0._gpu_chunk_digests(buf, total, vc)
+    # boundary chunks no segment covered (chunk-unaligned bounds), plus
+    # any too-small segment's range: hash the gap runs now — every
+    # landing is host-synced at this point
+    covered = [c for c in covered if c is not None]
+    ivs = sorted((lo_c, hi_c) for _, lo_c, hi_c in covered)
+    events = [ev for ev, _, _ in covered]
+    h = lander0._h
+    at = 0
+    gaps = []
+    for lo_c, hi_c in ivs:
+        if lo_c > at:
+            gaps.append((at, lo_c))
+        at = max(at, hi_c)
+    if at < n_chunks:
+        gaps.append((at, n_chunks))
+    for a, b in gaps:
+        span = min(total, b * vc) - a * vc
+        h.sha256_batch(buf.ptr + a * vc, span, vc, dig_dev.ptr + a * 32,
+                       b - a, lander0.verify_stream.handle)
+    blob.digest_blob = lander0.collect_digests(dig_dev, n_chunks, events)
     if expected_digests is not None:
         from .pipeline import check_digests
 
