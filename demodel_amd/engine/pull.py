@@ -374,9 +374,11 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     # stream), so the end-of-blob verify work shrinks to the few
     # boundary chunks + one digest D2H — round 1 hashed the whole
     # buffer serially after the last byte.
+    inc_verify = os.environ.get("DEMODEL_INC_VERIFY", "1") != "0"
     vc = verify_chunk or lander0.verify_chunk
     n_chunks = (total + vc - 1) // vc
-    dig_dev = lander0._h.DeviceBuffer(n_chunks * 32)
+    dig_dev = (lander0._h.DeviceBuffer(n_chunks * 32)
+               if inc_verify else None)
 
     def land_range(i):
         lo, hi = bounds[i], bounds[i + 1]
@@ -409,7 +411,8 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
             finally:
                 s.close()
         lander.sync()
-        hashed = lander.hash_range_into(buf, lo, hi, vc, dig_dev, total)
+        hashed = (lander.hash_range_into(buf, lo, hi, vc, dig_dev, total)
+                  if inc_verify else None)
         if on_range is not None:
             on_range(name, lo, hi, buf, None)
         return hashed
@@ -448,7 +451,7 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
                 s.close()
     lander0.sync()
     covered = [lander0.hash_range_into(buf, 0, bounds[1], vc, dig_dev,
-                                       total)]
+                                       total) if inc_verify else None]
     if head is None:
         # seg-0 fallback dropped its in-flight head capture: rebuild the
         # SEGMENT-0 portion now (other segments may still be landing)
@@ -465,6 +468,13 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     blob = LandedBlob(nbytes=total, device=f"cuda:{lander0.device_index}",
                       buffer=buf, verify_chunk=vc, head=bytes(head))
     blob._ranged = True  # on_range already fired per segment
+    if not inc_verify:  # A/B fallback: whole-buffer tail hash
+        blob.digest_blob = lander0._gpu_chunk_digests(buf, total, vc)
+        if expected_digests is not None:
+            from .pipeline import check_digests
+
+            check_digests(blob.digest_blob, expected_digests, vc)
+        return blob
     # boundary chunks no segment covered (chunk-unaligned bounds), plus
     # any too-small segment's range: hash the gap runs now — every
     # landing is host-synced at this point
