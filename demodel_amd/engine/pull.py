@@ -374,7 +374,14 @@ def _pull_segmented(landers: LanderPool, url: str, total: int, src0,
     # stream), so the end-of-blob verify work shrinks to the few
     # boundary chunks + one digest D2H — round 1 hashed the whole
     # buffer serially after the last byte.
-    inc_verify = os.environ.get("DEMODEL_INC_VERIFY", "1") != "0"
+    # Incremental per-segment verification measured a ~25% REGRESSION
+    # on the flagship (same-box A/B, profiles/bench_history.md): the
+    # many small hash launches across per-lander verify streams
+    # multiplex onto the 4 HW queues and SERIALIZE with the H2D copy
+    # streams mid-landing, while the single whole-buffer tail hash is a
+    # big efficient launch that overlaps across files anyway.  Kept
+    # selectable for re-evaluation (DEMODEL_INC_VERIFY=1), default off.
+    inc_verify = os.environ.get("DEMODEL_INC_VERIFY", "0") == "1"
     vc = verify_chunk or lander0.verify_chunk
     n_chunks = (total + vc - 1) // vc
     dig_dev = (lander0._h.DeviceBuffer(n_chunks * 32)
