@@ -281,3 +281,30 @@ def test_gpu_buffer_recycle(stack, tmp_path):
     f2 = res2.files[0]
     assert f2.blob.buffer.ptr == ptr0  # same buffer came back
     assert bytes(f2.blob.torch_u8().cpu().numpy().tobytes()) == data
+
+
+@pytest.mark.parametrize("inc", ["0", "1"])
+def test_segmented_digests_match_host(stack, tmp_path, monkeypatch, inc):
+    """Segmented-pull chunk digests equal a host-computed record for
+    BOTH verification schedules (tail hash = default; incremental =
+    DEMODEL_INC_VERIFY=1, kept as a documented experiment), including
+    ragged, chunk-unaligned segment bounds."""
+    import hashlib
+
+    import demodel_amd.engine.pull as pm
+
+    monkeypatch.setenv("DEMODEL_INC_VERIFY", inc)
+    monkeypatch.setattr(pm, "SEGMENT_MIN", 3 << 20)
+    monkeypatch.setattr(pm, "MAX_SEGMENTS", 7)  # unaligned bounds
+    data = os.urandom((19 << 20) + 4321)
+    p = tmp_path / "rag.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/rag", {"rag.bin": str(p)})
+    res = pm.pull_hf("org/rag", endpoint=stack.origin_base,
+                     verify="chunked", workers=4)
+    f = [x for x in res.files if x.name == "rag.bin"][0]
+    vc = f.blob.verify_chunk
+    want = b"".join(hashlib.sha256(data[o:o + vc]).digest()
+                    for o in range(0, len(data), vc))
+    assert f.blob.digest_blob == want
+    assert bytes(f.blob.torch_u8().cpu().numpy().tobytes()) == data

@@ -85,3 +85,35 @@ def test_peer_verify_detects_tamper(stack, tmp_path):
         pull_mod.pull_hf("org/p", endpoint=stack.endpoint,
                          verify="chunked", workers=1, peer_verify=True)
     assert ei.value.chunk_index == 2_000_000 // (64 << 10)
+
+
+def test_pull_cli_peer_verify(stack, tmp_path, capsys):
+    """`demodel pull --peer-verify` verifies against the peer's digest
+    record end-to-end through the CLI."""
+    import json as _json
+    import time
+
+    from demodel_amd.cli import main as cli_main
+    from demodel_amd.engine.pull import fetch_peer_digests
+
+    data = os.urandom(300_000)
+    p = tmp_path / "pv.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/pvcli", {"pv.bin": str(p)})
+    # prime the peer's cache + wait for its async digest record
+    import urllib.request
+
+    urllib.request.urlopen(
+        f"{stack.endpoint}/org/pvcli/resolve/main/pv.bin",
+        timeout=20).read()
+    path = "/org/pvcli/resolve/main/pv.bin"
+    t0 = time.time()
+    while fetch_peer_digests(stack.endpoint, path) is None:
+        assert time.time() - t0 < 30
+        time.sleep(0.05)
+    rc = cli_main(["pull", "hf://org/pvcli", "--cpu", "--peer-verify",
+                   "--endpoint", stack.endpoint])
+    assert rc == 0
+    out = _json.loads(capsys.readouterr().out)
+    f = [x for x in out["files"] if x["name"] == "pv.bin"][0]
+    assert f["bytes"] == len(data)
