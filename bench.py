@@ -210,6 +210,9 @@ def main():
 
     proxy = None
     if args.via != "direct":
+        assert not args.virtual, (
+            "--via proxy/peer with --virtual would spool the virtual "
+            "blob set to the proxy's DISK cache (e.g. 141 GB)")
         # the client-facing data plane (reference hot loop,
         # start.go:201-204): engine pulls go THROUGH the proxy
         from demodel_amd.config import Config
