@@ -97,7 +97,13 @@ def _cmd_stats(args) -> int:
     from .engine import fetch
 
     base = (args.endpoint or "http://127.0.0.1:8080").rstrip("/")
-    print(json.dumps(fetch.get_json(base + "/__demodel/stats"), indent=1))
+    try:
+        print(json.dumps(fetch.get_json(base + "/__demodel/stats"),
+                         indent=1))
+    except (fetch.FetchError, OSError) as e:
+        print(f"error: no demodel proxy reachable at {base} ({e})",
+              file=sys.stderr)
+        return 1
     return 0
 
 
@@ -108,16 +114,25 @@ def _cmd_prefetch(args) -> int:
 
     base = (args.endpoint or "http://127.0.0.1:8080").rstrip("/")
     url = base + "/__demodel/prefetch"
-    if not args.paths:
-        with urllib.request.urlopen(url, timeout=30) as r:
-            print(json.dumps(json.loads(r.read()), indent=1))
-        return 0
-    req = urllib.request.Request(
-        url, method="POST",
-        data=json.dumps({"paths": args.paths}).encode(),
-        headers={"Content-Type": "application/json"})
-    with urllib.request.urlopen(req, timeout=30) as r:
-        out = json.loads(r.read())
+    try:
+        if not args.paths:
+            with urllib.request.urlopen(url, timeout=30) as r:
+                print(json.dumps(json.loads(r.read()), indent=1))
+            return 0
+        req = urllib.request.Request(
+            url, method="POST",
+            data=json.dumps({"paths": args.paths}).encode(),
+            headers={"Content-Type": "application/json"})
+        with urllib.request.urlopen(req, timeout=30) as r:
+            out = json.loads(r.read())
+    except urllib.error.HTTPError as e:  # before OSError: subclass
+        print(f"error: {e} — is the proxy running with prefetch "
+              f"landers (--gpu-prefetch)?", file=sys.stderr)
+        return 1
+    except OSError as e:
+        print(f"error: no demodel proxy reachable at {base} ({e})",
+              file=sys.stderr)
+        return 1
     print(json.dumps(out, indent=1))
     return 0 if out.get("queued") is not None else 1
 
