@@ -63,7 +63,8 @@ def make_model_files(model: str, data_dir: str, n_shards_override=None,
     if model == "parquet":
         return synth.write_parquet_shards(data_dir,
                                           compression=parquet_codec)
-    if model in ("gguf-8b", "gguf-tiny"):
+    if model.startswith("gguf"):
+        assert model != "gguf-70b", "gguf-70b needs --virtual (41 GB)"
         geom = synth.LLAMA3_8B if model == "gguf-8b" else TINY_GEOM
         path = os.path.join(data_dir, "model.gguf")
         os.makedirs(data_dir, exist_ok=True)
@@ -506,7 +507,7 @@ def main():
 
     def one_step(record_digests=False):
         nonlocal step_bytes
-        if args.model in ("gguf-8b", "gguf-tiny"):
+        if args.model.startswith("gguf"):
             return gguf_step(record_digests)
         if args.model == "dataset":
             res, n = dataset_step(record_digests)
