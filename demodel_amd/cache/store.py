@@ -246,7 +246,17 @@ class CacheStore:
         under max_bytes.  Recency = body file atime (falls back to
         mtime).  Returns {kept, evicted, bytes}."""
         entries = []
+        now = time.time()
         for fn in os.listdir(self.root):
+            # orphaned fill temps (process died mid-fill): reap after 1 h
+            if fn.endswith(".part"):
+                p = os.path.join(self.root, fn)
+                try:
+                    if now - os.path.getmtime(p) > 3600:
+                        os.unlink(p)
+                except OSError:
+                    pass
+                continue
             if not fn.endswith(".meta"):
                 continue
             body = os.path.join(self.root, fn[:-5])

@@ -133,3 +133,22 @@ def test_verify_cache_cli(tmp_path):
         f.write(b"\x00\x01\x02")
     res2 = verify_cache(cfg)
     assert not res2["ok"] and res2["bad"] == ["https://host/v"]
+
+
+def test_gc_reaps_orphaned_part_files(tmp_path):
+    """A fill temp left by a crashed process is reaped by gc once
+    stale; fresh temps (an active fill) are left alone."""
+    import os
+    import time
+
+    from demodel_amd.cache import CacheStore
+
+    store = CacheStore(str(tmp_path))
+    stale = tmp_path / ".deadbeef.abc.part"
+    stale.write_bytes(b"x" * 100)
+    os.utime(stale, (time.time() - 7200, time.time() - 7200))
+    fresh = tmp_path / ".cafef00d.xyz.meta.part"
+    fresh.write_bytes(b"y")
+    store.gc(10 << 30)
+    assert not stale.exists()
+    assert fresh.exists()
