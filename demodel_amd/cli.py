@@ -133,6 +133,28 @@ def _cmd_prefetch(args) -> int:
         print(f"error: no demodel proxy reachable at {base} ({e})",
               file=sys.stderr)
         return 1
+    if getattr(args, "wait", False):
+        import time
+
+        deadline = time.time() + args.timeout
+        left = set(args.paths)
+        while left:
+            if time.time() > deadline:
+                print(f"error: timed out waiting for {sorted(left)}",
+                      file=sys.stderr)
+                return 1
+            time.sleep(0.2)
+            with urllib.request.urlopen(url, timeout=30) as r:
+                st = json.loads(r.read())
+            left -= set(st.get("registered", []))
+            # a path that is neither registered nor in flight failed
+            # (lands register BEFORE leaving the in-flight set)
+            gone = left - set(st.get("in_flight", []))
+            if gone:
+                print(f"error: prefetch failed for {sorted(gone)}",
+                      file=sys.stderr)
+                return 1
+        out["registered"] = args.paths
     print(json.dumps(out, indent=1))
     return 0 if out.get("queued") is not None else 1
 
@@ -212,6 +234,10 @@ def main(argv: list[str] | None = None) -> int:
                          "/org/repo/resolve/main/model.safetensors")
     pf.add_argument("--endpoint", default=None,
                     help="proxy base URL (default http://127.0.0.1:8080)")
+    pf.add_argument("--wait", action="store_true",
+                    help="block until every path is registered "
+                         "(HBM-resident) or --timeout expires")
+    pf.add_argument("--timeout", type=float, default=600.0)
     pf.set_defaults(fn=_cmd_prefetch)
 
     args = p.parse_args(argv)

@@ -204,3 +204,21 @@ def test_recycle_skips_shared_blobs():
     r.files = [f]
     assert pool.recycle(r) == 0
     assert f.blob.buffer is not None
+
+
+def test_prefetch_cli_wait(stack, tmp_path, capsys):
+    from demodel_amd.cli import main as cli_main
+
+    data = _mk_repo(stack, tmp_path, "org/cliw", "w2.safetensors",
+                    nbytes=200_000)
+    path = "/org/cliw/resolve/main/w2.safetensors"
+    rc = cli_main(["prefetch", path, "--wait", "--timeout", "30",
+                   "--endpoint", stack.endpoint])
+    assert rc == 0
+    out = json.loads(capsys.readouterr().out)
+    assert out["registered"] == [path]
+    assert bytes(stack.proxy.registry.get(path).buffer) == data
+    # a bogus path fails within the timeout instead of hanging
+    rc = cli_main(["prefetch", "/no/such/thing.safetensors", "--wait",
+                   "--timeout", "20", "--endpoint", stack.endpoint])
+    assert rc == 1
