@@ -96,8 +96,18 @@ class _Reader:
         self.o += struct.calcsize(fmt)
         return v[0] if len(v) == 1 else v
 
+    def _need(self, n: int) -> None:
+        # untrusted input: a declared length beyond the remaining bytes
+        # must fail loudly, not hang a billion-element loop or walk
+        # struct.unpack into the weeds
+        if n < 0 or n > len(self.d) - self.o:
+            raise ValueError(
+                f"gguf: declared length {n} exceeds remaining "
+                f"{len(self.d) - self.o} bytes")
+
     def take_str(self) -> str:
         n = self.take("<Q")
+        self._need(n)
         s = self.d[self.o:self.o + n].decode("utf-8", "replace")
         self.o += n
         return s
@@ -110,6 +120,7 @@ class _Reader:
         if vt == 9:
             et = self.take("<I")
             n = self.take("<Q")
+            self._need(n)  # every element is >= 1 byte
             return [self.take_value(et) for _ in range(n)]
         raise ValueError(f"unknown gguf value type {vt}")
 
@@ -122,6 +133,12 @@ def parse_bytes(head: bytes) -> GGUFModel:
     version = r.take("<I")
     n_tensors = r.take("<Q")
     n_kv = r.take("<Q")
+    # every entry consumes multiple bytes; counts beyond the data size
+    # are crafted (would loop for ages before running dry)
+    if n_tensors > len(head) or n_kv > len(head):
+        raise ValueError(
+            f"gguf: implausible counts ({n_tensors} tensors, {n_kv} kv "
+            f"in {len(head)} bytes)")
     kv = {}
     for _ in range(n_kv):
         k = r.take_str()

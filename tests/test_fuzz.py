@@ -112,3 +112,41 @@ def test_http1_request_head_roundtrip(method, path, headers):
     assert got.target == path
     # full ordered list (duplicate names allowed; .get returns first)
     assert got.headers == [(k, v) for k, v in headers]
+
+
+# --- GGUF header parser: untrusted-registry input ---------------------
+
+@settings(max_examples=200, deadline=None)
+@given(garbage=st.binary(min_size=0, max_size=3000))
+def test_gguf_parse_garbage_fails_loudly(garbage):
+    """Arbitrary bytes (including a valid magic prefix) must raise a
+    clean error — never hang, crash, or allocate per declared counts."""
+    import struct as _struct
+
+    from demodel_amd.engine.formats import gguf
+
+    for payload in (garbage, b"GGUF" + garbage):
+        try:
+            gguf.parse_bytes(payload)
+        except (ValueError, _struct.error):
+            pass
+
+
+@settings(max_examples=60, deadline=None)
+@given(names=st.lists(
+    st.text(alphabet="abcdefgh.0123456789_", min_size=1, max_size=30),
+    min_size=0, max_size=6, unique=True),
+    qtype=st.sampled_from([0, 2, 8, 12, 14]))
+def test_gguf_header_roundtrip(names, qtype):
+    from demodel_amd.engine.formats import gguf
+
+    blocked = 32 if qtype in (2, 8) else 256
+    dims = (blocked * 2, 3) if qtype != 0 else (5, 7)
+    tensors = [(n, dims, qtype) for n in names]
+    prefix, total = gguf.build_virtual(tensors)
+    gg = gguf.parse_bytes(prefix)
+    assert [t.name for t in gg.tensors] == names
+    for t in gg.tensors:
+        assert t.dims == dims
+        assert t.type_id == qtype
+    assert total >= gg.data_offset
