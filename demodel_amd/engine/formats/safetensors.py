@@ -72,6 +72,11 @@ def parse_header(prefix: bytes) -> SafetensorsHeader:
             raise ValueError(f"unsupported safetensors dtype {tag!r}")
         tname, isz = _DTYPES[tag]
         b, e = spec["data_offsets"]
+        # untrusted input: negative offsets would make the python-slice
+        # views silently read from the blob's END
+        if not (isinstance(b, int) and isinstance(e, int)
+                and 0 <= b <= e):
+            raise ValueError(f"bad data_offsets for {name!r}: {b}, {e}")
         tensors.append(TensorInfo(
             name=name, st_dtype=tag, torch_dtype=tname, itemsize=isz,
             shape=tuple(spec["shape"]), begin=b, end=e))

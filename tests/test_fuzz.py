@@ -150,3 +150,32 @@ def test_gguf_header_roundtrip(names, qtype):
         assert t.dims == dims
         assert t.type_id == qtype
     assert total >= gg.data_offset
+
+
+# --- safetensors header parser: untrusted input -----------------------
+
+@settings(max_examples=200, deadline=None)
+@given(garbage=st.binary(min_size=0, max_size=2000))
+def test_safetensors_parse_garbage_fails_loudly(garbage):
+    import json as _json
+
+    from demodel_amd.engine.formats import safetensors as stf
+
+    try:
+        stf.parse_header(garbage)
+    except (ValueError, _json.JSONDecodeError, KeyError, TypeError,
+            AttributeError):
+        pass
+
+
+def test_safetensors_rejects_negative_offsets():
+    import json as _json
+    import struct as _struct
+
+    from demodel_amd.engine.formats import safetensors as stf
+
+    hdr = _json.dumps({"w": {"dtype": "F32", "shape": [2],
+                             "data_offsets": [-8, 0]}}).encode()
+    blob = _struct.pack("<Q", len(hdr)) + hdr
+    with pytest.raises(ValueError):
+        stf.parse_header(blob)
