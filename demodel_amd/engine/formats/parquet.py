@@ -46,6 +46,22 @@ class _Compact:
         self.p += 1
         return b
 
+    def _need(self, n: int) -> None:
+        # untrusted input: a declared element count beyond the
+        # remaining bytes would spin a near-infinite skip loop
+        if n < 0 or n > len(self.d) - self.p:
+            raise ValueError(
+                f"thrift: declared {n} elements with "
+                f"{len(self.d) - self.p} bytes left")
+
+    def skip_elem(self, ctype):
+        """Skip one LIST/SET/MAP element: unlike struct FIELDS, bool
+        elements are encoded as one byte in the compact protocol."""
+        if ctype in (_CT_TRUE, _CT_FALSE):
+            self.byte()
+        else:
+            self.skip(ctype)
+
     def varint(self):
         out = 0
         shift = 0
@@ -78,15 +94,17 @@ class _Compact:
             et = b & 0xF
             if n == 15:
                 n = self.varint()
+            self._need(n)  # every element is >= 1 byte
             for _ in range(n):
-                self.skip(et)
+                self.skip_elem(et)
         elif ctype == _CT_MAP:
             n = self.varint()
+            self._need(n)
             if n:
                 kv = self.byte()
                 for _ in range(n):
-                    self.skip(kv >> 4)
-                    self.skip(kv & 0xF)
+                    self.skip_elem(kv >> 4)
+                    self.skip_elem(kv & 0xF)
         elif ctype == _CT_STRUCT:
             last = 0
             while True:

@@ -179,3 +179,34 @@ def test_safetensors_rejects_negative_offsets():
     blob = _struct.pack("<Q", len(hdr)) + hdr
     with pytest.raises(ValueError):
         stf.parse_header(blob)
+
+
+# --- parquet thrift compact walker: untrusted input -------------------
+
+def test_thrift_boolean_list_bomb_raises_fast():
+    """Regression: a crafted bool-list header declared 10^15 elements;
+    the walker treated struct-bool encoding (zero bytes) as the element
+    encoding and would spin for hours.  Compact-protocol list elements
+    of bool are ONE byte each, and declared counts are bounded by the
+    remaining bytes."""
+    import time
+
+    from demodel_amd.engine.formats import parquet as pqf
+
+    # struct field 1, type LIST; list header: n=15 -> varint, elem=TRUE
+    bomb = bytes([0x19, 0xF1]) + b"\xff\xff\xff\xff\xff\xff\x7f"
+    t0 = time.time()
+    with pytest.raises((ValueError, IndexError)):
+        pqf.parse_page_header(bomb, 0)
+    assert time.time() - t0 < 1.0
+
+
+@settings(max_examples=300, deadline=None)
+@given(garbage=st.binary(min_size=0, max_size=400))
+def test_parquet_page_header_garbage_fails_loudly(garbage):
+    from demodel_amd.engine.formats import parquet as pqf
+
+    try:
+        pqf.parse_page_header(garbage, 0)
+    except (ValueError, IndexError, RecursionError):
+        pass
