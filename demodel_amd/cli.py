@@ -34,6 +34,32 @@ def _cmd_start(args) -> int:
         cfg.port = args.port
     if getattr(args, "gpu_prefetch", None) is not None:
         cfg.gpu_prefetch = args.gpu_prefetch
+    if getattr(args, "loops", None) is not None:
+        cfg.loops = args.loops
+    if getattr(cfg, "loops", 1) > 1:
+        # SO_REUSEPORT fleet: one event loop per core slice, so MITM
+        # TLS crypto scales (proxy/server.py ProxyFleet)
+        import threading
+
+        from .ca import read_or_new_ca
+        from .certs import LeafStore
+        from .proxy.server import ProxyFleet
+
+        ca = read_or_new_ca(cfg.ca_use_ecdsa)
+        landers = None
+        if getattr(cfg, "gpu_prefetch", "off") != "off":
+            from .engine.pull import LanderPool
+            from .gpu import have_gpu
+
+            landers = LanderPool(0, gpu=True if have_gpu() else False)
+        fleet = ProxyFleet(cfg, leafs=LeafStore(ca),
+                           prefetch_landers=landers)
+        fleet.start()
+        try:
+            threading.Event().wait()
+        except KeyboardInterrupt:
+            fleet.close()
+        return 0
     try:
         asyncio.run(run_proxy(cfg))
     except KeyboardInterrupt:
@@ -178,6 +204,10 @@ def main(argv: list[str] | None = None) -> int:
 
     sp = sub.add_parser("start", help="run the caching proxy")
     sp.add_argument("--port", type=int, default=None)
+    sp.add_argument("--loops", type=int, default=None,
+                    help="acceptor event loops on one port "
+                         "(SO_REUSEPORT); >1 scales MITM TLS across "
+                         "cores (also DEMODEL_LOOPS)")
     sp.add_argument("--gpu-prefetch", default=None,
                     choices=["off", "auto"],
                     help="land proxy-cached blobs into HBM ahead of "

@@ -64,6 +64,9 @@ class Config:
     mitm_hosts: list[str] = field(default_factory=lambda: list(DEFAULT_MITM_HOSTS))
 
     # --- listener ---
+    # acceptor event loops on ONE port (SO_REUSEPORT).  >1 scales MITM
+    # TLS crypto across cores (proxy/server.py ProxyFleet).
+    loops: int = 1
     # 0.0.0.0:8080 mirrors the reference (start.go:206) for drop-in
     # parity, but note what it means: an UNAUTHENTICATED forward proxy
     # with a CONNECT tunnel reachable on every interface.  Set
@@ -116,6 +119,7 @@ def load_config(**overrides) -> Config:
         no_mitm=_env_bool("DEMODEL_PROXY_NO_MITM"),
         mitm_hosts=hosts,
         host=os.environ.get("DEMODEL_HOST", "0.0.0.0"),
+        loops=_env_int("DEMODEL_LOOPS", 1),
         port=_env_int("DEMODEL_PORT", 8080),
         cache_dir=os.environ.get("DEMODEL_CACHE_DIR", ".cache"),
         cache_max_bytes=(
