@@ -296,3 +296,43 @@ def test_progressive_dequant_range_merging():
     assert pd._ranges == [(0, 400)]
     pd._add_range(50, 150)               # duplicate/overlap is harmless
     assert pd._ranges == [(0, 400)]
+
+
+def test_virtual_origin_prefix_and_ranges(stack, tmp_path):
+    """Virtual blobs with a prefix (the gguf-70b bench's REAL header +
+    patterned payload) serve correct bytes for full GETs and Ranges
+    spanning the prefix/pattern boundary."""
+    import urllib.request
+
+    from demodel_amd.testing.origin import FakeOrigin
+
+    prefix = bytes(range(256)) * 4          # 1024-byte "header"
+    total = 200_000
+    stack.origin.add_hf_repo_virtual(
+        "org/virt", {"v.bin": total}, prefixes={"v.bin": prefix})
+    url = f"{stack.origin_base}/cdn/org/virt/x/v.bin"
+    pattern = FakeOrigin._virtual_pattern()
+
+    def expect(lo, hi):  # [lo, hi)
+        out = bytearray()
+        for p in range(lo, hi):
+            out += (prefix[p:p + 1] if p < len(prefix)
+                    else pattern[p % len(pattern):p % len(pattern) + 1])
+        return bytes(out)
+
+    with urllib.request.urlopen(url, timeout=30) as r:
+        body = r.read()
+    assert len(body) == total
+    assert body[:2048] == expect(0, 2048)
+    assert body[-64:] == expect(total - 64, total)
+    # range straddling the prefix boundary
+    req = urllib.request.Request(
+        url, headers={"Range": "bytes=1000-1100"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        assert r.status == 206
+        assert r.read() == expect(1000, 1101)
+    # range fully past the prefix
+    req = urllib.request.Request(
+        url, headers={"Range": "bytes=50000-50099"})
+    with urllib.request.urlopen(req, timeout=30) as r:
+        assert r.read() == expect(50000, 50100)
