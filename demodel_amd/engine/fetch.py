@@ -254,6 +254,12 @@ def http_get(url: str, headers: dict | None = None, cafile=None,
                 else:
                     base = u.path.rsplit("/", 1)[0]
                     url = f"{u.scheme}://{u.netloc}{base}/{loc}"
+                # hub -> CDN hop: never forward credentials to a
+                # DIFFERENT host (presigned CDN URLs reject them)
+                if headers and urlsplit(url).netloc != u.netloc:
+                    headers = {k: v for k, v in headers.items()
+                               if k.lower() not in ("authorization",
+                                                    "cookie")}
                 continue
         src = BlobSource(sock, resp, leftover, is_tls)
         src.resp.url = url

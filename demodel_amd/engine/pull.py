@@ -537,7 +537,7 @@ def pull_hf_stream(repo: str, rev: str = "main",
                    digest_map: dict[str, bytes] | None = None,
                    peer_verify: bool = False, batched: bool = False,
                    on_range=None, repo_type: str = "model",
-                   registry=None):
+                   registry=None, token: str | None = None):
     """Streaming pull: returns (info, names, generator) where the
     generator yields each PulledFile AS IT FINISHES landing, so a
     consumer (e.g. stream_dataset's GPU decompression) overlaps with the
@@ -548,12 +548,19 @@ def pull_hf_stream(repo: str, rev: str = "main",
     unstarted pulls."""
     endpoint = (endpoint or os.environ.get("HF_ENDPOINT")
                 or HF_DEFAULT_ENDPOINT).rstrip("/")
+    # gated/private repos: Bearer token on hub requests; fetch strips
+    # it on the cross-host CDN hop (presigned URLs reject credentials)
+    if token is None:
+        token = (os.environ.get("HF_TOKEN")
+                 or os.environ.get("HUGGING_FACE_HUB_TOKEN"))
+    auth = {"Authorization": f"Bearer {token}"} if token else None
     # HF dataset repos live under /api/datasets and resolve their blobs
     # at /datasets/{repo}/resolve/... (models have no path prefix)
     api = "datasets" if repo_type == "dataset" else "models"
     prefix = "datasets/" if repo_type == "dataset" else ""
     info = fetch.get_json(f"{endpoint}/api/{api}/{repo}/revision/{rev}",
-                          cafile=cafile, insecure=insecure)
+                          cafile=cafile, insecure=insecure,
+                          headers=auth)
     names = [s["rfilename"] for s in info.get("siblings", [])]
     if patterns:
         names = [n for n in names
@@ -603,7 +610,7 @@ def pull_hf_stream(repo: str, rev: str = "main",
                     futs[ex.submit(
                         _pull_blob, landers, n,
                         f"{endpoint}/{prefix}{repo}/resolve/{rev}/{n}",
-                        None, verify, cafile, insecure, None,
+                        None, verify, cafile, insecure, auth,
                         exp, seg_ex, vc, on_range)] = n
                 try:
                     if batched:
@@ -638,7 +645,8 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
             slab_bytes: int = 32 << 20,
             digest_map: dict[str, bytes] | None = None,
             peer_verify: bool = False, on_range=None,
-            repo_type: str = "model", registry=None) -> PullResult:
+            repo_type: str = "model", registry=None,
+            token: str | None = None) -> PullResult:
     """peer_verify: when `endpoint` is another demodel node, fetch its
     recorded chunk digests per blob and GPU-verify the pull against them
     (verified distribution).  on_range: progress hook, see _pull_blob.
@@ -651,7 +659,8 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
         workers=workers, verify=verify, cafile=cafile, insecure=insecure,
         patterns=patterns, landers=landers, slab_bytes=slab_bytes,
         digest_map=digest_map, peer_verify=peer_verify,
-        on_range=on_range, repo_type=repo_type, registry=registry)
+        on_range=on_range, repo_type=repo_type, registry=registry,
+        token=token)
     result = PullResult(spec=f"hf://{repo}@{rev}")
     result.files = list(gen)
     result.files.sort(key=lambda f: f.name)

@@ -56,6 +56,14 @@ class FakeOrigin:
         # fault injection: path-substrings that hang (no response bytes)
         # until the client gives up — exercises proxy read timeouts
         self.hang_once: set[str] = set()
+        # gated-repo emulation: /api and /resolve require this Bearer
+        # token (401 without); /cdn REJECTS requests that still carry
+        # Authorization (403 — presigned-URL semantics, like S3)
+        self.require_token: str | None = None
+        # absolute base for blob redirects (e.g. a second FakeOrigin
+        # playing the CDN host — the real hub/S3 topology); None keeps
+        # same-host /cdn/ paths
+        self.cdn_base: str | None = None
 
     # ------------------------------------------------------------------ #
     # content registration
@@ -170,6 +178,15 @@ class FakeOrigin:
         path = head.target.split("?")[0]
         parts = [p for p in path.split("/") if p]
 
+        if self.require_token is not None:
+            auth = head.get("authorization", "")
+            if parts and parts[0] == "cdn":
+                if auth:
+                    return await self._error(writer, 403)
+            elif parts and parts[0] in ("api",) or "resolve" in parts:
+                if auth != f"Bearer {self.require_token}":
+                    return await self._error(writer, 401)
+
         # ---- POST/PUT echo (proxy pass-through tests): digest the
         # streamed request body, reply with its size + sha256 ----
         if parts == ["echo"] and head.method in ("POST", "PUT"):
@@ -277,7 +294,8 @@ class FakeOrigin:
                      ("X-Linked-Etag", f'"{etag}"'),
                      ("X-Linked-Size", str(size))]
             if self.redirect_blobs:
-                loc = f"/cdn/{repo_id}/{repo['sha']}/{fname}"
+                loc = ((self.cdn_base or "")
+                       + f"/cdn/{repo_id}/{repo['sha']}/{fname}")
                 return await self._reply(
                     writer, head, 302,
                     extra + [("Location", loc),

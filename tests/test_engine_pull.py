@@ -336,3 +336,45 @@ def test_virtual_origin_prefix_and_ranges(stack, tmp_path):
         url, headers={"Range": "bytes=50000-50099"})
     with urllib.request.urlopen(req, timeout=30) as r:
         assert r.read() == expect(50000, 50100)
+
+
+def test_gated_repo_token_flow(stack, tmp_path, monkeypatch):
+    """Private/gated repos: the engine sends the HF token to hub
+    endpoints, and fetch STRIPS it on the cross-host CDN redirect
+    (presigned-URL hosts reject stray credentials)."""
+    import pytest as _pytest
+
+    from demodel_amd.engine import fetch as fetch_mod
+
+    from demodel_amd.testing.origin import FakeOrigin
+
+    data = os.urandom(100_000)
+    p = tmp_path / "g.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/gated", {"g.bin": str(p)})
+    stack.origin.require_token = "s3cret"
+    # a SECOND origin plays the CDN host (the real hub/S3 topology):
+    # it rejects requests that still carry Authorization
+    cdn = FakeOrigin(str(tmp_path))
+    cdn.hf_repos = stack.origin.hf_repos
+    cdn.require_token = "s3cret"
+    cdn_port = stack.lt.call(cdn.start())
+    stack.origin.cdn_base = f"http://127.0.0.1:{cdn_port}"
+
+    # without a token: 401 on the repo-info call
+    monkeypatch.delenv("HF_TOKEN", raising=False)
+    monkeypatch.delenv("HUGGING_FACE_HUB_TOKEN", raising=False)
+    with _pytest.raises(fetch_mod.FetchError):
+        pull_mod.pull_hf("org/gated", endpoint=stack.origin_base,
+                         workers=1)
+    # with the token (env surface), the pull succeeds END TO END —
+    # which proves the CDN hop did NOT receive Authorization (the
+    # origin 403s any /cdn request that carries it)
+    monkeypatch.setenv("HF_TOKEN", "s3cret")
+    res = pull_mod.pull_hf("org/gated", endpoint=stack.origin_base,
+                           workers=1)
+    f = [x for x in res.files if x.name == "g.bin"][0]
+    assert bytes(f.blob.buffer) == data
+    # and the CDN host really served it (cross-host redirect taken)
+    assert any("/cdn/" in r for r in cdn.requests)
+    stack.lt.call(cdn.close())
