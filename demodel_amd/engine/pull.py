@@ -297,6 +297,15 @@ def _pull_blob(landers: LanderPool, name: str, url: str,
                         first[0] = False
                         return src.fill
                     rng_headers = dict(headers or {})
+                    # the resume hits the FINAL (post-redirect) URL:
+                    # drop credentials if that host differs from the
+                    # hub (presigned CDN URLs reject them)
+                    from urllib.parse import urlsplit as _us
+
+                    if _us(src.resp.url).netloc != _us(url).netloc:
+                        for k in list(rng_headers):
+                            if k.lower() in ("authorization", "cookie"):
+                                del rng_headers[k]
                     if offset:
                         rng_headers["Range"] = f"bytes={offset}-"
                     s2 = fetch.http_get(src.resp.url, cafile=cafile,

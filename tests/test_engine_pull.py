@@ -378,3 +378,29 @@ def test_gated_repo_token_flow(stack, tmp_path, monkeypatch):
     # and the CDN host really served it (cross-host redirect taken)
     assert any("/cdn/" in r for r in cdn.requests)
     stack.lt.call(cdn.close())
+
+
+def test_gated_repo_resume_does_not_leak_token(stack, tmp_path,
+                                               monkeypatch):
+    """A mid-stream drop resumes against the CDN host directly — the
+    token must NOT ride along (the cdn origin 403s it)."""
+    from demodel_amd.testing.origin import FakeOrigin
+
+    data = os.urandom(2 << 20)
+    p = tmp_path / "gr.bin"
+    p.write_bytes(data)
+    stack.origin.add_hf_repo("org/gres", {"gr.bin": str(p)})
+    stack.origin.require_token = "tok"
+    cdn = FakeOrigin(str(tmp_path))
+    cdn.hf_repos = stack.origin.hf_repos
+    cdn.require_token = "tok"
+    cdn.drop_once["gr.bin"] = 300 << 10  # resume mid-blob on the CDN
+    cdn_port = stack.lt.call(cdn.start())
+    stack.origin.cdn_base = f"http://127.0.0.1:{cdn_port}"
+    monkeypatch.setenv("HF_TOKEN", "tok")
+    res = pull_mod.pull_hf("org/gres", endpoint=stack.origin_base,
+                           workers=1, slab_bytes=256 << 10)
+    f = [x for x in res.files if x.name == "gr.bin"][0]
+    assert bytes(f.blob.buffer) == data
+    assert not cdn.drop_once  # the drop fired -> a resume happened
+    stack.lt.call(cdn.close())
