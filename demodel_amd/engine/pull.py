@@ -683,6 +683,57 @@ def pull_hf(repo: str, rev: str = "main", endpoint: str | None = None,
     return result
 
 
+def _registry_token(www_auth: str, cafile=None,
+                    insecure: bool = False) -> str | None:
+    """Docker-registry Bearer handshake: parse a 401's
+    ``WWW-Authenticate: Bearer realm="...",service="...",scope="..."``
+    and fetch a token from the realm (how ollama.com / docker
+    registries gate private models; registry.ollama.ai's public
+    library never 401s)."""
+    import re
+    from urllib.parse import quote
+
+    if not www_auth or not www_auth.lower().startswith("bearer"):
+        return None
+    parts = dict(re.findall(r'(\w+)="([^"]*)"', www_auth))
+    realm = parts.get("realm")
+    if not realm:
+        return None
+    q = "&".join(f"{k}={quote(parts[k], safe='')}"
+                 for k in ("service", "scope") if parts.get(k))
+    try:
+        obj = fetch.get_json(realm + (f"?{q}" if q else ""),
+                             cafile=cafile, insecure=insecure)
+    except (fetch.FetchError, OSError, ValueError):
+        return None
+    return obj.get("token") or obj.get("access_token")
+
+
+def _registry_get_json(url: str, cafile, insecure,
+                       auth: dict | None) -> tuple[dict, dict | None]:
+    """GET a registry JSON document, performing the Bearer handshake on
+    a 401.  Returns (json, auth_headers_used)."""
+    import json as _json
+
+    src = fetch.http_get(url, cafile=cafile, insecure=insecure,
+                         headers=auth)
+    try:
+        if src.status == 401 and auth is None:
+            tok = _registry_token(src.resp.get("www-authenticate", ""),
+                                  cafile, insecure)
+            if tok is not None:
+                src.close()
+                auth = {"Authorization": f"Bearer {tok}"}
+                src = fetch.http_get(url, cafile=cafile,
+                                     insecure=insecure, headers=auth)
+        if src.status != 200:
+            raise fetch.FetchError(
+                f"GET {url} -> {src.status} {src.resp.reason}")
+        return _json.loads(src.read_all(limit=256 << 20)), auth
+    finally:
+        src.close()
+
+
 def pull_ollama(name: str, tag: str = "latest",
                 endpoint: str | None = None, device_index: int = 0,
                 workers: int = 4, verify: str = "digest",
@@ -703,8 +754,8 @@ def pull_ollama(name: str, tag: str = "latest",
     if "/" not in name:
         name = f"library/{name}"
     t0 = time.perf_counter()
-    manifest = fetch.get_json(f"{endpoint}/v2/{name}/manifests/{tag}",
-                              cafile=cafile, insecure=insecure)
+    manifest, reg_auth = _registry_get_json(
+        f"{endpoint}/v2/{name}/manifests/{tag}", cafile, insecure, None)
     layers = list(manifest.get("layers", []))
     if manifest.get("config"):
         layers.append(manifest["config"])
@@ -729,7 +780,7 @@ def pull_ollama(name: str, tag: str = "latest",
             is_model = (layer.get("mediaType")
                         == "application/vnd.ollama.image.model")
             futs[ex.submit(_pull_blob, landers, digest, url, expected,
-                           verify, cafile, insecure, None, None,
+                           verify, cafile, insecure, reg_auth, None,
                            seg_ex, None,
                            pd.on_range if (pd and is_model) else None)
                  ] = layer

@@ -64,6 +64,10 @@ class FakeOrigin:
         # playing the CDN host — the real hub/S3 topology); None keeps
         # same-host /cdn/ paths
         self.cdn_base: str | None = None
+        # docker-registry Bearer gate: /v2/ requests 401 with a
+        # WWW-Authenticate pointing at our /token endpoint until the
+        # client presents the issued token (the ollama.com/docker flow)
+        self.docker_token: str | None = None
 
     # ------------------------------------------------------------------ #
     # content registration
@@ -203,8 +207,31 @@ class FakeOrigin:
                 writer, head, 200,
                 [("Content-Type", "application/json")], body)
 
+        # ---- docker token endpoint ----
+        if parts and parts[0] == "token" and self.docker_token:
+            body = json.dumps({"token": self.docker_token}).encode()
+            return await self._reply(
+                writer, head, 200,
+                [("Content-Type", "application/json")], body)
+
         # ---- Ollama registry v2 ----
         if parts and parts[0] == "v2":
+            if (self.docker_token
+                    and head.get("authorization", "")
+                    != f"Bearer {self.docker_token}"):
+                scheme = "https" if self.tls_ctx else "http"
+                hdr = (f'Bearer realm="{scheme}://127.0.0.1:{self.port}'
+                       f'/token",service="fake-registry",'
+                       f'scope="repository:{"/".join(parts[1:-2])}:pull"')
+                body = b'{"errors": [{"code": "UNAUTHORIZED"}]}'
+                out = ResponseHead(
+                    "HTTP/1.1", 401, "Unauthorized",
+                    [("Content-Type", "application/json"),
+                     ("Content-Length", str(len(body))),
+                     ("WWW-Authenticate", hdr)])
+                writer.write(http1.serialize_response(out) + body)
+                await writer.drain()
+                return
             if len(parts) >= 4 and parts[-2] == "manifests":
                 name = "/".join(parts[1:-2])
                 tag = parts[-1]

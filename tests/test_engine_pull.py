@@ -404,3 +404,25 @@ def test_gated_repo_resume_does_not_leak_token(stack, tmp_path,
     assert bytes(f.blob.buffer) == data
     assert not cdn.drop_once  # the drop fired -> a resume happened
     stack.lt.call(cdn.close())
+
+
+def test_ollama_registry_token_handshake(stack, tmp_path):
+    """Docker-registry Bearer flow (ollama.com/private models): a 401
+    with WWW-Authenticate drives a token fetch from the realm; manifest
+    AND blob requests then carry the token."""
+    from demodel_amd.engine.formats import gguf as gguf_mod
+
+    gg_path = tmp_path / "tok.gguf"
+    gguf_mod.build_file(str(gg_path), [("w.weight", (64, 4), 2)])
+    stack.origin.add_ollama_model(
+        "library/gated", "latest",
+        [("application/vnd.ollama.image.model", str(gg_path))])
+    stack.origin.docker_token = "registry-jwt"
+    res = pull_mod.pull_ollama("gated", "latest",
+                               endpoint=stack.origin_base,
+                               verify="digest", workers=1)
+    assert all(f.digest_ok for f in res.files)
+    assert "gguf" in res.meta
+    # the token endpoint was hit exactly once
+    assert sum(1 for r in stack.origin.requests
+               if r.startswith("GET /token")) == 1
