@@ -201,7 +201,7 @@ def test_upstream_pool_unit(stack, tmp_path):
     host, port = "127.0.0.1", stack.origin_port
 
     async def scenario():
-        pool = UpstreamPool(max_idle_per_key=1, idle_ttl=0.05)
+        pool = UpstreamPool(max_idle_per_key=1, idle_ttl=0.3)
         r1, w1, reused = await pool.acquire(host, port, None)
         assert reused is False
         pool.release(host, port, False, r1, w1)
@@ -214,7 +214,7 @@ def test_upstream_pool_unit(stack, tmp_path):
         assert len(pool._idle[(host, port, False)]) == 1
         assert w3.is_closing()  # the overflow one was closed
         # TTL expiry: after the idle window the conn is discarded
-        await asyncio.sleep(0.1)
+        await asyncio.sleep(0.9)
         r4, w4, reused = await pool.acquire(host, port, None)
         assert reused is False
         pool.close_all()
