@@ -134,3 +134,70 @@ def test_sharded_fanout_gloo_world2(tmp_path):
     torch.multiprocessing.spawn(
         _worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
     assert (tmp_path / "ok0").exists() and (tmp_path / "ok1").exists()
+
+
+def _worker_http_resume(rank, world, port, tmpdir):
+    """Sharded fan-out where each rank's FIRST owned blob drops
+    mid-stream: Range-resume must complete inside the collective run
+    (the realistic multi-GPU failure mode — one flaky origin conn must
+    not poison the broadcast schedule)."""
+    import torch.distributed as dist
+
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}",
+        rank=rank, world_size=world)
+    try:
+        from demodel_amd.engine.pull import LanderPool, _pull_blob
+        from demodel_amd.parallel.fanout import (shard_assignment,
+                                                 sharded_pull_fanout)
+        from demodel_amd.testing.origin import FakeOrigin
+        from helpers import LoopThread
+
+        names = sorted(n for n in os.listdir(tmpdir)
+                       if n.endswith(".bin"))
+        files = {n: os.path.join(tmpdir, n) for n in names}
+        sizes = [(n, os.path.getsize(p)) for n, p in files.items()]
+        lt = LoopThread()
+        origin = FakeOrigin(tmpdir)
+        origin.add_hf_repo("org/m", files)
+        oport = lt.call(origin.start())
+        landers = LanderPool(0, gpu=False,
+                             slab_bytes=64 << 10)  # small slabs: real
+        plan = shard_assignment(sizes, world, rank)  # resume mid-file
+        if plan.my_files:
+            origin.drop_once[plan.my_files[0]] = 100 << 10
+
+        def pull_one(name):
+            pf = _pull_blob(
+                landers, name,
+                f"http://127.0.0.1:{oport}/org/m/resolve/main/{name}",
+                None, "chunked", None, False)
+            return pf.blob.torch_u8()
+
+        def alloc(nb):
+            return torch.zeros(nb, dtype=torch.uint8)
+
+        out = sharded_pull_fanout(plan, pull_one, alloc,
+                                  bucket_bytes=128 << 10)
+        assert not origin.drop_once  # the fault fired on this rank
+        for n, p in files.items():
+            want = open(p, "rb").read()
+            assert bytes(out[n].numpy().tobytes()) == want, n
+        with open(os.path.join(tmpdir, f"okr{rank}"), "w") as f:
+            f.write("ok")
+        lt.call(origin.close())
+        lt.stop()
+    finally:
+        dist.destroy_process_group()
+
+
+def test_fanout_survives_mid_pull_drops_world2(tmp_path):
+    rng = __import__("random").Random(5)
+    for i in range(4):
+        (tmp_path / f"m{i}.bin").write_bytes(
+            bytes(rng.getrandbits(8) for _ in range(300_000 + i * 70_000)))
+    port = _free_port()
+    torch.multiprocessing.spawn(
+        _worker_http_resume, args=(2, port, str(tmp_path)), nprocs=2,
+        join=True)
+    assert (tmp_path / "okr0").exists() and (tmp_path / "okr1").exists()
